@@ -1,0 +1,176 @@
+// Python bindings for the gfx950 flat-arena kernels (flat_ops.hip).
+//
+// Host side is written against the native ROCm/HIP ATen surface of
+// PyTorch-ROCm (c10::hip) — no CUDA compatibility layer.  All launches go
+// onto the current torch stream; no host synchronization anywhere.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void launch_pseudo_grad(float*, const float*, const float*, float, long long,
+                        hipStream_t);
+void launch_axpy(float*, const float*, float, long long, hipStream_t);
+void launch_scale(float*, float, long long, hipStream_t);
+void launch_sum_sumsq(const float*, long long, double*, hipStream_t);
+void launch_clip_apply(float*, long long, const double*, float, float, float*,
+                       hipStream_t);
+void launch_add_gaussian_noise(float*, long long, float, unsigned long long,
+                               unsigned long long, hipStream_t);
+void launch_sgd_step(float*, const float*, float*, float, float, float, float,
+                     int, int, long long, hipStream_t);
+void launch_adam_step(float*, const float*, float*, float*, float*, float,
+                      float, float, float, float, float, float, int, int,
+                      long long, hipStream_t);
+void launch_adamax_step(float*, const float*, float*, float*, float, float,
+                        float, float, float, float, long long, hipStream_t);
+void launch_segmented_sqnorm(const float*, const long long*, int, double*,
+                             hipStream_t);
+void launch_quant_bin_mask(float*, long long, const float*, const float*,
+                           const float*, int, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_flat(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+}
+
+void pseudo_grad(torch::Tensor out, torch::Tensor ws, torch::Tensor wt,
+                 double weight) {
+  check_flat(out, "out"); check_flat(ws, "ws"); check_flat(wt, "wt");
+  TORCH_CHECK(out.numel() == ws.numel() && ws.numel() == wt.numel());
+  launch_pseudo_grad(out.data_ptr<float>(), ws.data_ptr<float>(),
+                     wt.data_ptr<float>(), (float)weight, out.numel(),
+                     cur_stream());
+}
+
+void axpy(torch::Tensor y, torch::Tensor x, double alpha) {
+  check_flat(y, "y"); check_flat(x, "x");
+  TORCH_CHECK(y.numel() == x.numel());
+  launch_axpy(y.data_ptr<float>(), x.data_ptr<float>(), (float)alpha,
+              y.numel(), cur_stream());
+}
+
+void scale(torch::Tensor x, double alpha) {
+  check_flat(x, "x");
+  launch_scale(x.data_ptr<float>(), (float)alpha, x.numel(), cur_stream());
+}
+
+torch::Tensor sum_sumsq(torch::Tensor x) {
+  check_flat(x, "x");
+  auto acc = torch::zeros({2}, x.options().dtype(torch::kFloat64));
+  launch_sum_sumsq(x.data_ptr<float>(), x.numel(), acc.data_ptr<double>(),
+                   cur_stream());
+  return acc.to(torch::kFloat32);
+}
+
+torch::Tensor clip_by_norm(torch::Tensor x, double max_norm, double eps) {
+  check_flat(x, "x");
+  auto acc = torch::zeros({2}, x.options().dtype(torch::kFloat64));
+  auto norm = torch::empty({}, x.options());
+  launch_sum_sumsq(x.data_ptr<float>(), x.numel(), acc.data_ptr<double>(),
+                   cur_stream());
+  launch_clip_apply(x.data_ptr<float>(), x.numel(), acc.data_ptr<double>(),
+                    (float)max_norm, (float)eps, norm.data_ptr<float>(),
+                    cur_stream());
+  return norm;
+}
+
+void add_gaussian_noise(torch::Tensor x, double sigma, int64_t seed,
+                        int64_t offset) {
+  check_flat(x, "x");
+  launch_add_gaussian_noise(x.data_ptr<float>(), x.numel(), (float)sigma,
+                            (unsigned long long)seed,
+                            (unsigned long long)offset, cur_stream());
+}
+
+void sgd_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
+              double momentum, double dampening, double weight_decay,
+              bool nesterov, bool first_step) {
+  check_flat(p, "p"); check_flat(g, "g");
+  if (momentum != 0.0) {
+    check_flat(buf, "buf");
+    TORCH_CHECK(buf.numel() == p.numel(), "momentum buffer size mismatch");
+  }
+  launch_sgd_step(p.data_ptr<float>(), g.data_ptr<float>(),
+                  buf.numel() ? buf.data_ptr<float>() : nullptr, (float)lr,
+                  (float)momentum, (float)dampening, (float)weight_decay,
+                  nesterov ? 1 : 0, first_step ? 1 : 0, p.numel(),
+                  cur_stream());
+}
+
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, torch::Tensor vmax, int64_t step, double lr,
+               double beta1, double beta2, double eps, double weight_decay,
+               bool amsgrad, bool adamw) {
+  check_flat(p, "p"); check_flat(g, "g"); check_flat(m, "m"); check_flat(v, "v");
+  if (amsgrad) check_flat(vmax, "vmax");
+  float bc1 = 1.0f - (float)std::pow(beta1, (double)step);
+  float bc2 = 1.0f - (float)std::pow(beta2, (double)step);
+  launch_adam_step(p.data_ptr<float>(), g.data_ptr<float>(),
+                   m.data_ptr<float>(), v.data_ptr<float>(),
+                   vmax.numel() ? vmax.data_ptr<float>() : nullptr, (float)lr,
+                   (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                   bc1, bc2, amsgrad ? 1 : 0, adamw ? 1 : 0, p.numel(),
+                   cur_stream());
+}
+
+void adamax_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                 torch::Tensor u, int64_t step, double lr, double beta1,
+                 double beta2, double eps, double weight_decay) {
+  check_flat(p, "p"); check_flat(g, "g"); check_flat(m, "m"); check_flat(u, "u");
+  float bc1 = 1.0f - (float)std::pow(beta1, (double)step);
+  launch_adamax_step(p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), u.data_ptr<float>(), (float)lr,
+                     (float)beta1, (float)beta2, (float)eps,
+                     (float)weight_decay, bc1, p.numel(), cur_stream());
+}
+
+torch::Tensor segmented_sqnorm(torch::Tensor x, torch::Tensor seg_offsets) {
+  check_flat(x, "x");
+  TORCH_CHECK(seg_offsets.is_cuda() && seg_offsets.is_contiguous() &&
+              seg_offsets.scalar_type() == torch::kInt64,
+              "seg_offsets must be contiguous int64 on GPU");
+  int n_segs = (int)seg_offsets.numel() - 1;
+  auto out = torch::zeros({n_segs}, x.options().dtype(torch::kFloat64));
+  launch_segmented_sqnorm(
+      x.data_ptr<float>(),
+      reinterpret_cast<const long long*>(seg_offsets.data_ptr<int64_t>()),
+      n_segs, out.data_ptr<double>(), cur_stream());
+  return out.to(torch::kFloat32);
+}
+
+void quant_bin_mask(torch::Tensor x, torch::Tensor min_t, torch::Tensor max_t,
+                    torch::Tensor thresh_t, int64_t n_bins) {
+  check_flat(x, "x");
+  check_flat(min_t, "min_t"); check_flat(max_t, "max_t");
+  check_flat(thresh_t, "thresh_t");
+  launch_quant_bin_mask(x.data_ptr<float>(), x.numel(),
+                        min_t.data_ptr<float>(), max_t.data_ptr<float>(),
+                        thresh_t.data_ptr<float>(), (int)n_bins, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "msrflute_amd gfx950 flat-arena kernels";
+  m.def("pseudo_grad", &pseudo_grad);
+  m.def("axpy", &axpy);
+  m.def("scale", &scale);
+  m.def("sum_sumsq", &sum_sumsq);
+  m.def("clip_by_norm", &clip_by_norm);
+  m.def("add_gaussian_noise", &add_gaussian_noise);
+  m.def("sgd_step", &sgd_step);
+  m.def("adam_step", &adam_step);
+  m.def("adamax_step", &adamax_step);
+  m.def("segmented_sqnorm", &segmented_sqnorm);
+  m.def("quant_bin_mask", &quant_bin_mask);
+}

@@ -1,0 +1,152 @@
+"""Flat-arena op dispatch: CDNA4 HIP kernels on GPU, torch reference on CPU.
+
+The HIP extension (``msrflute_amd._C``, built in-tree from
+``msrflute_amd/csrc`` for gfx950) is REQUIRED whenever tensors live on a
+GPU — there is no silent eager fallback on device (a GPU box without the
+extension raises, per the build contract).  CPU tensors always use the
+torch reference implementations in ``msrflute_amd.ops.reference``, which
+also serve as the numerics oracle for the kernels.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+from . import reference as ref
+
+_C = None
+_C_ERR: Optional[str] = None
+try:
+    from msrflute_amd import _C as _C  # built by setup.py build_ext --inplace
+except ImportError as e:  # extension not built — allowed on CPU only
+    _C = None
+    _C_ERR = str(e)
+
+HAS_EXT = _C is not None
+
+# Escape hatch for A/B benchmarking the eager path on GPU (never default).
+_ALLOW_EAGER_GPU = os.environ.get("MSRFLUTE_AMD_ALLOW_EAGER_GPU", "0") == "1"
+
+
+def _use_ext(*tensors: torch.Tensor) -> bool:
+    if not tensors[0].is_cuda:
+        return False
+    if HAS_EXT:
+        return True
+    if _ALLOW_EAGER_GPU:
+        return False
+    raise RuntimeError(
+        "msrflute_amd._C HIP extension is not built but tensors are on GPU "
+        f"(import error: {_C_ERR}). Build it with `python setup.py "
+        "build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950); refusing to fall "
+        "back to eager ops on device.")
+
+
+def pseudo_grad(out, w_server, w_trained, weight: float):
+    if _use_ext(out):
+        _C.pseudo_grad(out, w_server, w_trained, float(weight))
+        return out
+    return ref.pseudo_grad(out, w_server, w_trained, weight)
+
+
+def axpy(y, x, alpha: float = 1.0):
+    if _use_ext(y):
+        _C.axpy(y, x, float(alpha))
+        return y
+    return ref.axpy(y, x, alpha)
+
+
+def scale(x, alpha: float):
+    if _use_ext(x):
+        _C.scale(x, float(alpha))
+        return x
+    return ref.scale(x, alpha)
+
+
+def sum_sumsq(x) -> torch.Tensor:
+    if _use_ext(x):
+        return _C.sum_sumsq(x)
+    return ref.sum_sumsq(x)
+
+
+def l2_norm(x) -> torch.Tensor:
+    if _use_ext(x):
+        return _C.sum_sumsq(x)[1].sqrt()
+    return ref.l2_norm(x)
+
+
+def clip_by_norm(x, max_norm: float, eps: float = 1e-6) -> torch.Tensor:
+    """Returns the pre-clip L2 norm as a 0-dim device tensor (no host sync)."""
+    if _use_ext(x):
+        return _C.clip_by_norm(x, float(max_norm), float(eps))
+    return ref.clip_by_norm(x, max_norm, eps)
+
+
+def add_gaussian_noise(x, sigma: float, seed: int, offset: int = 0,
+                       generator=None):
+    """x += sigma*N(0,1).  On GPU the noise stream is a Philox keyed by
+    (seed, offset, element index) — deterministic for a given grid-independent
+    (seed, offset)."""
+    if _use_ext(x):
+        _C.add_gaussian_noise(x, float(sigma), int(seed), int(offset))
+        return x
+    return ref.add_gaussian_noise(x, sigma, generator=generator)
+
+
+def sgd_step(param, grad, momentum_buf, *, lr, momentum=0.0, dampening=0.0,
+             weight_decay=0.0, nesterov=False, first_step=False):
+    if _use_ext(param):
+        _C.sgd_step(param, grad,
+                    momentum_buf if momentum_buf is not None else param.new_empty(0),
+                    float(lr), float(momentum), float(dampening),
+                    float(weight_decay), bool(nesterov), bool(first_step))
+        return
+    ref.sgd_step(param, grad, momentum_buf, lr=lr, momentum=momentum,
+                 dampening=dampening, weight_decay=weight_decay,
+                 nesterov=nesterov, first_step=first_step)
+
+
+def adam_step(param, grad, exp_avg, exp_avg_sq, max_exp_avg_sq=None, *, step,
+              lr, beta1=0.9, beta2=0.999, eps=1e-8, weight_decay=0.0,
+              amsgrad=False, adamw=False):
+    if _use_ext(param):
+        _C.adam_step(param, grad, exp_avg, exp_avg_sq,
+                     max_exp_avg_sq if max_exp_avg_sq is not None else param.new_empty(0),
+                     int(step), float(lr), float(beta1), float(beta2),
+                     float(eps), float(weight_decay), bool(amsgrad), bool(adamw))
+        return
+    ref.adam_step(param, grad, exp_avg, exp_avg_sq, max_exp_avg_sq, step=step,
+                  lr=lr, beta1=beta1, beta2=beta2, eps=eps,
+                  weight_decay=weight_decay, amsgrad=amsgrad, adamw=adamw)
+
+
+def adamax_step(param, grad, exp_avg, exp_inf, *, step, lr, beta1=0.9,
+                beta2=0.999, eps=1e-8, weight_decay=0.0):
+    if _use_ext(param):
+        _C.adamax_step(param, grad, exp_avg, exp_inf, int(step), float(lr),
+                       float(beta1), float(beta2), float(eps), float(weight_decay))
+        return
+    ref.adamax_step(param, grad, exp_avg, exp_inf, step=step, lr=lr,
+                    beta1=beta1, beta2=beta2, eps=eps, weight_decay=weight_decay)
+
+
+def segmented_sqnorm(x, seg_offsets) -> torch.Tensor:
+    if _use_ext(x):
+        return _C.segmented_sqnorm(x, seg_offsets)
+    return ref.segmented_sqnorm(x, seg_offsets)
+
+
+def quantize_dequantize(x, n_bins: int, threshold_quantile: float):
+    if _use_ext(x):
+        # Stats via torch's device reductions (stay on device), fused
+        # bin+mask via the HIP kernel — no host synchronization.
+        min_t, max_t = torch.aminmax(x)
+        thresh_t = torch.quantile(x.abs(), threshold_quantile)
+        _C.quant_bin_mask(x, min_t.reshape(1).contiguous(),
+                          max_t.reshape(1).contiguous(),
+                          thresh_t.reshape(1).float().contiguous(), int(n_bins))
+        return x
+    return ref.quantize_dequantize(x, n_bins, threshold_quantile)
