@@ -1,0 +1,161 @@
+"""Symmetric-replica distributed runtime (RCCL over xGMI).
+
+The reference runs a hub-and-spoke pool: rank 0 orchestrates, ranks 1..K-1
+wait for per-client P2P commands, every model movement is a per-tensor
+point-to-point send (reference core/federated.py).
+
+This runtime replaces that with a *symmetric replicated* design built for
+one MI355X node (one process per GPU, backend "nccl" == RCCL on ROCm):
+
+* every rank holds an identical replica of the server state (model arena +
+  server optimizer + schedulers) and deterministically derives the same
+  round decisions (client sampling, partitioning) from a shared seed —
+  so there is NO model broadcast per round at all;
+* every rank trains its partition of the round's sampled clients locally,
+  accumulating weighted pseudo-gradients into its grad arena;
+* the only bulk communication per round is ONE all-reduce of the flat
+  gradient arena (+ one tiny all-reduce for scalar sums), after which every
+  rank applies the identical server-optimizer update — mathematically
+  identical to the reference's sum-then-normalize (fedavg.py:140-147)
+  because aggregation is a commutative weighted sum;
+* per-client scalar metadata travels once per round via all_gather_object.
+
+Ring all-reduce over xGMI is per-link bound (7 links × ~153 GB/s), so the
+one fused arena all-reduce (vs the reference's 3+T messages per client)
+is the right shape for this fabric.
+
+A `gloo` backend path keeps the same code running on CPU for tests and the
+plumbing config (BASELINE.json config 1).
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+import random
+from typing import Any, List, Optional, Sequence
+
+import torch
+import torch.distributed as dist
+
+from ..utils import print_rank
+
+
+def rank() -> int:
+    return int(os.environ.get("RANK", 0))
+
+
+def local_rank() -> int:
+    return int(os.environ.get("LOCAL_RANK", 0))
+
+
+def size() -> int:
+    return int(os.environ.get("WORLD_SIZE", 1))
+
+
+class FedRuntime:
+    """Process-group wrapper + deterministic round derivations."""
+
+    def __init__(self, backend: str = "nccl", seed: int = 0):
+        self.backend = backend
+        self.seed = seed
+        self.rank = rank()
+        self.size = size()
+        self.local_rank = local_rank()
+        self.initialized = False
+        if self.size > 1 and not dist.is_initialized():
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29500")
+            dist.init_process_group(
+                backend=backend, rank=self.rank, world_size=self.size,
+                timeout=datetime.timedelta(minutes=30))
+            self.initialized = True
+        if backend == "nccl" and torch.cuda.is_available():
+            torch.cuda.set_device(self.local_rank % torch.cuda.device_count())
+
+    # -- collectives (no-ops at world_size == 1) --------------------------
+    def all_reduce_(self, t: torch.Tensor):
+        if self.size > 1:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        return t
+
+    def broadcast_(self, t: torch.Tensor, src: int = 0):
+        if self.size > 1:
+            dist.broadcast(t, src=src)
+        return t
+
+    def all_gather_object(self, obj: Any) -> List[Any]:
+        if self.size == 1:
+            return [obj]
+        out: List[Any] = [None] * self.size
+        dist.all_gather_object(out, obj)
+        return out
+
+    def barrier(self):
+        if self.size > 1:
+            if self.backend == "nccl" and torch.cuda.is_available():
+                dist.barrier(device_ids=[torch.cuda.current_device()])
+            else:
+                dist.barrier()
+
+    def shutdown(self):
+        if self.initialized and dist.is_initialized():
+            dist.destroy_process_group()
+            self.initialized = False
+
+    # -- deterministic round derivations ----------------------------------
+    def round_rng(self, round_no: int, salt: int = 0) -> random.Random:
+        """Same generator on every rank for a given round — replaces the
+        reference's rank-0 ``random.sample`` + broadcast of decisions."""
+        return random.Random((self.seed * 1_000_003 + salt) * 2_654_435_761
+                             + round_no)
+
+    def sample_clients(self, client_idx_list: Sequence[int], n: int,
+                       round_no: int) -> List[int]:
+        """Round client sampling (reference: core/server.py:301-302)."""
+        if n <= 0 or n >= len(client_idx_list):
+            return list(client_idx_list)
+        return self.round_rng(round_no, salt=1).sample(list(client_idx_list), n)
+
+    def partition(self, items: Sequence[Any],
+                  weights: Optional[Sequence[float]] = None) -> List[List[Any]]:
+        """Deterministic size-aware partition of ``items`` over all ranks:
+        greedy longest-processing-time bin packing by ``weights`` (defaults
+        to uniform).  Every rank computes the identical result (SURVEY.md
+        §7.4 item 6 — replaces the reference's dynamic work queue)."""
+        k = self.size
+        bins: List[List[Any]] = [[] for _ in range(k)]
+        loads = [0.0] * k
+        if weights is None:
+            weights = [1.0] * len(items)
+        order = sorted(range(len(items)), key=lambda i: (-weights[i], i))
+        for i in order:
+            j = min(range(k), key=lambda b: (loads[b], b))
+            bins[j].append(items[i])
+            loads[j] += weights[i]
+        return bins
+
+    def my_share(self, items: Sequence[Any],
+                 weights: Optional[Sequence[float]] = None) -> List[Any]:
+        return self.partition(items, weights)[self.rank]
+
+
+_RUNTIME: Optional[FedRuntime] = None
+
+
+def init_runtime(backend: str = "nccl", seed: int = 0) -> FedRuntime:
+    global _RUNTIME
+    if _RUNTIME is None:
+        _RUNTIME = FedRuntime(backend=backend, seed=seed)
+    return _RUNTIME
+
+
+def get_runtime() -> FedRuntime:
+    if _RUNTIME is None:
+        return init_runtime(backend="gloo" if not torch.cuda.is_available() else "nccl")
+    return _RUNTIME
+
+
+def set_runtime(rt: Optional[FedRuntime]):
+    global _RUNTIME
+    _RUNTIME = rt

@@ -1,0 +1,355 @@
+"""Client logic (reference: core/client.py).
+
+Two pieces:
+
+* ``Client`` — the lightweight per-task handle (same constructor contract
+  as the reference: client ids + config + send_gradients) with the shared
+  train-dataset cache and the data-slicing helper;
+* ``ClientExecutor`` — NEW: a per-rank persistent workspace holding ONE
+  resident model replica + flat arena + fused optimizer that every local
+  client reuses.  The reference pays per-client model construction,
+  parameter copy via Python loops and a fresh optimizer object per client
+  (client.py:280-344); here per-client setup is one flat arena copy (K13)
+  plus an optimizer-state reset.
+"""
+
+from __future__ import annotations
+
+import copy
+import logging
+import os
+import time
+from typing import Dict, List, Optional
+
+import torch
+
+from .. import ops
+from ..models import make_model
+from ..ops.arena import ParameterArena
+from ..ops.fused_optim import make_arena_optimizer
+from ..strategies import select_strategy
+from ..utils import (ScheduledSamplingScheduler, alpha_update, make_optimizer,
+                     print_rank, to_device)
+from ..utils.dataloaders_utils import (get_dataset, make_test_dataloader,
+                                       make_train_dataloader,
+                                       make_val_dataloader)
+from .trainer import Trainer, run_validation_generic, set_component_wise_lr
+
+# Worker-wide dataset cache (reference: client.py:45-47, 76-99)
+train_dataset = None
+trainset_unlab = None
+trainset_unlab_rand = None
+
+
+class Client:
+    """Per-client handle (reference: core/client.py:49-124)."""
+
+    def __init__(self, client_id, config, send_gradients):
+        self.client_id = client_id
+        self.config = config
+        self.send_gradients = send_gradients
+
+    def get_client_data(self, dataset=None):
+        client_data = self.get_data(self.client_id, dataset)
+        return self.client_id, client_data, self.config, self.send_gradients
+
+    @staticmethod
+    def get_train_dataset(data_path, client_train_config, task):
+        """Load + cache the full train dataset once per worker
+        (reference: client.py:76-99)."""
+        global train_dataset, trainset_unlab, trainset_unlab_rand
+        train_dataset = get_dataset(data_path, client_train_config, task, mode="train")
+        if task == "semisupervision":
+            trainset_unlab = get_dataset(data_path, client_train_config, task,
+                                         mode="train", user_idx=-2)
+            trainset_unlab_rand = get_dataset(data_path, client_train_config,
+                                              task, mode="train", user_idx=-3)
+        else:
+            trainset_unlab = None
+            trainset_unlab_rand = None
+        return len(train_dataset.user_list)
+
+    @staticmethod
+    def get_data(clients, dataset):
+        """Slice the cached dataset into one-or-more users' data blobs
+        (reference: client.py:101-124)."""
+        if dataset is None:
+            datasets = ([train_dataset, trainset_unlab, trainset_unlab_rand]
+                        if trainset_unlab is not None else [train_dataset])
+        else:
+            datasets = [dataset]
+        data_with_labels = hasattr(datasets[0], "user_data_label")
+        strcts = []
+        for ds in datasets:
+            s = {"users": [], "num_samples": [], "user_data": {}}
+            if data_with_labels:
+                s["user_data_label"] = {}
+            for client in clients:
+                user = ds.user_list[client]
+                s["users"].append(user)
+                s["num_samples"].append(ds.num_samples[client])
+                s["user_data"][user] = ds.user_data[user]
+                if data_with_labels:
+                    s["user_data_label"][user] = ds.user_data_label[user]
+            strcts.append(s)
+        return strcts
+
+
+class ClientExecutor:
+    """Per-rank persistent client-training workspace."""
+
+    def __init__(self, config, task, data_path, server_arena: ParameterArena,
+                 model_path: Optional[str] = None):
+        self.config = config
+        self.task = task
+        self.data_path = data_path
+        self.server_arena = server_arena
+        self.model_path = model_path or config.get("model_path")
+        self.model_config = config["model_config"]
+        self.client_config = config["client_config"]
+        self.server_config = config["server_config"]
+
+        # resident client model replica + arena
+        self.model = make_model(self.model_config)
+        self.arena = ParameterArena(self.model, bind_grads=True)
+        # reusable fused optimizer (reset per client)
+        opt_cfg = dict(self.client_config["optimizer_config"])
+        opt_cfg.setdefault("lr", self.server_config.get("initial_lr_client", 1.0))
+        self._fused_opt = make_arena_optimizer(opt_cfg, self.arena)
+        self._torch_opt_cfg = opt_cfg
+
+        strategy_cls = select_strategy(config["strategy"])
+        self.client_strategy = strategy_cls("client", config, self.model_path)
+        self.send_dicts = self.server_config.get("send_dicts", False)
+
+    # ------------------------------------------------------------------
+    def _make_optimizer(self, initial_lr):
+        if self._fused_opt is not None:
+            self._fused_opt.reset_state()
+            if initial_lr > 0:
+                self._fused_opt.param_groups[0]["lr"] = initial_lr
+            return self._fused_opt
+        cfg = dict(self._torch_opt_cfg)
+        if initial_lr > 0:
+            cfg["lr"] = initial_lr
+        return make_optimizer(cfg, self.model)
+
+    def process_round(self, client: Client, initial_lr: float, iteration: int,
+                      round_seed: int = 0) -> Dict:
+        """Run one client's local training (reference: client.py:226-511)."""
+        config = self.config
+        client_config = self.client_config
+        data_config = client_config["data_config"]["train"]
+        privacy_metrics_config = config.get("privacy_metrics_config", None)
+        begin = time.time()
+        client_stats = {}
+
+        client_id, data_strcts, _, send_gradients = client.get_client_data()
+        data_strct = data_strcts[0]
+        user = data_strct["users"][0]
+
+        train_dataloader = make_train_dataloader(
+            data_config, self.data_path, task=self.task, clientx=0,
+            data_strct=data_strct)
+
+        # one flat copy-in instead of the reference's per-tensor clone loop
+        # (client.py:294-301, K13)
+        self.arena.copy_data_(self.server_arena.data)
+        self.arena.zero_grad()
+
+        trainer_config = client_config.get("trainer_config", {})
+        if "updatable_names" in trainer_config:
+            set_component_wise_lr(self.model, client_config["optimizer_config"],
+                                  trainer_config["updatable_names"])
+
+        optimizer = self._make_optimizer(initial_lr)
+
+        ss_scheduler = None
+        if client_config.get("ss_config") is not None:
+            ss_scheduler = ScheduledSamplingScheduler(model=self.model,
+                                                      **client_config["ss_config"])
+
+        trainer = Trainer(
+            model=self.model,
+            optimizer=optimizer,
+            ss_scheduler=ss_scheduler,
+            train_dataloader=train_dataloader,
+            server_replay_config=client_config,
+            max_grad_norm=data_config.get("max_grad_norm", None),
+            anneal_config=client_config.get("annealing_config", None),
+            num_skips_threshold=client_config.get("num_skips_threshold", -1),
+            ignore_subtask=client_config["ignore_subtask"],
+            arena=self.arena,
+        )
+
+        desired_max_samples = data_config.get("desired_max_samples", None)
+        apply_privacy_metrics = bool(privacy_metrics_config
+                                     and privacy_metrics_config["apply_metrics"])
+
+        client_stats["setup"] = time.time() - begin
+        begin_training = time.time()
+
+        self.model.train()
+
+        algo_payload = None
+        strategy_algo = config["strategy"]
+        if strategy_algo == "FedLabels":
+            datasets = [get_dataset(self.data_path, config, self.task,
+                                    mode="train", test_only=False,
+                                    data_strct=data_strcts[i], user_idx=0)
+                        for i in range(3)]
+            algo_payload = {"strategy": "FedLabels", "data": datasets,
+                            "iter": iteration,
+                            "config": client_config.get("semisupervision")}
+        elif strategy_algo == "FedProx":
+            algo_payload = {"strategy": "FedProx",
+                            "mu": client_config.get("mu", 0.001)}
+
+        train_loss, num_samples, algo_computation = trainer.train_desired_samples(
+            desired_max_samples=desired_max_samples,
+            apply_privacy_metrics=apply_privacy_metrics,
+            algo_payload=algo_payload)
+
+        trainer.train_loss = train_loss
+        trainer.num_samples = num_samples
+        trainer.algo_computation = algo_computation
+
+        # pseudo-gradient g = w_server − w_trained (K1; reference client.py:380-383)
+        if not self.send_dicts:
+            ops.pseudo_grad(self.arena.grad, self.server_arena.data,
+                            self.arena.data, 1.0)
+
+        payload = self.client_strategy.generate_client_payload(trainer) \
+            if send_gradients else None
+
+        if self.server_config.get("type") == "personalization":
+            self._personalization_round(client, data_strct, data_config,
+                                        client_config, trainer, initial_lr,
+                                        user, desired_max_samples)
+
+        client_stats["training"] = time.time() - begin_training
+        client_stats["full cost"] = time.time() - begin
+
+        client_output = {
+            "cs": client_stats,
+            "tl": train_loss,
+            "mg": trainer.sufficient_stats["mag"],
+            "vg": trainer.sufficient_stats["var"],
+            "ng": trainer.sufficient_stats["mean"],
+            "rg": trainer.sufficient_stats["norm"],
+            "ns": num_samples,
+            "pl": payload,
+        }
+
+        if apply_privacy_metrics:
+            self._apply_privacy_metrics(client_output, trainer,
+                                        privacy_metrics_config)
+
+        client_output["ts"] = time.time()
+        return client_output
+
+    # ------------------------------------------------------------------
+    def _personalization_round(self, client, data_strct, data_config,
+                               client_config, trainer, initial_lr, user,
+                               desired_max_samples):
+        """Per-user local model + convex-interpolation alpha update
+        (reference: client.py:387-443)."""
+        alpha = client_config.get("convex_model_interp", 0.75)
+        local_model = make_model(self.config["model_config"])
+        train_dataloader = make_train_dataloader(
+            data_config, self.data_path, task=self.task, clientx=0,
+            data_strct=data_strct)
+        local_optimizer = make_optimizer(dict(client_config["optimizer_config"]),
+                                         local_model)
+        local_trainer = Trainer(
+            model=local_model, optimizer=local_optimizer, ss_scheduler=None,
+            train_dataloader=train_dataloader, server_replay_config=client_config,
+            max_grad_norm=data_config.get("max_grad_norm", None),
+            anneal_config=client_config.get("annealing_config", None),
+            ignore_subtask=client_config["ignore_subtask"])
+
+        local_model_name = os.path.join(self.model_path, f"{user}_model.tar")
+        local_alpha_name = os.path.join(self.model_path, f"{user}_alpha")
+        if os.path.exists(local_model_name):
+            local_trainer.load(local_model_name, update_lr_scheduler=False,
+                               update_ss_scheduler=False)
+        if os.path.exists(local_alpha_name):
+            alpha = torch.load(local_alpha_name, weights_only=False)
+
+        original_local_model = local_trainer.get_model()
+        local_model.train()
+        train_loss, num_samples, _ = local_trainer.train_desired_samples(
+            desired_max_samples=desired_max_samples, apply_privacy_metrics=False)
+        print_rank(f"user {user}: LOCAL training loss={train_loss}",
+                   loglevel=logging.DEBUG)
+        local_trainer.save(model_path=self.model_path, config=self.config,
+                           token=user)
+        for p, orig in zip(local_trainer.model.parameters(),
+                           original_local_model.parameters()):
+            p.grad = to_device(orig.data) - p.data
+        alpha = alpha_update(local_trainer.model, trainer.model, alpha, initial_lr)
+        torch.save(alpha, local_alpha_name)
+
+    # ------------------------------------------------------------------
+    def _apply_privacy_metrics(self, client_output, trainer,
+                               privacy_metrics_config):
+        """Token-extraction + leakage attacks; may zero the client's weight
+        (reference: client.py:466-508)."""
+        from ..extensions.privacy import metrics as privacy_metrics
+
+        privacy_stats = {"Dropped clients": 0}
+        batches = trainer.cached_batches
+        trainer.cached_batches = []
+        gradients = self.arena.grad  # already flat
+
+        if privacy_metrics_config.get("apply_indices_extraction", False):
+            allowed_word_rank = privacy_metrics_config.get("allowed_word_rank", 9000)
+            embed_dim = self.model_config["embed_dim"]
+            vocab_size = self.model_config["vocab_size"]
+            overlap, indices = privacy_metrics.extract_indices_from_embeddings(
+                gradients, batches, embed_dim, vocab_size)
+            max_overlap = privacy_metrics_config.get("max_allowed_overlap", None)
+            if max_overlap is not None and overlap > max_overlap:
+                print_rank(f"Removing client: extracted {overlap*100}% words, "
+                           f"max allowed {max_overlap*100}%")
+                client_output["wt"] = 0.0
+                privacy_stats["Dropped clients"] = 1
+            privacy_stats["Extracted indices percentage"] = overlap
+            privacy_stats[f"Words percentage above {allowed_word_rank} word rank"] = \
+                float((indices > allowed_word_rank).mean()) if len(indices) > 0 else 0
+
+        if privacy_metrics_config.get("apply_leakage_metric", False):
+            import numpy as np
+            orig_params = {n: self.server_arena.data[
+                self.server_arena.offsets[i]:self.server_arena.offsets[i]
+                + self.server_arena.numels[i]].view(self.server_arena.shapes[i])
+                for i, n in enumerate(self.server_arena.names)}
+            # practical_epsilon_leakage needs a full state dict
+            sd = {k: v.clone() for k, v in self.model.state_dict().items()}
+            for n, p in orig_params.items():
+                if n in sd:
+                    sd[n] = p.clone()
+            max_ratio = float(np.exp(privacy_metrics_config["max_leakage"]))
+            leakage = privacy_metrics.practical_epsilon_leakage(
+                sd, self.model, batches,
+                privacy_metrics_config.get("is_leakage_weighted", False),
+                max_ratio,
+                privacy_metrics_config.get("attacker_optimizer_config"))
+            max_leakage = privacy_metrics_config.get("max_allowed_leakage", None)
+            if max_leakage is not None and leakage > max_leakage:
+                print_rank(f"Removing client: leakage {leakage} > {max_leakage}")
+                client_output["wt"] = 0.0
+                privacy_stats["Dropped clients"] = 1
+            privacy_stats["Practical epsilon (Max leakage)"] = leakage
+
+        client_output["ps"] = privacy_stats
+
+def convex_inference(model_global, model_personal, alpha):
+    """Personalized logit interpolation accuracy
+    (reference: utils/utils.py:598-603)."""
+    import numpy as np
+    targets = torch.tensor(model_global["labels"])
+    probs = alpha * model_personal["probabilities"] + \
+        (1 - alpha) * model_global["probabilities"]
+    preds = torch.argmax(torch.tensor(np.asarray(probs)), dim=1)
+    return torch.mean((preds == targets).float()).item()

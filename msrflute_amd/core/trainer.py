@@ -1,0 +1,423 @@
+"""Training engine: client-side Trainer + server-side ModelUpdater.
+
+Reference: core/trainer.py (TrainerBase 30-79, ModelUpdater 82-197,
+Trainer 200-687, run_validation_generic 690-723, save_model 753-775).
+
+MI355X-first differences (behavior-preserving):
+
+* the model's params/grads are bound to a flat ``ParameterArena``; gradient
+  clipping, gradient sufficient statistics and (when the optimizer type
+  allows) the optimizer step are single fused kernels over the arena
+  instead of per-tensor loops;
+* gradient statistics (reference trainer.py:271-292 round-trips every
+  tensor through ``.cpu().numpy()`` per batch) accumulate in a device
+  tensor and hit the host exactly once per epoch;
+* the FedProx proximal term (reference trainer.py:463-466, which buggily
+  re-adds partial sums inside the parameter loop — SURVEY.md §7.5) is
+  applied exactly: ``grad += mu*(w - w_global)`` as two fused axpys and
+  ``loss += (mu/2)*||w - w_global||^2`` for reporting.
+
+Known deliberate deviation: the reference's ``var`` sufficient stat is
+identically zero by construction (``sq_sum/n - (sqrt(sq_sum/n))**2``,
+trainer.py:300-301); we compute the actual variance ``sq_sum/n - mean**2``.
+"""
+
+from __future__ import annotations
+
+import copy
+import logging
+import math
+import os
+from typing import Optional
+
+import torch
+import yaml
+
+from .. import ops
+from ..ops.arena import ParameterArena
+from ..ops.fused_optim import make_arena_optimizer
+from ..utils import (make_lr_scheduler, make_optimizer, print_rank, to_device)
+from .metrics import Metrics
+
+
+def get_lr(optimizer) -> float:
+    return optimizer.param_groups[0]["lr"]
+
+
+class TrainerBase:
+    """Common trainer interface (reference: core/trainer.py:30-79)."""
+
+    def __init__(self, model, train_dataloader, optimizer, max_grad_norm=None,
+                 ignore_subtask=True, model_type="LanguageModel",
+                 decoder_config=None, arena: Optional[ParameterArena] = None):
+        self.model = model
+        self.train_dataloader = train_dataloader
+        self.optimizer = optimizer
+        self.max_grad_norm = max_grad_norm
+        self.model_type = model_type
+        self.decoder_config = decoder_config
+        self.arena = arena
+        self.step = 0
+        self.ignore_subtask = ignore_subtask
+
+    def epoch_boundary(self):
+        return self.step % len(self.train_dataloader.create_loader()) == 0 and self.step != 0
+
+    # -- gradient plumbing -------------------------------------------------
+    def zero_grad(self):
+        if self.arena is not None:
+            self.arena.zero_grad()
+        else:
+            for p in self.model.parameters():
+                if p.grad is not None:
+                    p.grad.detach_()
+                    p.grad.zero_()
+
+    def clip_gradients(self):
+        if self.max_grad_norm is None:
+            return None
+        if self.arena is not None:
+            return ops.clip_by_norm(self.arena.grad, float(self.max_grad_norm))
+        return torch.nn.utils.clip_grad_norm_(self.model.parameters(),
+                                              self.max_grad_norm)
+
+    def train_desired_samples(self, desired_max_samples, apply_privacy_metrics=False):
+        pass
+
+    def save(self):
+        pass
+
+    def load(self):
+        pass
+
+
+class ModelUpdater(TrainerBase):
+    """Server-side trainer without data: applies the aggregated
+    pseudo-gradient (reference: core/trainer.py:82-197)."""
+
+    def __init__(self, model, optimizer, ss_scheduler, train_dataloader,
+                 val_dataloader, max_grad_norm, anneal_config,
+                 model_type="LanguageModel", decoder_config=None,
+                 arena: Optional[ParameterArena] = None, val_fn=None):
+        super().__init__(model=model, train_dataloader=train_dataloader,
+                         optimizer=optimizer, max_grad_norm=max_grad_norm,
+                         model_type=model_type, decoder_config=decoder_config,
+                         arena=arena)
+        self.val_dataloader = val_dataloader
+        self.annealing_type = anneal_config["type"] if anneal_config is not None else None
+        self.lr_scheduler = make_lr_scheduler(anneal_config, self.optimizer) \
+            if anneal_config is not None else None
+        self.ss_scheduler = ss_scheduler
+        # callback returning (val_loss, val_acc); wired to the distributed
+        # evaluation by the server (replaces reference's local val loader).
+        self.val_fn = val_fn
+
+    def update_model(self):
+        """Clip aggregated gradient then step the server optimizer
+        (reference: core/trainer.py:127-137)."""
+        self.clip_gradients()
+        self.optimizer.step()
+        self.optimizer.zero_grad() if self.arena is None else self.arena.zero_grad()
+
+    def run_lr_scheduler(self, force_run_val=False):
+        val_loss = val_acc = None
+        if (force_run_val or self.annealing_type == "val_loss") and self.val_fn is not None:
+            val_loss, val_acc = self.val_fn()
+        if self.lr_scheduler is not None:
+            if self.annealing_type == "val_loss":
+                self.lr_scheduler.step(val_loss)
+            else:
+                self.lr_scheduler.step()
+        return (val_loss, val_acc)
+
+    def run_ss_scheduler(self):
+        if self.ss_scheduler is not None:
+            self.ss_scheduler.step()
+
+    def save(self, model_path, token=None, config=None):
+        save_model(model_path=model_path, config=config, model=self.model,
+                   optimizer=self.optimizer, lr_scheduler=self.lr_scheduler,
+                   ss_scheduler=self.ss_scheduler, token=token)
+
+    def load(self, save_path, update_lr_scheduler, update_ss_scheduler):
+        _load_checkpoint(self, save_path, update_lr_scheduler, update_ss_scheduler)
+
+
+class Trainer(TrainerBase):
+    """Client-side local-SGD engine (reference: core/trainer.py:200-687)."""
+
+    def __init__(self, model, ss_scheduler, train_dataloader,
+                 server_replay_config=None, optimizer=None, max_grad_norm=None,
+                 anneal_config=None, num_skips_threshold=-1,
+                 ignore_subtask=True, arena: Optional[ParameterArena] = None):
+        super().__init__(model=model, train_dataloader=train_dataloader,
+                         optimizer=optimizer, max_grad_norm=max_grad_norm,
+                         ignore_subtask=ignore_subtask, arena=arena)
+        self.server_replay_config = server_replay_config
+        self.anneal_config = anneal_config
+        self.lr_scheduler = None
+        if self.optimizer is None and server_replay_config is not None \
+                and "optimizer_config" in server_replay_config:
+            self.optimizer = make_optimizer(server_replay_config["optimizer_config"], model)
+        if self.optimizer is not None and self.anneal_config is not None:
+            self.lr_scheduler = make_lr_scheduler(self.anneal_config, self.optimizer)
+        self.cached_batches = []
+        self.ss_scheduler = ss_scheduler
+        self.reset_gradient_power()
+
+    # -- gradient sufficient statistics (K5) ------------------------------
+    def reset_gradient_power(self):
+        dev = self.arena.device if self.arena is not None else "cpu"
+        self._stats_acc = torch.zeros(2, dtype=torch.float32, device=dev)
+        self.counter = 0
+        self.sum_grad = 0.0
+        self.sum_grad2 = 0.0
+        self.sufficient_stats = {}
+
+    def accumulate_gradient_power(self):
+        """Accumulate Σg, Σg² for the current gradient — one fused pass on
+        the arena, no host transfer (reference: trainer.py:271-292)."""
+        if self.arena is not None:
+            self._stats_acc += ops.sum_sumsq(self.arena.grad)
+            self.counter += self.arena.total
+        else:
+            for p in self.model.parameters():
+                if p.grad is None:
+                    continue
+                g = p.grad.detach().reshape(-1)
+                self._stats_acc[0] += g.sum()
+                self._stats_acc[1] += g.dot(g)
+                self.counter += g.numel()
+
+    def estimate_sufficient_stats(self):
+        """Finalize stats dict (reference: trainer.py:294-312; one host
+        sync here instead of one per tensor per batch)."""
+        self.accumulate_gradient_power()
+        acc = self._stats_acc.tolist()
+        self.sum_grad, self.sum_grad2 = acc[0], acc[1]
+        n = max(self.counter, 1)
+        mean_grad = self.sum_grad / n
+        mag_grad = math.sqrt(max(self.sum_grad2 / n, 0.0))
+        var_grad = max(self.sum_grad2 / n - mean_grad ** 2, 0.0)
+        norm_grad = math.sqrt(max(self.sum_grad2, 0.0))
+        self.sufficient_stats = {
+            "n": n, "sum": self.sum_grad, "sq_sum": self.sum_grad2,
+            "var": var_grad, "mean": mean_grad, "mag": mag_grad,
+            "norm": norm_grad,
+        }
+        return self.sufficient_stats
+
+    # -- training loops ----------------------------------------------------
+    def train_desired_samples(self, desired_max_samples=None,
+                              apply_privacy_metrics=False, algo_payload=None):
+        """One local-training pass; returns (train_loss, num_samples,
+        algo_computation) (reference: trainer.py:314-339)."""
+        algo_computation = None
+        if algo_payload is None:
+            num_samples, train_loss = self.run_train_epoch(
+                desired_max_samples, apply_privacy_metrics)
+        elif algo_payload["strategy"] == "FedProx":
+            num_samples, train_loss = self.run_train_epoch_fedprox(
+                desired_max_samples, apply_privacy_metrics, algo_payload)
+        elif algo_payload["strategy"] == "FedLabels":
+            num_samples, train_loss, algo_computation = self.run_train_epoch_sup(
+                desired_max_samples, apply_privacy_metrics, algo_payload)
+        else:
+            raise ValueError(f"unknown algo payload {algo_payload['strategy']}")
+        return train_loss, num_samples, algo_computation
+
+    def _batch_loss(self, batch, apply_privacy_metrics):
+        if self.ignore_subtask and hasattr(self.model, "single_task_loss"):
+            return self.model.single_task_loss(batch)
+        if apply_privacy_metrics:
+            if "x" in batch:
+                self.cached_batches.append(to_device(batch["x"]))
+            elif "input_ids" in batch:
+                self.cached_batches.append(to_device(batch["input_ids"]))
+        return self.model.loss(batch)
+
+    @staticmethod
+    def _batch_samples(batch):
+        if "attention_mask" in batch:
+            return int(torch.sum(batch["attention_mask"].detach().cpu() == 1).item())
+        if "total_frames" in batch:
+            return int(batch["total_frames"])
+        return len(batch["x"])
+
+    def _train_step(self, loss):
+        loss.backward()
+        self.clip_gradients()
+        self.accumulate_gradient_power()
+        if self.optimizer is not None:
+            self.optimizer.step()
+
+    def run_train_epoch(self, desired_max_samples=None,
+                        apply_privacy_metrics=False, prox=None):
+        """Reference: trainer.py:341-414 (and 416-501 when ``prox`` is set:
+        prox = (mu, w_global_flat))."""
+        sum_train_loss = 0.0
+        num_samples = 0
+        self.reset_gradient_power()
+        self.zero_grad()
+
+        train_loader = self.train_dataloader.create_loader()
+        for batch in train_loader:
+            if desired_max_samples is not None and num_samples >= desired_max_samples:
+                break
+            # NOTE: not optimizer.zero_grad() — torch's set_to_none default
+            # would unbind the arena grad views.
+            self.zero_grad()
+            loss = self._batch_loss(batch, apply_privacy_metrics)
+            loss_val = loss.item()
+            if prox is not None:
+                mu, w_global = prox
+                loss.backward()
+                if self.arena is not None:
+                    # exact prox gradient: g += mu*(w - w_global)
+                    ops.axpy(self.arena.grad, self.arena.data, mu)
+                    ops.axpy(self.arena.grad, w_global, -mu)
+                    loss_val += 0.5 * mu * float(
+                        (self.arena.data - w_global).square().sum())
+                else:
+                    off = 0
+                    reg = 0.0
+                    for p in self.model.parameters():
+                        wg = w_global[off:off + p.numel()].view(p.shape)
+                        p.grad.add_(p.data - wg, alpha=mu)
+                        reg += 0.5 * mu * float((p.data - wg).square().sum())
+                        off += p.numel()
+                    loss_val += reg
+                self.clip_gradients()
+                self.accumulate_gradient_power()
+                if self.optimizer is not None:
+                    self.optimizer.step()
+            else:
+                self._train_step(loss)
+            sum_train_loss += loss_val
+            num_samples += self._batch_samples(batch)
+            self.step += 1
+
+        if self.lr_scheduler is not None:
+            self.lr_scheduler.step()
+        self.estimate_sufficient_stats()
+        return num_samples, sum_train_loss
+
+    def run_train_epoch_fedprox(self, desired_max_samples=None,
+                                apply_privacy_metrics=False, algo_payload=None):
+        """FedProx local training (reference: trainer.py:416-501; the
+        proximal term is applied exactly once per batch — see module
+        docstring)."""
+        mu = algo_payload["mu"]
+        if self.arena is not None:
+            w_global = self.arena.clone_data()
+        else:
+            w_global = torch.cat([p.detach().reshape(-1).clone()
+                                  for p in self.model.parameters()])
+        return self.run_train_epoch(desired_max_samples, apply_privacy_metrics,
+                                    prox=(mu, w_global))
+
+    def run_train_epoch_sup(self, desired_max_samples=None,
+                            apply_privacy_metrics=False, algo_payload=None):
+        """FedLabels semi-supervised local training
+        (reference: trainer.py:503-619)."""
+        from ..extensions.fedlabels_train import run_train_epoch_sup as _sup
+        return _sup(self, desired_max_samples, apply_privacy_metrics, algo_payload)
+
+    # -- misc --------------------------------------------------------------
+    def get_model(self):
+        return copy.deepcopy(self.model)
+
+    def prepare_iteration(self, model=None):
+        """Reference: trainer.py:624-638."""
+        if model is not None:
+            self.model.load_state_dict(model.state_dict())
+            self.lr_scheduler = None
+            if self.optimizer is None and self.server_replay_config is not None \
+                    and "optimizer_config" in self.server_replay_config:
+                self.optimizer = make_optimizer(
+                    self.server_replay_config["optimizer_config"], self.model)
+            if self.optimizer is not None and self.anneal_config is not None:
+                self.lr_scheduler = make_lr_scheduler(self.anneal_config, self.optimizer)
+
+    def reset_optimizer(self, optimizer_state_dict, annealing_config=None):
+        assert self.optimizer is not None, "This trainer does not have an optimizer"
+        self.optimizer.load_state_dict(optimizer_state_dict)
+        self.lr_scheduler = None
+        if annealing_config is not None:
+            self.lr_scheduler = make_lr_scheduler(annealing_config, self.optimizer)
+
+    def save(self, model_path, token=None, config=None):
+        save_model(model_path=model_path, config=config, model=self.model,
+                   optimizer=self.optimizer, lr_scheduler=self.lr_scheduler,
+                   ss_scheduler=self.ss_scheduler, token=token)
+
+    def load(self, save_path, update_lr_scheduler, update_ss_scheduler):
+        _load_checkpoint(self, save_path, update_lr_scheduler, update_ss_scheduler)
+
+
+def run_validation_generic(model, val_dataloader):
+    """Run the generic metrics loop (reference: trainer.py:690-723)."""
+    model.set_eval()
+    val_loader = val_dataloader.create_loader()
+    return Metrics().compute_metrics(dataloader=val_loader, model=model)
+
+
+def set_component_wise_lr(model, optimizer_config, updatable_names):
+    """Freeze layers by zero LR (reference: trainer.py:725-751)."""
+    import re
+
+    def matched(name):
+        return any(re.match(u, name) is not None for u in updatable_names)
+
+    parameters = []
+    for name, params in model.named_parameters():
+        if matched(name):
+            parameters.append({"params": params, "lr": optimizer_config["lr"]})
+        else:
+            parameters.append({"params": params, "lr": 0.0})
+    return parameters
+
+
+def save_model(model_path, config, model, optimizer, lr_scheduler,
+               ss_scheduler, token=None):
+    """Checkpoint in the reference's exact .tar dict layout
+    (reference: trainer.py:753-775; format parity required by BASELINE)."""
+    save_state = {
+        "model_state_dict": model.state_dict(),
+        "optimizer_state_dict": optimizer.state_dict() if optimizer is not None else None,
+        "lr_scheduler_state_dict": lr_scheduler.state_dict() if lr_scheduler is not None else None,
+    }
+    if ss_scheduler is not None:
+        save_state["ss_scheduler_state_dict"] = ss_scheduler.state_dict()
+    name = f"{token}_model.tar" if token else "model.tar"
+    save_path = os.path.join(model_path, name)
+    for attempt in range(3):  # retry wrapper (reference: utils/utils.py:348-359)
+        try:
+            torch.save(save_state, save_path)
+            break
+        except Exception as e:
+            print_rank(f"save attempt {attempt} failed: {e}", loglevel=logging.WARNING)
+    if config is not None:
+        cfg = config.to_dict() if hasattr(config, "to_dict") else dict(config)
+        with open(os.path.join(model_path, "config.yaml"), "w") as f:
+            yaml.safe_dump(cfg, f)
+
+
+def _load_checkpoint(trainer, save_path, update_lr_scheduler, update_ss_scheduler):
+    if not os.path.isfile(save_path):
+        return
+    print_rank(f"Loading checkpoint: {save_path}")
+    checkpoint = torch.load(save_path, map_location="cpu", weights_only=False)
+    trainer.model.load_state_dict(checkpoint["model_state_dict"])
+    if trainer.arena is not None:
+        # state_dict load wrote through the arena views; nothing to rebind,
+        # but make sure grads are still views after any torn state.
+        trainer.arena.zero_grad()
+    if trainer.optimizer is not None and checkpoint.get("optimizer_state_dict"):
+        trainer.optimizer.load_state_dict(checkpoint["optimizer_state_dict"])
+    anl = checkpoint.get("lr_scheduler_state_dict")
+    if anl and trainer.lr_scheduler is not None and update_lr_scheduler:
+        trainer.lr_scheduler.load_state_dict(anl)
+    sss = checkpoint.get("ss_scheduler_state_dict")
+    if sss and trainer.ss_scheduler is not None and update_ss_scheduler:
+        trainer.ss_scheduler.load_state_dict(sss)

@@ -1,0 +1,131 @@
+"""Shared dataset/dataloader machinery for array-style tasks.
+
+Implements the universal FLUTE data-blob contract
+(``{users, num_samples, user_data, user_data_label}`` — reference
+doc/sphinx/scenarios.rst, testing/create_data.py:46-51) once, so each task
+plugin only declares its sample shape.  Accepts a path to a JSON/NPZ blob
+or an in-memory dict (the per-client slice handed out by
+``Client.get_data``).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+from typing import Optional
+
+import numpy as np
+import torch
+
+from ..core.dataloader import BaseDataLoader
+from ..core.dataset import BaseDataset
+
+
+def load_blob(data):
+    """Load a data blob from dict / .json / .npz / torch .pt file."""
+    if isinstance(data, dict):
+        return data
+    if data is None:
+        raise ValueError("no data provided")
+    path = str(data)
+    if path.endswith(".npz"):
+        z = np.load(path, allow_pickle=True)
+        return {k: z[k].item() if z[k].dtype == object and z[k].shape == () else z[k]
+                for k in z.files}
+    if path.endswith(".pt"):
+        return torch.load(path, weights_only=False)
+    with open(path, "r") as f:
+        return json.load(f)
+
+
+class ArrayDataset(BaseDataset):
+    """Array-feature dataset over the FLUTE blob convention.
+
+    ``user_idx == -1`` (or ``test_only``): concatenates every user's rows
+    (server-side eval usage); ``user_idx >= 0``: that single user's shard.
+    """
+
+    def __init__(self, data, test_only=False, user_idx=-1, args=None,
+                 x_dtype=torch.float32, y_dtype=torch.int64,
+                 x_shape: Optional[tuple] = None, **kwargs):
+        self.test_only = test_only
+        self.x_dtype = x_dtype
+        self.y_dtype = y_dtype
+        self.x_shape = x_shape
+        self.args = args or {}
+        self.load_data(data=data, user_idx=user_idx)
+
+    def load_data(self, data=None, user_idx=-1):
+        blob = load_blob(data)
+        self.user_list = list(blob["users"])
+        self.num_samples = list(blob["num_samples"])
+        self.user_data = blob["user_data"]
+        self.user_data_label = blob.get("user_data_label", None)
+
+        if self.test_only or user_idx == -1:
+            users = self.user_list
+        else:
+            users = [self.user_list[user_idx]]
+
+        xs, ys = [], []
+        for u in users:
+            ud = self.user_data[u]
+            x = ud["x"] if isinstance(ud, dict) else ud
+            xs.append(torch.as_tensor(np.asarray(x), dtype=self.x_dtype))
+            if self.user_data_label is not None:
+                y = self.user_data_label[u]
+                ys.append(torch.as_tensor(np.asarray(y), dtype=self.y_dtype))
+        self.x = torch.cat(xs) if xs else torch.empty(0)
+        if self.x_shape is not None and self.x.numel():
+            self.x = self.x.reshape(-1, *self.x_shape)
+        self.y = torch.cat(ys) if ys else None
+
+    def __len__(self):
+        return len(self.x)
+
+    def __getitem__(self, idx):
+        if self.y is not None:
+            return self.x[idx], self.y[idx]
+        return self.x[idx]
+
+
+class ArrayDataLoader(BaseDataLoader):
+    """Batching dataloader producing ``{'x': tensor, 'y': tensor}`` dicts.
+
+    Slices the dataset's packed tensors directly (no per-sample Python
+    collate) — batches are views, so the hot path does zero copies until
+    ``to_device`` in the model.
+    """
+
+    def __init__(self, data, user_idx=0, mode="train", args=None,
+                 dataset_cls=ArrayDataset, **kwargs):
+        args = args or {}
+        self.mode = mode
+        self.args = args
+        batch_size = args.get("batch_size", 40)
+        self.dataset = dataset_cls(
+            data, test_only=(mode != "train"),
+            user_idx=user_idx if mode == "train" else -1, args=args)
+        self.batch_size = max(1, int(batch_size))
+        self.shuffle = (mode == "train")
+        # intentionally NOT calling PyTorchDataLoader.__init__: this loader
+        # iterates tensor slices itself (faster for small client shards)
+
+    def create_loader(self):
+        return self
+
+    def __len__(self):
+        n = len(self.dataset)
+        return max(1, (n + self.batch_size - 1) // self.batch_size) if n else 0
+
+    def __iter__(self):
+        n = len(self.dataset)
+        if n == 0:
+            return
+        order = torch.randperm(n) if self.shuffle else torch.arange(n)
+        for s in range(0, n, self.batch_size):
+            idx = order[s:s + self.batch_size]
+            batch = {"x": self.dataset.x[idx]}
+            if self.dataset.y is not None:
+                batch["y"] = self.dataset.y[idx]
+            yield batch

@@ -32,6 +32,7 @@ class ParameterArena:
                  dtype: torch.dtype = torch.float32):
         self.model = model
         params = [p for p in model.parameters()]
+        self.names: List[str] = [n for n, _ in model.named_parameters()]
         self.shapes: List[torch.Size] = [p.shape for p in params]
         self.numels: List[int] = [p.numel() for p in params]
         self.total: int = sum(self.numels)
@@ -73,6 +74,15 @@ class ParameterArena:
             self._bind_grads()  # cheap; restores any views torn off by user code
 
     # -- views ------------------------------------------------------------
+    def segment_of(self, name: str) -> Tuple[int, int]:
+        """(offset, numel) of the named parameter's arena segment."""
+        i = self.names.index(name)
+        return self.offsets[i], self.numels[i]
+
+    def grad_segment(self, i: int) -> torch.Tensor:
+        off, n = self.offsets[i], self.numels[i]
+        return self.grad[off:off + n]
+
     def param_view(self, i: int) -> torch.Tensor:
         off, n = self.offsets[i], self.numels[i]
         return self.data[off:off + n].view(self.shapes[i])

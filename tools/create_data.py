@@ -1,0 +1,119 @@
+#!/usr/bin/env python3
+"""Synthetic federated-data generator.
+
+Equivalent role to the reference's dummy-data generator
+(testing/create_data.py:81-173): emits data blobs in the universal FLUTE
+layout ``{users, num_samples, user_data, user_data_label}``.  All data is
+synthetic (random): there is no network access for real datasets; bench
+and tests consume these shapes only.
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+
+import numpy as np
+import torch
+
+
+def make_classification_blob(n_users, samples_per_user, feature_shape,
+                             n_classes, seed=0, dtype=np.float32,
+                             flat=False):
+    rng = np.random.default_rng(seed)
+    users, num_samples, user_data, user_data_label = [], [], {}, {}
+    for u in range(n_users):
+        name = f"user{u:05d}"
+        n = samples_per_user if np.isscalar(samples_per_user) else int(
+            rng.integers(samples_per_user[0], samples_per_user[1] + 1))
+        shape = (n, int(np.prod(feature_shape))) if flat else (n, *feature_shape)
+        x = rng.standard_normal(shape).astype(dtype)
+        y = rng.integers(0, n_classes, size=n)
+        users.append(name)
+        num_samples.append(n)
+        user_data[name] = {"x": x}
+        user_data_label[name] = y
+    return {"users": users, "num_samples": num_samples,
+            "user_data": user_data, "user_data_label": user_data_label}
+
+
+def make_mnist_blob(n_users=1000, samples_per_user=60, seed=0):
+    return make_classification_blob(n_users, samples_per_user, (784,), 10,
+                                    seed=seed, flat=True)
+
+
+def make_femnist_blob(n_users=3400, samples_per_user=100, seed=0):
+    return make_classification_blob(n_users, samples_per_user, (28, 28), 62,
+                                    seed=seed)
+
+
+def make_char_lm_blob(n_users, samples_per_user, seq_len=80, vocab=90, seed=0):
+    """Shakespeare-style char-LM shards: x int sequences, y next-char
+    sequences."""
+    rng = np.random.default_rng(seed)
+    users, num_samples, user_data, user_data_label = [], [], {}, {}
+    for u in range(n_users):
+        name = f"user{u:05d}"
+        n = samples_per_user
+        x = rng.integers(1, vocab, size=(n, seq_len))
+        y = np.roll(x, -1, axis=1)
+        users.append(name)
+        num_samples.append(n)
+        user_data[name] = {"x": x}
+        user_data_label[name] = y
+    return {"users": users, "num_samples": num_samples,
+            "user_data": user_data, "user_data_label": user_data_label}
+
+
+def save_blob(blob, path):
+    os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
+    if path.endswith(".pt"):
+        torch.save(blob, path)
+    else:
+        import json
+
+        def clean(o):
+            if isinstance(o, np.ndarray):
+                return o.tolist()
+            if isinstance(o, dict):
+                return {k: clean(v) for k, v in o.items()}
+            if isinstance(o, list):
+                return [clean(v) for v in o]
+            return o
+        with open(path, "w") as f:
+            json.dump(clean(blob), f)
+
+
+TASKS = {
+    "cv_lr_mnist": (make_mnist_blob, dict(n_users=1000, samples_per_user=60)),
+    "cv_cnn_femnist": (make_femnist_blob, dict(n_users=3400, samples_per_user=100)),
+}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--task", required=True, choices=sorted(TASKS))
+    ap.add_argument("--out", required=True, help="output directory")
+    ap.add_argument("--users", type=int, default=None)
+    ap.add_argument("--samples", type=int, default=None)
+    ap.add_argument("--seed", type=int, default=0)
+    args = ap.parse_args()
+
+    fn, kw = TASKS[args.task]
+    kw = dict(kw)
+    if args.users:
+        kw["n_users"] = args.users
+    if args.samples:
+        kw["samples_per_user"] = args.samples
+    train = fn(seed=args.seed, **kw)
+    kw_eval = dict(kw, n_users=max(kw["n_users"] // 20, 2))
+    val = fn(seed=args.seed + 1, **kw_eval)
+    test = fn(seed=args.seed + 2, **kw_eval)
+    save_blob(train, os.path.join(args.out, args.task, "train_data.pt"))
+    save_blob(val, os.path.join(args.out, args.task, "val_data.pt"))
+    save_blob(test, os.path.join(args.out, args.task, "test_data.pt"))
+    print(f"wrote {args.task} blobs to {os.path.join(args.out, args.task)}")
+
+
+if __name__ == "__main__":
+    main()
