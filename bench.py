@@ -1,0 +1,200 @@
+#!/usr/bin/env python3
+"""Benchmark driver: CNN-FEMNIST FedAvg FL rounds/sec (BASELINE.json metric).
+
+One step = one federated-learning round of the reference's benchmark task 2
+(Fed-EMNIST, CNN 2conv+2FC, FedAvg, 3400 clients, 10 clients
+per round, batch 20, client SGD lr=0.1, 1 local epoch): deterministic
+client sampling, local training of every sampled client, weighted
+pseudo-gradient aggregation (all-reduce at N>1) and the server optimizer
+step.  Synthetic federated shards (random normal images, random labels)
+and random-init weights — no network access for the real dataset; shapes
+and sample counts match the benchmark task.
+
+Run directly (1 GPU) or under torch.distributed.run with --nproc-per-node N
+(one rank per GPU over RCCL).  Rank 0 prints one JSON line.
+
+Reference wall-clock for this workload: 1500 rounds in 00:08:22 ≈ 2.99
+rounds/s on an unspecified GPU (BASELINE.md / reference README.md:39).
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+REF_ROUNDS_PER_SEC = 1500 / 502.0  # reference CNN_FEMNIST: 1500 rounds / 8:22
+
+
+def build_config(args):
+    from msrflute_amd.config import FLUTEConfig
+    cfg = {
+        "model_config": {
+            "model_type": "CNN",
+            "model_folder": "experiments/cv_cnn_femnist/model.py",
+            "num_classes": 62,
+        },
+        "dp_config": {"enable_local_dp": False},
+        "privacy_metrics_config": {"apply_metrics": False},
+        "strategy": "FedAvg",
+        "server_config": {
+            "wantRL": False,
+            "resume_from_checkpoint": False,
+            "do_profiling": False,
+            "optimizer_config": {"type": "sgd", "lr": 1.0},
+            "annealing_config": {"type": "step_lr", "step_interval": "epoch",
+                                 "gamma": 1.0, "step_size": 10000},
+            "val_freq": 10 ** 9, "rec_freq": 10 ** 9,
+            "initial_val": False, "initial_rec": False,
+            "max_iteration": args.warmup + args.steps,
+            "num_clients_per_iteration": args.clients_per_round,
+            "data_config": {
+                "val": {"batch_size": 2048, "val_data": None},
+                "test": {"batch_size": 2048, "test_data": None},
+            },
+            "type": "model_optimization",
+            "aggregate_median": "mean",
+            "weight_train_loss": "train_loss",
+            "softmax_beta": 1.0,
+            "initial_lr_client": 0.1,
+            "lr_decay_factor": 1.0,
+            "best_model_criterion": "loss",
+            "fall_back_to_best_model": False,
+            "seed": 1234,
+        },
+        "client_config": {
+            "do_profiling": False,
+            "ignore_subtask": False,
+            "data_config": {
+                "train": {"batch_size": 20, "list_of_train_data": None,
+                          "desired_max_samples": 100000,
+                          "max_grad_norm": 10.0},
+            },
+            "type": "optimization",
+            "optimizer_config": {"type": "sgd", "lr": 0.1},
+        },
+    }
+    return FLUTEConfig.from_dict(cfg)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--clients", type=int, default=3400,
+                    help="total client pool size")
+    ap.add_argument("--clients-per-round", type=int, default=10)
+    ap.add_argument("--samples-per-client", type=int, default=100)
+    args = ap.parse_args()
+
+    from msrflute_amd.comm import runtime as rt_mod
+    from msrflute_amd.core import client as client_mod
+    from msrflute_amd.core.client import Client
+    from msrflute_amd.core.server import OptimizationServer
+    from msrflute_amd.models import make_model
+    from msrflute_amd.ops.arena import ParameterArena
+    from msrflute_amd.ops.fused_optim import make_arena_optimizer
+    from tools.create_data import make_femnist_blob
+
+    world_size = int(os.environ.get("WORLD_SIZE", 1))
+    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    rt = rt_mod.init_runtime(backend=backend, seed=1234)
+
+    config = build_config(args)
+    config["model_path"] = os.path.join("gpurun_out", "bench_models")
+    os.makedirs(config["model_path"], exist_ok=True)
+
+    # synthetic federated shards, cached the way the engine expects
+    blob = make_femnist_blob(n_users=args.clients,
+                             samples_per_user=args.samples_per_client, seed=7)
+    from msrflute_amd.models.generic_data import ArrayDataset
+    ds = ArrayDataset(blob, test_only=False, user_idx=-1, args={},
+                      x_shape=(28, 28))
+    # keep per-user arrays for slicing
+    ds.user_data = blob["user_data"]
+    ds.user_data_label = blob["user_data_label"]
+    client_mod.train_dataset = ds
+
+    torch.manual_seed(1234 + 12345)
+    model = make_model(config["model_config"])
+    arena = ParameterArena(model, bind_grads=True)
+    rt.broadcast_(arena.data, src=0)
+    optimizer = make_arena_optimizer(
+        dict(config["server_config"]["optimizer_config"]), arena)
+
+    server = OptimizationServer(
+        num_clients=args.clients, model=model, optimizer=optimizer,
+        ss_scheduler=None, data_path=None, model_path=config["model_path"],
+        server_train_dataloader=None, config=config, idx_val_clients=[],
+        idx_test_clients=[], runtime=rt, arena=arena, task="cv_cnn_femnist")
+
+    server.run_stats = {k: [] for k in [
+        "secsPerClientRound", "secsPerClient", "secsPerClientTraining",
+        "secsPerClientSetup", "secsPerClientFull",
+        "secsPerRoundHousekeeping", "secsPerRoundTotal", "communicationCosts"]}
+    server.worker_trainer.model.train()
+
+    def sync():
+        rt.barrier()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        server.run_one_round(i, housekeeping=False)
+
+    if torch.cuda.is_available():
+        torch.cuda.reset_peak_memory_stats()
+    sync()
+    t0 = time.time()
+    for i in range(args.warmup, args.warmup + args.steps):
+        server.run_one_round(i, housekeeping=False)
+    sync()
+    elapsed = time.time() - t0
+
+    # max over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device="cuda" if backend == "nccl" else "cpu")
+    if rt.size > 1:
+        import torch.distributed as dist
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    peak_mb = (torch.cuda.max_memory_allocated() / 2 ** 20
+               if torch.cuda.is_available() else 0.0)
+    value = args.steps / elapsed
+    if rt.rank == 0:
+        print(json.dumps({
+            "metric": "fl_rounds_per_sec",
+            "value": value,
+            "unit": "rounds/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": value / REF_ROUNDS_PER_SEC,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "cnn_femnist (2conv+2fc, 62 classes)",
+                "global_batch": args.clients_per_round * 20,
+                "seq_len": 784,
+                "parallelism": f"client-parallel fedavg, {world_size} ranks",
+                "clients": args.clients,
+                "clients_per_round": args.clients_per_round,
+                "samples_per_client": args.samples_per_client,
+                "local_epochs": 1,
+                "client_batch_size": 20,
+                "client_lr": 0.1,
+                "peak_gpu_mem_mb": round(peak_mb, 1),
+                "eval_and_checkpoint": "outside timed region",
+            },
+        }))
+    rt.shutdown()
+
+
+if __name__ == "__main__":
+    main()

@@ -148,6 +148,13 @@ class ClientExecutor:
         data_strct = data_strcts[0]
         user = data_strct["users"][0]
 
+        # Per-client seed discipline: a client's
+        # local-SGD trajectory (shuffle order, dropout) is a function of
+        # (round, client) only — independent of which rank runs it — so the
+        # aggregated round result is world-size-invariant up to fp
+        # reduction order.
+        torch.manual_seed(round_seed & 0x7FFFFFFFFFFF)
+
         train_dataloader = make_train_dataloader(
             data_config, self.data_path, task=self.task, clientx=0,
             data_strct=data_strct)

@@ -169,15 +169,19 @@ class OptimizationServer:
     # ------------------------------------------------------------------
     def train(self):
         """Main loop (reference: server.py:215-528)."""
+        self.train_setup()
+        for i in range(self.cur_iter_no, self.max_iteration):
+            self.run_one_round(i)
+
+    def train_setup(self):
+        """Initial eval + initial checkpoint saves (reference: server.py:218-258)."""
         rt = self.runtime
-        is_chief = rt.rank == 0
         self.run_stats = {
             "secsPerClientRound": [], "secsPerClient": [],
             "secsPerClientTraining": [], "secsPerClientSetup": [],
             "secsPerClientFull": [], "secsPerRoundHousekeeping": [],
             "secsPerRoundTotal": [], "communicationCosts": [],
         }
-
         eval_list = []
         if self.cur_iter_no == 0:
             if self.config["server_config"]["initial_rec"]:
@@ -187,16 +191,21 @@ class OptimizationServer:
             print_rank(f"Running {eval_list} at itr={self.cur_iter_no}")
             self.metrics = self.evaluation.run(eval_list, self.metrics,
                                                metric_logger=log_metric)
-            eval_list = []
-
-        if is_chief:
+        if rt.rank == 0:
             for token in ["best_val_loss", "best_val_acc", "best_test_acc", "latest"]:
                 self.worker_trainer.save(model_path=self.model_path,
                                          token=token,
                                          config=self.config["server_config"])
-
         self.worker_trainer.model.train()
-        for i in range(self.cur_iter_no, self.max_iteration):
+
+    def run_one_round(self, i, housekeeping=True):
+        """One FL round: sample → local client training → all-reduce
+        aggregation → replicated server update (+ eval/checkpoint cadence
+        when ``housekeeping``).  Reference round body: server.py:259-525."""
+        rt = self.runtime
+        is_chief = rt.rank == 0
+        eval_list = []
+        if True:
             begin = time.time()
             metrics_payload = {}
 
@@ -361,9 +370,9 @@ class OptimizationServer:
             self.worker_trainer.run_ss_scheduler()
 
             # ---- evaluation cadence --------------------------------------
-            if ((i + 1) % self.val_freq) == 0:
+            if housekeeping and ((i + 1) % self.val_freq) == 0:
                 eval_list.append("val")
-            if ((i + 1) % self.req_freq) == 0:
+            if housekeeping and ((i + 1) % self.req_freq) == 0:
                 eval_list.append("test")
 
             ran_val = "val" in eval_list
@@ -388,13 +397,13 @@ class OptimizationServer:
                     print_rank(f"LOG: Client weight of learning rate {self.lr_weight}..")
 
             # ---- checkpoint / backup / fallback --------------------------
-            if is_chief:
+            if is_chief and housekeeping:
                 self.backup_models(i)
-            if self.fall_back_to_best_model:
+            if self.fall_back_to_best_model and housekeeping:
                 rt.barrier()
-            self.fall_back_to_prev_best_status()
+                self.fall_back_to_prev_best_status()
 
-            if len(self.metrics) > 1 and is_chief:
+            if housekeeping and len(self.metrics) > 1 and is_chief:
                 update_json_log(self.log_path, {
                     "i": i + 1,
                     "best_val_loss": float(self.metrics.get("best_val_loss", float("inf"))),
