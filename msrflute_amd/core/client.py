@@ -158,6 +158,8 @@ class ClientExecutor:
         train_dataloader = make_train_dataloader(
             data_config, self.data_path, task=self.task, clientx=0,
             data_strct=data_strct)
+        if hasattr(train_dataloader, "to_device"):
+            train_dataloader.to_device()
 
         # one flat copy-in instead of the reference's per-tensor clone loop
         # (client.py:294-301, K13)

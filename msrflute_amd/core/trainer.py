@@ -255,10 +255,13 @@ class Trainer(TrainerBase):
                         apply_privacy_metrics=False, prox=None):
         """Reference: trainer.py:341-414 (and 416-501 when ``prox`` is set:
         prox = (mu, w_global_flat))."""
-        sum_train_loss = 0.0
         num_samples = 0
         self.reset_gradient_power()
         self.zero_grad()
+        # loss accumulates in a device scalar: ONE host sync per epoch
+        # instead of the reference's per-batch loss.item()
+        dev = self.arena.device if self.arena is not None else "cpu"
+        loss_acc = torch.zeros((), device=dev)
 
         train_loader = self.train_dataloader.create_loader()
         for batch in train_loader:
@@ -268,7 +271,6 @@ class Trainer(TrainerBase):
             # would unbind the arena grad views.
             self.zero_grad()
             loss = self._batch_loss(batch, apply_privacy_metrics)
-            loss_val = loss.item()
             if prox is not None:
                 mu, w_global = prox
                 loss.backward()
@@ -276,31 +278,31 @@ class Trainer(TrainerBase):
                     # exact prox gradient: g += mu*(w - w_global)
                     ops.axpy(self.arena.grad, self.arena.data, mu)
                     ops.axpy(self.arena.grad, w_global, -mu)
-                    loss_val += 0.5 * mu * float(
-                        (self.arena.data - w_global).square().sum())
+                    loss_acc += loss.detach() + 0.5 * mu * \
+                        (self.arena.data - w_global).square().sum()
                 else:
                     off = 0
-                    reg = 0.0
+                    reg = torch.zeros((), device=dev)
                     for p in self.model.parameters():
                         wg = w_global[off:off + p.numel()].view(p.shape)
                         p.grad.add_(p.data - wg, alpha=mu)
-                        reg += 0.5 * mu * float((p.data - wg).square().sum())
+                        reg += 0.5 * mu * (p.data - wg).square().sum()
                         off += p.numel()
-                    loss_val += reg
+                    loss_acc += loss.detach() + reg
                 self.clip_gradients()
                 self.accumulate_gradient_power()
                 if self.optimizer is not None:
                     self.optimizer.step()
             else:
+                loss_acc += loss.detach()
                 self._train_step(loss)
-            sum_train_loss += loss_val
             num_samples += self._batch_samples(batch)
             self.step += 1
 
         if self.lr_scheduler is not None:
             self.lr_scheduler.step()
         self.estimate_sufficient_stats()
-        return num_samples, sum_train_loss
+        return num_samples, float(loss_acc)
 
     def run_train_epoch_fedprox(self, desired_max_samples=None,
                                 apply_privacy_metrics=False, algo_payload=None):

@@ -33,8 +33,16 @@ def resolve_task_path(*parts: str) -> str:
     return rel  # let the caller fail with a clear path
 
 
+_MODULE_CACHE = {}
+
+
 def _load_module(path: str, name: str):
-    return SourceFileLoader(name, path).load_module()
+    key = (os.path.abspath(path), name)
+    mod = _MODULE_CACHE.get(key)
+    if mod is None:
+        mod = SourceFileLoader(name, path).load_module()
+        _MODULE_CACHE[key] = mod
+    return mod
 
 
 def make_model(model_config, dataloader_type=None, input_dim=-1, output_dim=-1):

@@ -114,6 +114,15 @@ class ArrayDataLoader(BaseDataLoader):
     def create_loader(self):
         return self
 
+    def to_device(self):
+        """Move the whole shard to the GPU once; batches become device
+        slices (kills the per-batch H2D copy of the reference path)."""
+        if torch.cuda.is_available() and not self.dataset.x.is_cuda:
+            self.dataset.x = self.dataset.x.cuda(non_blocking=True)
+            if self.dataset.y is not None:
+                self.dataset.y = self.dataset.y.cuda(non_blocking=True)
+        return self
+
     def __len__(self):
         n = len(self.dataset)
         return max(1, (n + self.batch_size - 1) // self.batch_size) if n else 0
@@ -122,7 +131,11 @@ class ArrayDataLoader(BaseDataLoader):
         n = len(self.dataset)
         if n == 0:
             return
+        # shuffle order is drawn on CPU from the torch global RNG (keeps the
+        # per-client seed discipline device-independent), then moved once
         order = torch.randperm(n) if self.shuffle else torch.arange(n)
+        if self.dataset.x.is_cuda:
+            order = order.cuda(non_blocking=True)
         for s in range(0, n, self.batch_size):
             idx = order[s:s + self.batch_size]
             batch = {"x": self.dataset.x[idx]}
