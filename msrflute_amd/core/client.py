@@ -122,6 +122,19 @@ class ClientExecutor:
         self.client_strategy = strategy_cls("client", config, self.model_path)
         self.send_dicts = self.server_config.get("send_dicts", False)
 
+        # hipGraph fast path for the per-batch client step (ops/graphs.py);
+        # enabled on GPU for plain-SGD clients unless disabled by config
+        self.graph_cache = None
+        if (torch.cuda.is_available()
+                and self.client_config.get("use_hip_graphs", True)):
+            from ..ops.graphs import GraphCache
+            cache = GraphCache(
+                self.model, self.arena,
+                dict(self.client_config["optimizer_config"]),
+                self.client_config["data_config"]["train"].get("max_grad_norm"))
+            if cache.supports():
+                self.graph_cache = cache
+
     # ------------------------------------------------------------------
     def _make_optimizer(self, initial_lr):
         if self._fused_opt is not None:
@@ -183,6 +196,8 @@ class ClientExecutor:
             optimizer=optimizer,
             ss_scheduler=ss_scheduler,
             train_dataloader=train_dataloader,
+            # graph path engaged inside run_train_epoch when applicable
+
             server_replay_config=client_config,
             max_grad_norm=data_config.get("max_grad_norm", None),
             anneal_config=client_config.get("annealing_config", None),
@@ -190,6 +205,7 @@ class ClientExecutor:
             ignore_subtask=client_config["ignore_subtask"],
             arena=self.arena,
         )
+        trainer.graph_cache = self.graph_cache
 
         desired_max_samples = data_config.get("desired_max_samples", None)
         apply_privacy_metrics = bool(privacy_metrics_config

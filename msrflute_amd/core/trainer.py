@@ -153,6 +153,8 @@ class Trainer(TrainerBase):
         super().__init__(model=model, train_dataloader=train_dataloader,
                          optimizer=optimizer, max_grad_norm=max_grad_norm,
                          ignore_subtask=ignore_subtask, arena=arena)
+        # optional hipGraph fast path, injected by ClientExecutor
+        self.graph_cache = None
         self.server_replay_config = server_replay_config
         self.anneal_config = anneal_config
         self.lr_scheduler = None
@@ -190,9 +192,14 @@ class Trainer(TrainerBase):
                 self.counter += g.numel()
 
     def estimate_sufficient_stats(self):
-        """Finalize stats dict (reference: trainer.py:294-312; one host
-        sync here instead of one per tensor per batch)."""
+        """Accumulate current grad stats + finalize (reference API,
+        trainer.py:294-312: called per batch there; here the per-batch flow
+        calls accumulate_gradient_power and the epoch end finalizes once)."""
         self.accumulate_gradient_power()
+        return self._finalize_sufficient_stats()
+
+    def _finalize_sufficient_stats(self):
+        """One host sync: turn the device accumulator into the stats dict."""
         acc = self._stats_acc.tolist()
         self.sum_grad, self.sum_grad2 = acc[0], acc[1]
         n = max(self.counter, 1)
@@ -255,6 +262,10 @@ class Trainer(TrainerBase):
                         apply_privacy_metrics=False, prox=None):
         """Reference: trainer.py:341-414 (and 416-501 when ``prox`` is set:
         prox = (mu, w_global_flat))."""
+        if (self.graph_cache is not None and prox is None
+                and not apply_privacy_metrics
+                and self.graph_cache.supports()):
+            return self._run_train_epoch_graphed(desired_max_samples)
         num_samples = 0
         self.reset_gradient_power()
         self.zero_grad()
@@ -301,8 +312,63 @@ class Trainer(TrainerBase):
 
         if self.lr_scheduler is not None:
             self.lr_scheduler.step()
-        self.estimate_sufficient_stats()
+        self._finalize_sufficient_stats()
         return num_samples, float(loss_acc)
+
+    def _run_train_epoch_graphed(self, desired_max_samples=None):
+        """hipGraph fast path: each full-shape batch is one graph replay
+        (capture: zero-grad → fwd → bwd → fused clip+stats → fused SGD).
+        Ragged tail batches run the identical ops eagerly against the same
+        graph-owned accumulators/momentum so semantics match the eager
+        epoch exactly (ops/graphs.py)."""
+        cache = self.graph_cache
+        num_samples = 0
+        self.reset_gradient_power()
+        g = None
+        n_graph_batches = 0
+
+        cache.set_lr(get_lr(self.optimizer) if self.optimizer is not None
+                     else float(cache.lr_t[0]))
+        train_loader = self.train_dataloader.create_loader()
+        for batch in train_loader:
+            if desired_max_samples is not None and num_samples >= desired_max_samples:
+                break
+            x, y = batch["x"], batch["y"]
+            if g is None:
+                g = cache.get(x, y)
+                g.reset_client()
+            if tuple(x.shape) == tuple(g.static_x.shape):
+                g.run_batch(x, y)
+            else:
+                # ragged tail: same ops, eager, same accumulators
+                self.arena.grad.zero_()
+                loss = self.model.loss({"x": x, "y": y})
+                loss.backward()
+                ops.clip_stats_accumulate(
+                    self.arena.grad,
+                    float(self.max_grad_norm) if self.max_grad_norm else -1.0,
+                    g.stats_acc)
+                ops.sgd_step_devlr(self.arena.data, self.arena.grad,
+                                   g.momentum_buf, cache.lr_t,
+                                   momentum=cache.momentum, dampening=0.0,
+                                   weight_decay=cache.weight_decay,
+                                   nesterov=cache.nesterov, first_step=False)
+                g.loss_acc += loss.detach()
+            n_graph_batches += 1
+            num_samples += self._batch_samples(batch)
+            self.step += 1
+
+        if self.lr_scheduler is not None:
+            self.lr_scheduler.step()
+
+        if g is not None:
+            self._stats_acc += g.stats_acc
+            self.counter += n_graph_batches * self.arena.total
+            loss_total = float(g.loss_acc)
+        else:
+            loss_total = 0.0
+        self._finalize_sufficient_stats()
+        return num_samples, loss_total
 
     def run_train_epoch_fedprox(self, desired_max_samples=None,
                                 apply_privacy_metrics=False, algo_payload=None):

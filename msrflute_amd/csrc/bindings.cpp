@@ -18,8 +18,10 @@ void launch_clip_apply(float*, long long, const double*, float, float, float*,
                        hipStream_t);
 void launch_add_gaussian_noise(float*, long long, float, unsigned long long,
                                unsigned long long, hipStream_t);
-void launch_sgd_step(float*, const float*, float*, float, float, float, float,
-                     int, int, long long, hipStream_t);
+void launch_sgd_step(float*, const float*, float*, float, const float*, float,
+                     float, float, int, int, long long, hipStream_t);
+void launch_clip_apply_stats(float*, long long, const double*, float, float,
+                             float*, hipStream_t);
 void launch_adam_step(float*, const float*, float*, float*, float*, float,
                       float, float, float, float, float, float, int, int,
                       long long, hipStream_t);
@@ -102,9 +104,39 @@ void sgd_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
   }
   launch_sgd_step(p.data_ptr<float>(), g.data_ptr<float>(),
                   buf.numel() ? buf.data_ptr<float>() : nullptr, (float)lr,
-                  (float)momentum, (float)dampening, (float)weight_decay,
-                  nesterov ? 1 : 0, first_step ? 1 : 0, p.numel(),
-                  cur_stream());
+                  nullptr, (float)momentum, (float)dampening,
+                  (float)weight_decay, nesterov ? 1 : 0, first_step ? 1 : 0,
+                  p.numel(), cur_stream());
+}
+
+// graph-replayable SGD: lr read from a 1-element device tensor
+void sgd_step_devlr(torch::Tensor p, torch::Tensor g, torch::Tensor buf,
+                    torch::Tensor lr_t, double momentum, double dampening,
+                    double weight_decay, bool nesterov, bool first_step) {
+  check_flat(p, "p"); check_flat(g, "g"); check_flat(lr_t, "lr_t");
+  if (momentum != 0.0) {
+    check_flat(buf, "buf");
+    TORCH_CHECK(buf.numel() == p.numel(), "momentum buffer size mismatch");
+  }
+  launch_sgd_step(p.data_ptr<float>(), g.data_ptr<float>(),
+                  buf.numel() ? buf.data_ptr<float>() : nullptr, 0.f,
+                  lr_t.data_ptr<float>(), (float)momentum, (float)dampening,
+                  (float)weight_decay, nesterov ? 1 : 0, first_step ? 1 : 0,
+                  p.numel(), cur_stream());
+}
+
+// fused per-batch clip + sufficient-stats accumulate (one reduction pass):
+// clips x to max_norm and adds post-clip {Σx, Σx²} into stats_acc[0..1]
+void clip_stats_accumulate(torch::Tensor x, double max_norm, double eps,
+                           torch::Tensor stats_acc) {
+  check_flat(x, "x"); check_flat(stats_acc, "stats_acc");
+  TORCH_CHECK(stats_acc.numel() >= 2, "stats_acc must have 2 elements");
+  auto acc = torch::zeros({2}, x.options().dtype(torch::kFloat64));
+  launch_sum_sumsq(x.data_ptr<float>(), x.numel(), acc.data_ptr<double>(),
+                   cur_stream());
+  launch_clip_apply_stats(x.data_ptr<float>(), x.numel(),
+                          acc.data_ptr<double>(), (float)max_norm, (float)eps,
+                          stats_acc.data_ptr<float>(), cur_stream());
 }
 
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
@@ -169,6 +201,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("clip_by_norm", &clip_by_norm);
   m.def("add_gaussian_noise", &add_gaussian_noise);
   m.def("sgd_step", &sgd_step);
+  m.def("sgd_step_devlr", &sgd_step_devlr);
+  m.def("clip_stats_accumulate", &clip_stats_accumulate);
   m.def("adam_step", &adam_step);
   m.def("adamax_step", &adamax_step);
   m.def("segmented_sqnorm", &segmented_sqnorm);

@@ -160,6 +160,36 @@ __global__ void k_clip_apply(float* __restrict__ x, long long n,
     x[i] *= coef;
 }
 
+// Fused clip + sufficient stats: given out2 = {Σg, Σg²} from k_sum_sumsq,
+// scale g by coef = min(1, max_norm/(‖g‖+eps)) and ACCUMULATE the
+// post-clip stats {coef·Σg, coef²·Σg²} into stats_acc[0..1] — the whole
+// per-batch clip+stats pipeline (K5+K6) costs ONE reduction pass.
+__global__ void k_clip_apply_stats(float* __restrict__ x, long long n,
+                                   const double* __restrict__ out2,
+                                   float max_norm, float eps,
+                                   float* __restrict__ stats_acc) {
+  double norm = sqrt(out2[1]);
+  float coef = (float)(max_norm / (norm + (double)eps));
+  if (coef > 1.0f || max_norm <= 0.f) coef = 1.0f;
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    atomicAdd(&stats_acc[0], (float)(out2[0] * coef));
+    atomicAdd(&stats_acc[1], (float)(out2[1] * coef * coef));
+  }
+  if (coef >= 1.0f) return;
+  long long n4 = n >> 2;
+  float4* x4 = reinterpret_cast<float4*>(x);
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n4;
+       i += (long long)gridDim.x * blockDim.x) {
+    float4 a = x4[i];
+    a.x *= coef; a.y *= coef; a.z *= coef; a.w *= coef;
+    x4[i] = a;
+  }
+  long long tail = n4 << 2;
+  for (long long i = tail + blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < n; i += (long long)gridDim.x * blockDim.x)
+    x[i] *= coef;
+}
+
 // ---------------------------------------------------------------------------
 // Philox Gaussian noise (DP): x[i] += sigma * N(0,1)
 // ---------------------------------------------------------------------------
@@ -186,9 +216,13 @@ __global__ void k_add_gaussian_noise(float* __restrict__ x, long long n,
 // ---------------------------------------------------------------------------
 
 __global__ void k_sgd_step(float* __restrict__ p, const float* __restrict__ g,
-                           float* __restrict__ buf, float lr, float momentum,
+                           float* __restrict__ buf, float lr,
+                           const float* __restrict__ lr_ptr, float momentum,
                            float dampening, float weight_decay, int nesterov,
                            int first_step, long long n) {
+  // lr from a device scalar when lr_ptr != null — lets a hipGraph replay
+  // the step while the host retunes the LR between rounds
+  if (lr_ptr) lr = *lr_ptr;
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n;
        i += (long long)gridDim.x * blockDim.x) {
     float dp = g[i];
@@ -346,11 +380,19 @@ void launch_add_gaussian_noise(float* x, long long n, float sigma,
 }
 
 void launch_sgd_step(float* p, const float* g, float* buf, float lr,
-                     float momentum, float dampening, float weight_decay,
-                     int nesterov, int first_step, long long n, hipStream_t s) {
+                     const float* lr_ptr, float momentum, float dampening,
+                     float weight_decay, int nesterov, int first_step,
+                     long long n, hipStream_t s) {
   hipLaunchKernelGGL(k_sgd_step, dim3(grid_for(n)), dim3(BLOCK), 0, s,
-                     p, g, buf, lr, momentum, dampening, weight_decay,
+                     p, g, buf, lr, lr_ptr, momentum, dampening, weight_decay,
                      nesterov, first_step, n);
+}
+
+void launch_clip_apply_stats(float* x, long long n, const double* out2,
+                             float max_norm, float eps, float* stats_acc,
+                             hipStream_t s) {
+  hipLaunchKernelGGL(k_clip_apply_stats, dim3(grid_for(n >> 2)), dim3(BLOCK),
+                     0, s, x, n, out2, max_norm, eps, stats_acc);
 }
 
 void launch_adam_step(float* p, const float* g, float* m, float* v,

@@ -109,6 +109,34 @@ def sgd_step(param, grad, momentum_buf, *, lr, momentum=0.0, dampening=0.0,
                  nesterov=nesterov, first_step=first_step)
 
 
+def sgd_step_devlr(param, grad, momentum_buf, lr_t, *, momentum=0.0,
+                   dampening=0.0, weight_decay=0.0, nesterov=False,
+                   first_step=False):
+    """SGD step with the LR read from a 1-element device tensor — the form
+    a hipGraph can replay across rounds while the host retunes the LR."""
+    if _use_ext(param):
+        _C.sgd_step_devlr(param, grad,
+                          momentum_buf if momentum_buf is not None else param.new_empty(0),
+                          lr_t, float(momentum), float(dampening),
+                          float(weight_decay), bool(nesterov), bool(first_step))
+        return
+    ref.sgd_step(param, grad, momentum_buf, lr=float(lr_t.item()),
+                 momentum=momentum, dampening=dampening,
+                 weight_decay=weight_decay, nesterov=nesterov,
+                 first_step=first_step)
+
+
+def clip_stats_accumulate(x, max_norm, stats_acc, eps: float = 1e-6):
+    """Fused per-batch clip-to-norm + post-clip {Σx, Σx²} accumulation into
+    ``stats_acc`` (one reduction pass; pass max_norm <= 0 to skip clipping)."""
+    if _use_ext(x):
+        _C.clip_stats_accumulate(x, float(max_norm), float(eps), stats_acc)
+        return
+    if max_norm is not None and max_norm > 0:
+        ref.clip_by_norm(x, max_norm, eps)
+    stats_acc += ref.sum_sumsq(x)
+
+
 def adam_step(param, grad, exp_avg, exp_avg_sq, max_exp_avg_sq=None, *, step,
               lr, beta1=0.9, beta2=0.999, eps=1e-8, weight_decay=0.0,
               amsgrad=False, adamw=False):

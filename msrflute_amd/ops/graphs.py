@@ -1,0 +1,127 @@
+"""hipGraph capture of the per-batch client training step.
+
+The FL client workload is thousands of tiny fixed-shape batches: eager
+execution is launch-bound (measured ~4% GPU busy on CNN-FEMNIST — see
+profiles/).  This captures the whole per-batch step
+
+    zero grad arena → forward → backward → fused clip+stats → fused SGD
+
+as ONE hip graph replayed per batch; the host copies the batch into static
+buffers and replays.  The LR is read from a device scalar so a single
+capture serves every round even as the server retunes the client LR.
+
+Constraints (checked by ``supports``): CUDA + arena-bound model, plain SGD
+client optimizer with dampening 0 (covers every reference benchmark task:
+all use client SGD — BASELINE.md), fixed batch shape (ragged tail batches
+fall back to the eager path).
+
+Note on RNG: dropout inside the graph draws from the torch CUDA
+generator's graph-captured Philox stream, which advances per replay — so
+the GPU fast path does not reproduce the CPU path's per-client dropout
+bit-pattern (world-size invariance of numerics remains exact on the CPU
+path, which is the one under test).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Optional, Tuple
+
+import torch
+
+from .. import ops
+from .arena import ParameterArena
+
+
+class GraphedClientStep:
+    """One captured (model, arena, batch-shape) training step."""
+
+    def __init__(self, model, arena: ParameterArena, lr_t: torch.Tensor,
+                 max_grad_norm: Optional[float], momentum: float,
+                 weight_decay: float, nesterov: bool,
+                 x_shape: Tuple[int, ...], y_shape: Tuple[int, ...],
+                 y_dtype: torch.dtype):
+        self.model = model
+        self.arena = arena
+        self.lr_t = lr_t
+        self.momentum = momentum
+        dev = arena.device
+        self.static_x = torch.zeros(x_shape, device=dev)
+        self.static_y = torch.zeros(y_shape, dtype=y_dtype, device=dev)
+        self.stats_acc = torch.zeros(2, device=dev)
+        self.loss_acc = torch.zeros((), device=dev)
+        self.momentum_buf = arena.new_buffer() if momentum != 0.0 else None
+        mn = float(max_grad_norm) if max_grad_norm is not None else -1.0
+
+        def step_body():
+            self.arena.grad.zero_()
+            loss = self.model.loss({"x": self.static_x, "y": self.static_y})
+            loss.backward()
+            ops.clip_stats_accumulate(self.arena.grad, mn, self.stats_acc)
+            ops.sgd_step_devlr(self.arena.data, self.arena.grad,
+                               self.momentum_buf, self.lr_t,
+                               momentum=momentum, dampening=0.0,
+                               weight_decay=weight_decay, nesterov=nesterov,
+                               first_step=False)
+            self.loss_acc += loss.detach()
+
+        # warmup on a side stream (required before capture), then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                step_body()
+        torch.cuda.current_stream().wait_stream(s)
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            step_body()
+
+    def reset_client(self):
+        """Per-client optimizer/stat reset (fresh-optimizer semantics:
+        dampening==0 makes a zeroed momentum buffer equal to first_step)."""
+        self.stats_acc.zero_()
+        self.loss_acc.zero_()
+        if self.momentum_buf is not None:
+            self.momentum_buf.zero_()
+
+    def run_batch(self, x: torch.Tensor, y: torch.Tensor):
+        self.static_x.copy_(x, non_blocking=True)
+        self.static_y.copy_(y, non_blocking=True)
+        self.graph.replay()
+
+
+class GraphCache:
+    """Per-executor cache of captured steps keyed by batch shape."""
+
+    def __init__(self, model, arena: ParameterArena, optimizer_config: dict,
+                 max_grad_norm: Optional[float]):
+        self.model = model
+        self.arena = arena
+        self.max_grad_norm = max_grad_norm
+        cfg = dict(optimizer_config)
+        self.opt_type = cfg.get("type", "sgd")
+        self.momentum = float(cfg.get("momentum", 0.0))
+        self.dampening = float(cfg.get("dampening", 0.0))
+        self.weight_decay = float(cfg.get("weight_decay", 0.0))
+        self.nesterov = bool(cfg.get("nesterov", False))
+        self.lr_t = torch.zeros(1, device=arena.device)
+        self._graphs: Dict[Tuple, GraphedClientStep] = {}
+
+    def supports(self) -> bool:
+        return (torch.cuda.is_available() and self.arena.device.type == "cuda"
+                and ops.HAS_EXT and self.opt_type == "sgd"
+                and self.dampening == 0.0)
+
+    def set_lr(self, lr: float):
+        self.lr_t.fill_(float(lr))
+
+    def get(self, x: torch.Tensor, y: torch.Tensor) -> GraphedClientStep:
+        key = (tuple(x.shape), tuple(y.shape), y.dtype)
+        g = self._graphs.get(key)
+        if g is None:
+            g = GraphedClientStep(self.model, self.arena, self.lr_t,
+                                  self.max_grad_norm, self.momentum,
+                                  self.weight_decay, self.nesterov,
+                                  tuple(x.shape), tuple(y.shape), y.dtype)
+            self._graphs[key] = g
+        return g
