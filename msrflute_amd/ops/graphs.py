@@ -64,7 +64,10 @@ class GraphedClientStep:
                                first_step=False)
             self.loss_acc += loss.detach()
 
-        # warmup on a side stream (required before capture), then capture
+        # warmup on a side stream (required before capture), then capture.
+        # warmup EXECUTES the step and would corrupt the live client
+        # weights — snapshot and restore around it.
+        saved = arena.data.clone()
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
@@ -75,6 +78,11 @@ class GraphedClientStep:
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             step_body()
+
+        arena.data.copy_(saved)
+        arena.grad.zero_()
+        self.reset_client()
+        torch.cuda.synchronize()
 
     def reset_client(self):
         """Per-client optimizer/stat reset (fresh-optimizer semantics:
