@@ -78,7 +78,7 @@ class Client:
                         if trainset_unlab is not None else [train_dataset])
         else:
             datasets = [dataset]
-        data_with_labels = hasattr(datasets[0], "user_data_label")
+        data_with_labels = getattr(datasets[0], "user_data_label", None) is not None
         strcts = []
         for ds in datasets:
             s = {"users": [], "num_samples": [], "user_data": {}}

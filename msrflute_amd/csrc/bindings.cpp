@@ -31,6 +31,8 @@ void launch_segmented_sqnorm(const float*, const long long*, int, double*,
                              hipStream_t);
 void launch_quant_bin_mask(float*, long long, const float*, const float*,
                            const float*, int, hipStream_t);
+void launch_gru_gates(const float*, const float*, const float*, float*, int,
+                      int, hipStream_t);
 }
 
 namespace {
@@ -190,6 +192,20 @@ void quant_bin_mask(torch::Tensor x, torch::Tensor min_t, torch::Tensor max_t,
                         thresh_t.data_ptr<float>(), (int)n_bins, cur_stream());
 }
 
+// fused GRU gate math (no-grad eval path of the nlg_gru recurrence)
+torch::Tensor gru_gates(torch::Tensor g_i, torch::Tensor g_h,
+                        torch::Tensor h) {
+  check_flat(g_i, "g_i"); check_flat(g_h, "g_h"); check_flat(h, "h");
+  int B = (int)h.size(0), H = (int)h.size(1);
+  TORCH_CHECK(g_i.numel() == 3LL * B * H && g_h.numel() == 3LL * B * H,
+              "g_i/g_h must be [B, 3H]");
+  auto out = torch::empty_like(h);
+  launch_gru_gates(g_i.data_ptr<float>(), g_h.data_ptr<float>(),
+                   h.data_ptr<float>(), out.data_ptr<float>(), B, H,
+                   cur_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -207,4 +223,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamax_step", &adamax_step);
   m.def("segmented_sqnorm", &segmented_sqnorm);
   m.def("quant_bin_mask", &quant_bin_mask);
+  m.def("gru_gates", &gru_gates);
 }

@@ -178,3 +178,13 @@ def quantize_dequantize(x, n_bins: int, threshold_quantile: float):
                           thresh_t.reshape(1).float().contiguous(), int(n_bins))
         return x
     return ref.quantize_dequantize(x, n_bins, threshold_quantile)
+
+
+def gru_gates(g_i, g_h, h):
+    """GRU gate fusion for the nlg_gru recurrence.  Autograd is required on
+    the training path, so the composite torch expression is used whenever
+    grad mode is on; the no-grad HIP fused kernel serves eval/serving."""
+    if (not torch.is_grad_enabled() or not g_i.requires_grad) and \
+            g_i.is_cuda and _use_ext(g_i) and hasattr(_C, "gru_gates"):
+        return _C.gru_gates(g_i.contiguous(), g_h.contiguous(), h.contiguous())
+    return ref.gru_gates(g_i, g_h, h)

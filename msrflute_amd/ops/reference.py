@@ -165,3 +165,15 @@ def quantize_dequantize(x: torch.Tensor, n_bins: int, threshold_quantile: float)
     binned = bins[idx.clamp_(0, n_bins - 1)]
     x.copy_(torch.where(x.abs() > thresh, binned, torch.zeros_like(x)))
     return x
+
+
+def gru_gates(g_i: torch.Tensor, g_h: torch.Tensor, h: torch.Tensor):
+    """Fused GRU gate math given precomputed input/hidden projections
+    (reference cell: experiments/nlg_gru/model.py:20-28).
+    g_i, g_h: [B, 3H]; h: [B, H] -> new hidden [B, H]."""
+    i_r, i_i, i_n = g_i.chunk(3, dim=-1)
+    h_r, h_i, h_n = g_h.chunk(3, dim=-1)
+    reset = torch.sigmoid(i_r + h_r)
+    update = torch.sigmoid(i_i + h_i)
+    new = torch.tanh(i_n + reset * h_n)
+    return new + update * (h - new)

@@ -65,6 +65,44 @@ def make_char_lm_blob(n_users, samples_per_user, seq_len=80, vocab=90, seed=0):
             "user_data": user_data, "user_data_label": user_data_label}
 
 
+def make_fedcifar100_blob(n_users=500, samples_per_user=100, seed=0):
+    return make_classification_blob(n_users, samples_per_user, (3, 24, 24),
+                                    100, seed=seed)
+
+
+def make_cifar10_blob(n_users=50, samples_per_user=100, seed=0):
+    return make_classification_blob(n_users, samples_per_user, (3, 32, 32),
+                                    10, seed=seed)
+
+
+def make_ecg_blob(n_users=50, samples_per_user=100, seed=0):
+    return make_classification_blob(n_users, samples_per_user, (187,), 5,
+                                    seed=seed, flat=True)
+
+
+def make_shakespeare_blob(n_users=715, samples_per_user=50, seed=0):
+    return make_char_lm_blob(n_users, samples_per_user, seq_len=80, vocab=90,
+                             seed=seed)
+
+
+def make_nlg_blob(n_users=25, utts_per_user=20, vocab=1000, max_words=25,
+                  seed=0):
+    """Reddit-style variable-length preencoded utterances (nlg_gru /
+    mlm_bert shape)."""
+    rng = np.random.default_rng(seed)
+    users, num_samples, user_data = [], [], {}
+    for u in range(n_users):
+        name = f"user{u:05d}"
+        utts = [rng.integers(1, vocab, size=int(
+            rng.integers(3, max_words + 1))).tolist()
+            for _ in range(utts_per_user)]
+        users.append(name)
+        num_samples.append(len(utts))
+        user_data[name] = {"x": utts}
+    return {"users": users, "num_samples": num_samples,
+            "user_data": user_data}
+
+
 def save_blob(blob, path):
     os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
     if path.endswith(".pt"):
@@ -87,6 +125,13 @@ def save_blob(blob, path):
 TASKS = {
     "cv_lr_mnist": (make_mnist_blob, dict(n_users=1000, samples_per_user=60)),
     "cv_cnn_femnist": (make_femnist_blob, dict(n_users=3400, samples_per_user=100)),
+    "cv_resnet_fedcifar100": (make_fedcifar100_blob,
+                              dict(n_users=500, samples_per_user=100)),
+    "classif_cnn": (make_cifar10_blob, dict(n_users=50, samples_per_user=100)),
+    "ecg_cnn": (make_ecg_blob, dict(n_users=50, samples_per_user=100)),
+    "nlp_rnn_fedshakespeare": (make_shakespeare_blob,
+                               dict(n_users=715, samples_per_user=50)),
+    "nlg_gru": (make_nlg_blob, dict(n_users=25, utts_per_user=20)),
 }
 
 
