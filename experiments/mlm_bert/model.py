@@ -1,0 +1,127 @@
+"""mlm_bert task: HuggingFace masked-LM wrapper (BERT/RoBERTa family).
+
+Reference: experiments/mlm_bert/model.py:39-473.  Capabilities kept:
+label smoothing, gradient accumulation, bottleneck adapters (base model
+frozen), perplexity/masked-accuracy eval.  Offline-first divergence: with
+no network, ``model_name_or_path`` must be a local directory; otherwise
+the architecture is built from explicit config keys (``vocab_size``,
+``hidden_size``, ``num_hidden_layers``, ``num_attention_heads``, ...)
+with random init via ``BertForMaskedLM(BertConfig(...))``.
+"""
+
+import torch
+from torch import nn
+
+from msrflute_amd.core.model import BaseModel
+from msrflute_amd.utils import to_device
+
+
+def _model_args(model_config):
+    args = model_config.get("BERT", model_config)
+    return args.get("model", args), args.get("training", {})
+
+
+class BottleneckAdapter(nn.Module):
+    """Houlsby-style adapter: LN -> down -> GELU -> up + residual."""
+
+    def __init__(self, hidden, bottleneck=64):
+        super().__init__()
+        self.ln = nn.LayerNorm(hidden)
+        self.down = nn.Linear(hidden, bottleneck)
+        self.up = nn.Linear(bottleneck, hidden)
+        nn.init.zeros_(self.up.weight)
+        nn.init.zeros_(self.up.bias)
+
+    def forward(self, x):
+        return x + self.up(torch.nn.functional.gelu(self.down(self.ln(x))))
+
+
+class BERT(BaseModel):
+    def __init__(self, model_config, **kwargs):
+        super().__init__()
+        from transformers import AutoConfig, AutoModelForMaskedLM, BertConfig, \
+            BertForMaskedLM
+
+        model_args, training_args = _model_args(model_config)
+        self.gradient_accumulation_steps = int(
+            model_args.get("gradient_accumulation_steps", 1))
+        smoothing = float(training_args.get("label_smoothing_factor", 0.0))
+        self.label_smoothing = smoothing
+
+        path = model_args.get("model_name_or_path")
+        import os
+        if path and os.path.isdir(str(path)):
+            config = AutoConfig.from_pretrained(path)
+            self.model = AutoModelForMaskedLM.from_pretrained(path,
+                                                              config=config)
+        else:
+            config = BertConfig(
+                vocab_size=int(model_args.get("vocab_size", 1000)),
+                hidden_size=int(model_args.get("hidden_size", 128)),
+                num_hidden_layers=int(model_args.get("num_hidden_layers", 2)),
+                num_attention_heads=int(model_args.get("num_attention_heads", 2)),
+                intermediate_size=int(model_args.get("intermediate_size", 512)),
+                max_position_embeddings=int(
+                    model_args.get("max_position_embeddings", 512)))
+            self.model = BertForMaskedLM(config)
+        self.output_layer_size = config.vocab_size
+
+        if model_args.get("adapter", False):
+            self._add_adapters(config.hidden_size,
+                               int(model_args.get("adapter_dim", 64)))
+
+    def _add_adapters(self, hidden, bottleneck):
+        """Freeze the base model; train only bottleneck adapters appended to
+        every encoder layer output (reference adapter path: model.py:126-131)."""
+        for p in self.model.parameters():
+            p.requires_grad = False
+        layers = self.model.bert.encoder.layer if hasattr(self.model, "bert") \
+            else self.model.base_model.encoder.layer
+        self.adapters = nn.ModuleList()
+        for layer in layers:
+            adapter = BottleneckAdapter(hidden, bottleneck)
+            self.adapters.append(adapter)
+            out_mod = layer.output
+
+            def hook(mod, inputs, output, _a=adapter):
+                return _a(output)
+            out_mod.register_forward_hook(hook)
+
+    def _prepare(self, inputs):
+        return {k: to_device(v) for k, v in inputs.items()
+                if k in ("input_ids", "attention_mask", "labels",
+                         "token_type_ids")}
+
+    def _loss_from_logits(self, logits, labels):
+        return nn.functional.cross_entropy(
+            logits.view(-1, self.output_layer_size), labels.view(-1),
+            ignore_index=-100, label_smoothing=self.label_smoothing)
+
+    def loss(self, inputs):
+        inputs = self._prepare(inputs)
+        labels = inputs.pop("labels")
+        out = self.model(**inputs)
+        loss = self._loss_from_logits(out.logits, labels)
+        return loss / self.gradient_accumulation_steps
+
+    def inference(self, inputs):
+        inputs = self._prepare(inputs)
+        labels = inputs.pop("labels")
+        out = self.model(**inputs)
+        loss = self._loss_from_logits(out.logits, labels)
+        mask = labels != -100
+        n_masked = int(mask.sum().item())
+        pred = out.logits.argmax(dim=-1)
+        acc = ((pred == labels) & mask).sum().item() / max(n_masked, 1)
+        batch_size = inputs["input_ids"].shape[0]
+        return {"output": loss.item(), "acc": acc, "batch_size": batch_size,
+                "perplexity": {"value": float(torch.exp(loss).item()),
+                               "higher_is_better": False}}
+
+    def set_eval(self):
+        self.eval()
+        self.model.eval()
+
+    def set_train(self):
+        self.train()
+        self.model.train()

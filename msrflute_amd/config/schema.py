@@ -49,7 +49,8 @@ SCHEMA = {
                         "schema": {
                             "model_name_or_path": {"type": "string"},
                             "model_name": {"required": True, "type": "string"},
-                            "process_line_by_line": {"required": True, "type": "boolean"},
+                            "process_line_by_line": {"type": "boolean",
+                                                     "default": False},
                         },
                     },
                 },

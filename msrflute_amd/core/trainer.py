@@ -249,7 +249,10 @@ class Trainer(TrainerBase):
             return int(torch.sum(batch["attention_mask"].detach().cpu() == 1).item())
         if "total_frames" in batch:
             return int(batch["total_frames"])
-        return len(batch["x"])
+        x = batch["x"]
+        if isinstance(x, (tuple, list)):  # multi-input tasks (fednewsrec)
+            x = x[0]
+        return len(x)
 
     def _train_step(self, loss):
         loss.backward()

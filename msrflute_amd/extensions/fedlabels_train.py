@@ -108,7 +108,10 @@ def run_train_epoch_sup(trainer, desired_max_samples=None,
 
     if round_ >= cfg["burnout_round"]:
         for _ in range(int(cfg["unsuptrain_ep"])):
-            data_idx = random.sample(range(len(unsupdataset)), cfg["unl_bs"])
+            unl_bs = min(int(cfg["unl_bs"]), len(unsupdataset))
+            if unl_bs == 0:
+                break
+            data_idx = random.sample(range(len(unsupdataset)), unl_bs)
             ldr = DataLoader(Subset(unsupdataset, indices=data_idx),
                              batch_size=cfg["bs"], shuffle=False)
             images, true_labels = next(iter(ldr))

@@ -23,6 +23,10 @@ TASK_MATRIX = {
     "ecg_cnn": {},
     "nlp_rnn_fedshakespeare": {},
     "nlg_gru": {},
+    "mlm_bert": {},
+    "cv": {},
+    "semisupervision": {},
+    "fednewsrec": {},
 }
 
 
@@ -36,10 +40,17 @@ def _make_data(task, data_dir, n_users=12, samples=8):
         kw["samples_per_user"] = samples
     if "utts_per_user" in kw:
         kw["utts_per_user"] = samples
+    if "n_samples" in kw:
+        kw["n_samples"] = 400
     train = fn(seed=0, **kw)
     kw_eval = dict(kw, n_users=3)
-    val = fn(seed=1, **kw_eval)
-    test = fn(seed=2, **kw_eval)
+    if task == "fednewsrec":
+        from tools.create_data import make_newsrec_eval_blob
+        val = make_newsrec_eval_blob(seed=1, **kw_eval)
+        test = make_newsrec_eval_blob(seed=2, **kw_eval)
+    else:
+        val = fn(seed=1, **kw_eval)
+        test = fn(seed=2, **kw_eval)
     cd.save_blob(train, os.path.join(data_dir, task, "train_data.pt"))
     cd.save_blob(val, os.path.join(data_dir, task, "val_data.pt"))
     cd.save_blob(test, os.path.join(data_dir, task, "test_data.pt"))
@@ -62,6 +73,20 @@ def _shrink_config(task, tmp_path):
         cfg["model_config"].update(hid_size=16)
     if task == "cv_resnet_fedcifar100":
         cfg["model_config"].update(num_classes=100)
+    if task == "semisupervision":
+        cfg["model_config"].update(num_classes=10)
+        cfg["client_config"]["semisupervision"].update(num_classes=10,
+                                                       unl_bs=3, bs=3)
+        cfg["client_config"]["data_config"]["train"][
+            "num_labeled_per_user"] = 4
+    if task == "fednewsrec":
+        cfg["model_config"].update(embed_dim=32)  # vocab must match the blob
+    if task == "cv":
+        for sec in (cfg["client_config"]["data_config"]["train"],
+                    cfg["server_config"]["data_config"]["val"],
+                    cfg["server_config"]["data_config"]["test"]):
+            sec["total_num_clients"] = 8
+        sc["num_clients_per_iteration"] = 2
     p = tmp_path / f"{task}.yaml"
     with open(p, "w") as f:
         yaml.safe_dump(cfg, f)

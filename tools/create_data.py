@@ -103,6 +103,50 @@ def make_nlg_blob(n_users=25, utts_per_user=20, vocab=1000, max_words=25,
             "user_data": user_data}
 
 
+def make_newsrec_blob(n_users=20, samples_per_user=8, vocab=5000, hist=50,
+                      title=30, k=1 + 4, seed=0):
+    """MIND-style pretokenized click/candidate slates (fednewsrec shape).
+    Train labels: clicked index; eval consumes the same as binary-ish."""
+    rng = np.random.default_rng(seed)
+    users, num_samples, user_data, user_labels = [], [], {}, {}
+    for u in range(n_users):
+        name = f"user{u:05d}"
+        n = samples_per_user
+        user_data[name] = {
+            "history": rng.integers(1, vocab, size=(n, hist, title)),
+            "candidates": rng.integers(1, vocab, size=(n, k, title)),
+            "labels": rng.integers(0, k, size=n),
+        }
+        users.append(name)
+        num_samples.append(n)
+        user_labels[name] = user_data[name]["labels"]
+    return {"users": users, "num_samples": num_samples,
+            "user_data": user_data, "user_data_label": user_labels}
+
+
+def make_newsrec_eval_blob(n_users=4, samples_per_user=8, vocab=5000,
+                           hist=50, title=30, k=1 + 4, seed=0):
+    """Eval variant: labels are [n, k] binary relevance vectors."""
+    blob = make_newsrec_blob(n_users, samples_per_user, vocab, hist, title,
+                             k, seed)
+    rng = np.random.default_rng(seed + 99)
+    for u in blob["users"]:
+        n = len(blob["user_data"][u]["labels"])
+        lab = np.zeros((n, k), dtype=np.int64)
+        lab[np.arange(n), rng.integers(0, k, size=n)] = 1
+        blob["user_data"][u]["labels"] = lab
+        blob["user_data_label"][u] = lab
+    return blob
+
+
+def make_cv_flat_blob(n_samples=2000, seed=0, img=32, n_classes=10, **kw):
+    """Flat (non-federated) CIFAR-shaped arrays; the cv task partitions
+    them Dirichlet-style on load (experiments/cv/data.py)."""
+    rng = np.random.default_rng(seed)
+    return {"x": rng.standard_normal((n_samples, 3, img, img)).astype(np.float32),
+            "y": rng.integers(0, n_classes, size=n_samples)}
+
+
 def save_blob(blob, path):
     os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
     if path.endswith(".pt"):
@@ -132,6 +176,11 @@ TASKS = {
     "nlp_rnn_fedshakespeare": (make_shakespeare_blob,
                                dict(n_users=715, samples_per_user=50)),
     "nlg_gru": (make_nlg_blob, dict(n_users=25, utts_per_user=20)),
+    "mlm_bert": (make_nlg_blob, dict(n_users=25, utts_per_user=20)),
+    "cv": (make_cv_flat_blob, dict(n_samples=2000)),
+    "semisupervision": (make_cifar10_blob,
+                        dict(n_users=20, samples_per_user=40)),
+    "fednewsrec": (make_newsrec_blob, dict(n_users=20, samples_per_user=8)),
 }
 
 
