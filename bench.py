@@ -164,6 +164,13 @@ def main():
     peak_mb = (torch.cuda.max_memory_allocated() / 2 ** 20
                if torch.cuda.is_available() else 0.0)
     value = args.steps / elapsed
+    if os.environ.get("BENCH_STATS") and rt.rank == 0:
+        acc = dict(server.executor.perf_acc)
+        n = max(acc.pop("clients", 1), 1)
+        per_client = {k: round(v / n * 1000, 3) for k, v in acc.items()}
+        print(json.dumps({"bench_stats_ms_per_client": per_client,
+                          "clients_processed": n,
+                          "ms_per_round_wall": elapsed / args.steps * 1000}))
     if rt.rank == 0:
         print(json.dumps({
             "metric": "fl_rounds_per_sec",

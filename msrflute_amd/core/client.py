@@ -121,6 +121,7 @@ class ClientExecutor:
         strategy_cls = select_strategy(config["strategy"])
         self.client_strategy = strategy_cls("client", config, self.model_path)
         self.send_dicts = self.server_config.get("send_dicts", False)
+        self.perf_acc = {}  # cumulative phase timings (bench diagnostics)
 
         # hipGraph fast path for the per-batch client step (ops/graphs.py);
         # enabled on GPU for plain-SGD clients unless disabled by config
@@ -168,11 +169,13 @@ class ClientExecutor:
         # reduction order.
         torch.manual_seed(round_seed & 0x7FFFFFFFFFFF)
 
+        t_dl = time.time()
         train_dataloader = make_train_dataloader(
             data_config, self.data_path, task=self.task, clientx=0,
             data_strct=data_strct)
         if hasattr(train_dataloader, "to_device"):
             train_dataloader.to_device()
+        client_stats["dataloader"] = time.time() - t_dl
 
         # one flat copy-in instead of the reference's per-tensor clone loop
         # (client.py:294-301, K13)
@@ -254,6 +257,9 @@ class ClientExecutor:
 
         client_stats["training"] = time.time() - begin_training
         client_stats["full cost"] = time.time() - begin
+        for k, v in client_stats.items():
+            self.perf_acc[k] = self.perf_acc.get(k, 0.0) + v
+        self.perf_acc["clients"] = self.perf_acc.get("clients", 0) + 1
 
         client_output = {
             "cs": client_stats,

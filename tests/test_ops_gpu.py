@@ -137,3 +137,15 @@ def test_quantize_kernel_matches_reference():
     assert close.float().mean().item() > 0.9999
     frac_zero = (got == 0).float().mean().item()
     assert 0.45 < frac_zero < 0.55
+
+
+def test_gru_gates_kernel_matches_reference():
+    torch.manual_seed(80)
+    B, H = 33, 129
+    g_i = torch.randn(B, 3 * H, device="cuda")
+    g_h = torch.randn(B, 3 * H, device="cuda")
+    h = torch.randn(B, H, device="cuda")
+    with torch.no_grad():
+        got = ops.gru_gates(g_i, g_h, h)
+    expected = ref.gru_gates(g_i.cpu(), g_h.cpu(), h.cpu())
+    assert torch.allclose(got.cpu(), expected, rtol=1e-4, atol=1e-5)
