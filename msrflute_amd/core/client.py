@@ -123,6 +123,16 @@ class ClientExecutor:
         self.send_dicts = self.server_config.get("send_dicts", False)
         self.perf_acc = {}  # cumulative phase timings (bench diagnostics)
 
+        # Device-resident client shard cache (SURVEY.md §7.1 divergence 3):
+        # a client's packed (x, y) tensors live in HBM after first use, so
+        # repeat sampling of a client costs zero host->device traffic and
+        # zero Dataset re-construction.  288 GB/GPU holds every reference
+        # dataset's full shard set; budget-capped for safety.
+        self._shard_cache: Dict[str, object] = {}
+        self._shard_cache_bytes = 0
+        self._shard_cache_budget = int(self.client_config.get(
+            "shard_cache_gb", 64)) << 30
+
         # hipGraph fast path for the per-batch client step (ops/graphs.py);
         # enabled on GPU for plain-SGD clients unless disabled by config
         self.graph_cache = None
@@ -170,11 +180,25 @@ class ClientExecutor:
         torch.manual_seed(round_seed & 0x7FFFFFFFFFFF)
 
         t_dl = time.time()
-        train_dataloader = make_train_dataloader(
-            data_config, self.data_path, task=self.task, clientx=0,
-            data_strct=data_strct)
-        if hasattr(train_dataloader, "to_device"):
-            train_dataloader.to_device()
+        train_dataloader = self._shard_cache.get(user)
+        if train_dataloader is None:
+            train_dataloader = make_train_dataloader(
+                data_config, self.data_path, task=self.task, clientx=0,
+                data_strct=data_strct)
+            if hasattr(train_dataloader, "to_device"):
+                train_dataloader.to_device()
+            if (torch.cuda.is_available()
+                    and hasattr(train_dataloader, "dataset")
+                    and getattr(train_dataloader.dataset, "x", None) is not None
+                    and torch.is_tensor(train_dataloader.dataset.x)
+                    and train_dataloader.dataset.x.is_cuda
+                    and self._shard_cache_bytes < self._shard_cache_budget):
+                ds = train_dataloader.dataset
+                nbytes = ds.x.numel() * ds.x.element_size()
+                if ds.y is not None and torch.is_tensor(ds.y):
+                    nbytes += ds.y.numel() * ds.y.element_size()
+                self._shard_cache[user] = train_dataloader
+                self._shard_cache_bytes += nbytes
         client_stats["dataloader"] = time.time() - t_dl
 
         # one flat copy-in instead of the reference's per-tensor clone loop
@@ -213,6 +237,13 @@ class ClientExecutor:
         desired_max_samples = data_config.get("desired_max_samples", None)
         apply_privacy_metrics = bool(privacy_metrics_config
                                      and privacy_metrics_config["apply_metrics"])
+        # FedAvg weights are num_samples (host-known), so the per-client
+        # loss/stats host syncs can be deferred and batched round-level
+        trainer.lazy_stats = (
+            config["strategy"] == "FedAvg" and not apply_privacy_metrics
+            and self.arena.device.type == "cuda"
+            and not getattr(self.client_strategy, "stats_on_smooth_grad", False)
+            and self.server_config.get("type") != "personalization")
 
         client_stats["setup"] = time.time() - begin
         begin_training = time.time()
@@ -261,16 +292,24 @@ class ClientExecutor:
             self.perf_acc[k] = self.perf_acc.get(k, 0.0) + v
         self.perf_acc["clients"] = self.perf_acc.get("clients", 0) + 1
 
-        client_output = {
-            "cs": client_stats,
-            "tl": train_loss,
-            "mg": trainer.sufficient_stats["mag"],
-            "vg": trainer.sufficient_stats["var"],
-            "ng": trainer.sufficient_stats["mean"],
-            "rg": trainer.sufficient_stats["norm"],
-            "ns": num_samples,
-            "pl": payload,
-        }
+        if train_loss is None and trainer.loss_dev is not None:
+            # deferred host sync: the server finalizes these in one batch
+            client_output = {
+                "cs": client_stats, "ns": num_samples, "pl": payload,
+                "_lazy": (trainer.loss_dev, trainer.stats_dev,
+                          trainer.counter),
+            }
+        else:
+            client_output = {
+                "cs": client_stats,
+                "tl": train_loss,
+                "mg": trainer.sufficient_stats["mag"],
+                "vg": trainer.sufficient_stats["var"],
+                "ng": trainer.sufficient_stats["mean"],
+                "rg": trainer.sufficient_stats["norm"],
+                "ns": num_samples,
+                "pl": payload,
+            }
 
         if apply_privacy_metrics:
             self._apply_privacy_metrics(client_output, trainer,

@@ -12,6 +12,7 @@ the reference's model broadcast, command protocol and worker pool
 from __future__ import annotations
 
 import copy
+import math
 import json
 import logging
 import os
@@ -281,6 +282,26 @@ class OptimizationServer:
                 meta["wt"] = payload["weight"] if payload is not None else 0.0
                 local_outputs.append((client_idx, meta))
                 self.run_stats["secsPerClient"][-1].append(time.time() - clients_begin)
+
+            # ---- batched finalize of deferred client stats ---------------
+            # lazy-stats clients carried device tensors; ONE host transfer
+            # materializes every client's loss/Σg/Σg² for this rank
+            lazy = [(idx, meta) for idx, (_, meta) in enumerate(local_outputs)
+                    if "_lazy" in meta]
+            if lazy:
+                flat = torch.stack(
+                    [torch.stack([meta["_lazy"][0].reshape(()),
+                                  meta["_lazy"][1][0], meta["_lazy"][1][1]])
+                     for _, meta in lazy]).cpu().tolist()
+                for (idx, meta), (tl, s, q) in zip(lazy, flat):
+                    n = max(meta["_lazy"][2], 1)
+                    mean = s / n
+                    meta["tl"] = tl
+                    meta["ng"] = mean
+                    meta["mg"] = math.sqrt(max(q / n, 0.0))
+                    meta["vg"] = max(q / n - mean ** 2, 0.0)
+                    meta["rg"] = math.sqrt(max(q, 0.0))
+                    del meta["_lazy"]
 
             # ---- metadata exchange (one all_gather per round) ------------
             gathered = sum(rt.all_gather_object(local_outputs), [])

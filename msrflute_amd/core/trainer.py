@@ -165,6 +165,11 @@ class Trainer(TrainerBase):
             self.lr_scheduler = make_lr_scheduler(self.anneal_config, self.optimizer)
         self.cached_batches = []
         self.ss_scheduler = ss_scheduler
+        # lazy-stats mode (set by ClientExecutor): keep per-epoch loss and
+        # Σg/Σg² on device; the server syncs all clients' stats at once
+        self.lazy_stats = False
+        self.loss_dev = None
+        self.stats_dev = None
         self.reset_gradient_power()
 
     # -- gradient sufficient statistics (K5) ------------------------------
@@ -197,6 +202,21 @@ class Trainer(TrainerBase):
         calls accumulate_gradient_power and the epoch end finalizes once)."""
         self.accumulate_gradient_power()
         return self._finalize_sufficient_stats()
+
+    def finalize_stats_from(self, s, q):
+        """Fill the stats dict from already-synced host scalars (lazy path:
+        the server syncs every client's accumulator in ONE transfer)."""
+        self.sum_grad, self.sum_grad2 = float(s), float(q)
+        n = max(self.counter, 1)
+        mean_grad = self.sum_grad / n
+        mag_grad = math.sqrt(max(self.sum_grad2 / n, 0.0))
+        var_grad = max(self.sum_grad2 / n - mean_grad ** 2, 0.0)
+        self.sufficient_stats = {
+            "n": n, "sum": self.sum_grad, "sq_sum": self.sum_grad2,
+            "var": var_grad, "mean": mean_grad, "mag": mag_grad,
+            "norm": math.sqrt(max(self.sum_grad2, 0.0)),
+        }
+        return self.sufficient_stats
 
     def _finalize_sufficient_stats(self):
         """One host sync: turn the device accumulator into the stats dict."""
@@ -315,6 +335,11 @@ class Trainer(TrainerBase):
 
         if self.lr_scheduler is not None:
             self.lr_scheduler.step()
+        if self.lazy_stats and self.arena is not None \
+                and self.arena.device.type == "cuda":
+            self.loss_dev = loss_acc
+            self.stats_dev = self._stats_acc.clone()
+            return num_samples, None
         self._finalize_sufficient_stats()
         return num_samples, float(loss_acc)
 
@@ -367,6 +392,12 @@ class Trainer(TrainerBase):
         if g is not None:
             self._stats_acc += g.stats_acc
             self.counter += n_graph_batches * self.arena.total
+            if self.lazy_stats:
+                # defer the host sync: the server batches every local
+                # client's (loss, Σg, Σg²) into ONE transfer per round
+                self.loss_dev = g.loss_acc.clone()
+                self.stats_dev = self._stats_acc.clone()
+                return num_samples, None
             loss_total = float(g.loss_acc)
         else:
             loss_total = 0.0
