@@ -86,17 +86,19 @@ class GRU(BaseModel):
         logits = self.forward(x[:, :-1])  # [B, T, V] incl. h0 position
         targets = x.reshape(-1)[mask]
         preds = logits.reshape(-1, self.vocab_size)[mask]
-        return preds, targets, x
+        return preds, targets, x, logits
 
     def loss(self, input) -> Tensor:
-        preds, targets, _ = self._masked_logits(input)
+        preds, targets, _, _ = self._masked_logits(input)
         return nn.functional.cross_entropy(preds, targets)
 
     def inference(self, input):
-        preds, targets, x = self._masked_logits(input)
+        preds, targets, x, logits = self._masked_logits(input)
         top = torch.argmax(preds, dim=1)
         if self.OOV_correct:
             acc = top.eq(targets).float().mean()
         else:
             acc = (top.eq(targets) & (top != 0)).float().mean()
-        return {"output": preds, "acc": acc.item(), "batch_size": x.shape[0]}
+        # 'output' is the full [B, T, V] logit tensor — the privacy leakage
+        # metric indexes it per position (extensions/privacy/metrics.py)
+        return {"output": logits, "acc": acc.item(), "batch_size": x.shape[0]}

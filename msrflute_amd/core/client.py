@@ -132,6 +132,8 @@ class ClientExecutor:
         self._shard_cache_bytes = 0
         self._shard_cache_budget = int(self.client_config.get(
             "shard_cache_gb", 64)) << 30
+        self._shard_store = None       # whole-dataset device pack
+        self._shard_store_tried = False
 
         # hipGraph fast path for the per-batch client step (ops/graphs.py);
         # enabled on GPU for plain-SGD clients unless disabled by config
@@ -147,6 +149,40 @@ class ClientExecutor:
                 self.graph_cache = cache
 
     # ------------------------------------------------------------------
+    def _get_shard_store(self, data_config):
+        """Lazily pack the whole train set into device HBM (one pinned H2D)
+        when the task uses the generic array machinery; None otherwise."""
+        if self._shard_store_tried:
+            return self._shard_store
+        self._shard_store_tried = True
+        import msrflute_amd.core.client as client_mod
+        from ..models import get_exp_dataloader
+        from ..models.generic_data import ArrayDataLoader, DeviceShardStore
+        ds = client_mod.train_dataset
+        try:
+            dl_cls = get_exp_dataloader(self.task) if self.task else None
+        except Exception:
+            dl_cls = None
+        eligible = (
+            ds is not None
+            and getattr(ds, "user_data_label", None) is not None
+            and dl_cls is not None and issubclass(dl_cls, ArrayDataLoader)
+            and not getattr(ds, "_want_transform", False)
+            and self.client_config.get("cache_client_shards", True))
+        if eligible:
+            try:
+                x_shape = tuple(getattr(ds, "x_shape", None) or ())
+                self._shard_store = DeviceShardStore(
+                    ds, x_shape, device=self.arena.device,
+                    budget_bytes=int(self.client_config.get(
+                        "shard_cache_gb", 128)) << 30)
+                print_rank(
+                    f"packed {len(ds.user_list)} client shards into device "
+                    f"memory ({self._shard_store.x.numel() * 4 >> 20} MiB)")
+            except MemoryError as e:
+                print_rank(f"shard store disabled: {e}")
+        return self._shard_store
+
     def _make_optimizer(self, initial_lr):
         if self._fused_opt is not None:
             self._fused_opt.reset_state()
@@ -181,6 +217,13 @@ class ClientExecutor:
 
         t_dl = time.time()
         train_dataloader = self._shard_cache.get(user)
+        if train_dataloader is None and torch.cuda.is_available():
+            store = self._get_shard_store(data_config)
+            if store is not None:
+                train_dataloader = store.loader_for(
+                    user, data_config.get("batch_size", 40))
+                if train_dataloader is not None:
+                    self._shard_cache[user] = train_dataloader
         if train_dataloader is None:
             train_dataloader = make_train_dataloader(
                 data_config, self.data_path, task=self.task, clientx=0,

@@ -41,7 +41,7 @@ def compute_perplexity(encoded_batch, model):
     out = outputs["output"]
     batch_size, seq_len, vocab_size = out.shape
     logp = torch.nn.functional.log_softmax(out, dim=-1)
-    flat_idx = encoded_batch.reshape(-1).cpu()
+    flat_idx = encoded_batch.reshape(-1).cpu().clamp(min=0)  # pads < 0 -> 0
     return logp.reshape(-1, vocab_size)[
         np.arange(batch_size * seq_len), flat_idx].reshape(batch_size, seq_len)
 

@@ -142,3 +142,74 @@ class ArrayDataLoader(BaseDataLoader):
             if self.dataset.y is not None:
                 batch["y"] = self.dataset.y[idx]
             yield batch
+
+
+class _PackedShardDataset:
+    """View-dataset over one packed device tensor (a user's slice)."""
+
+    __slots__ = ("x", "y")
+
+    def __init__(self, x, y):
+        self.x = x
+        self.y = y
+
+    def __len__(self):
+        return len(self.x)
+
+
+class PackedShardLoader(ArrayDataLoader):
+    """ArrayDataLoader over pre-packed device views (no per-client dataset
+    construction, no host->device traffic)."""
+
+    def __init__(self, x, y, batch_size):
+        self.mode = "train"
+        self.args = {}
+        self.dataset = _PackedShardDataset(x, y)
+        self.batch_size = max(1, int(batch_size))
+        self.shuffle = True
+
+    def to_device(self):
+        return self
+
+
+class DeviceShardStore:
+    """Entire federated train set packed into device HBM once
+    (SURVEY.md §7.1 divergence 3): per-user slices are views into two
+    contiguous tensors, staged through ONE pinned-host copy.
+
+    288 GB per MI355X holds every reference dataset whole; ``budget_bytes``
+    guards pathological cases (falls back to per-client loading)."""
+
+    def __init__(self, dataset, x_shape, device="cuda",
+                 budget_bytes=128 << 30):
+        import numpy as np
+        users = list(dataset.user_list)
+        xs, ys, offsets = [], [], [0]
+        for u in users:
+            ud = dataset.user_data[u]
+            x = np.asarray(ud["x"] if isinstance(ud, dict) else ud,
+                           dtype=np.float32)
+            y = np.asarray(dataset.user_data_label[u])
+            xs.append(x.reshape(len(x), -1))
+            ys.append(y)
+            offsets.append(offsets[-1] + len(x))
+        flat_x = np.concatenate(xs)
+        flat_y = np.concatenate(ys)
+        nbytes = flat_x.nbytes + flat_y.nbytes
+        if nbytes > budget_bytes:
+            raise MemoryError(f"shard store would need {nbytes >> 30} GiB")
+        hx = torch.from_numpy(flat_x).pin_memory()
+        hy = torch.from_numpy(flat_y).to(torch.int64).pin_memory()
+        self.x = hx.to(device, non_blocking=True)
+        self.y = hy.to(device, non_blocking=True)
+        if x_shape:
+            self.x = self.x.view(-1, *x_shape)
+        self.offsets = offsets
+        self.user_pos = {u: i for i, u in enumerate(users)}
+
+    def loader_for(self, user, batch_size):
+        i = self.user_pos.get(user)
+        if i is None:
+            return None
+        lo, hi = self.offsets[i], self.offsets[i + 1]
+        return PackedShardLoader(self.x[lo:hi], self.y[lo:hi], batch_size)

@@ -150,7 +150,10 @@ class DGA(FedAvg):
             return
 
         worker_trainer.update_model()
-        losses = worker_trainer.run_lr_scheduler(force_run_val=False)
+        # the RL reward is the val-metric delta vs the RL-weighted model, so
+        # RL mode forces a val pass (documented intent of arXiv:2106.07578;
+        # the reference's RL path is broken as shipped — SURVEY.md §7.5)
+        losses = worker_trainer.run_lr_scheduler(force_run_val=self.want_rl)
 
         if self.want_rl:
             losses = self._run_rl_training(worker_trainer, curr_iter, rl_model,

@@ -1,0 +1,110 @@
+"""Feature-flag e2e coverage: FedProx, DGA+RL, global DP, send_dicts,
+privacy metrics — each toggled by config on tiny synthetic data (mirrors
+reference capabilities listed in SURVEY.md §5.6)."""
+
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+import yaml
+
+from tests.test_tasks import REPO, _make_data
+
+
+def _run_task(task, cfg_dict, tmp_path, data_dir):
+    p = tmp_path / "cfg.yaml"
+    with open(p, "w") as f:
+        yaml.safe_dump(cfg_dict, f)
+    out = str(tmp_path / "out")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, "e2e_trainer.py", "-dataPath", data_dir,
+         "-outputPath", out, "-config", str(p), "-task", task,
+         "-backend", "gloo"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, (r.stdout[-2000:], r.stderr[-3000:])
+    return out
+
+
+def _base_cfg(task="cv_lr_mnist", rounds=3):
+    with open(os.path.join(REPO, "configs", f"{task}.yaml")) as f:
+        cfg = yaml.safe_load(f)
+    sc = cfg["server_config"]
+    sc.update(max_iteration=rounds, num_clients_per_iteration=3,
+              val_freq=1, rec_freq=rounds, initial_val=False,
+              initial_rec=False)
+    return cfg
+
+
+def test_fedprox_strategy(tmp_path):
+    data_dir = str(tmp_path / "data")
+    _make_data("cv_lr_mnist", data_dir)
+    cfg = _base_cfg()
+    cfg["strategy"] = "FedProx"
+    cfg["client_config"]["mu"] = 0.01
+    _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
+
+
+def test_dga_with_rl_reweighting(tmp_path):
+    data_dir = str(tmp_path / "data")
+    _make_data("cv_lr_mnist", data_dir)
+    cfg = _base_cfg(rounds=4)
+    cfg["strategy"] = "DGA"
+    cfg["server_config"].update(
+        wantRL=True, aggregate_median="softmax", softmax_beta=1.0,
+        RL={"RL_path": str(tmp_path / "rl"), "model_descriptor_RL": "t",
+            "network_params": [300, 128, 128, 64, 100],
+            "initial_epsilon": 0.5, "final_epsilon": 0.0001,
+            "epsilon_gamma": 0.90, "max_replay_memory_size": 1000,
+            "minibatch_size": 16, "gamma": 0.99,
+            "optimizer_config": {"type": "adam", "lr": 0.001,
+                                 "amsgrad": True},
+            "annealing_config": {"type": "step_lr", "step_interval": "epoch",
+                                 "gamma": 0.95, "step_size": 1}})
+    out = _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
+    # RL model checkpointed independently (reference RL.py:296-343)
+    assert any("model_RL" in f or "rl" in f.lower()
+               for f in os.listdir(str(tmp_path / "rl")))
+
+
+def test_global_dp(tmp_path):
+    data_dir = str(tmp_path / "data")
+    _make_data("cv_lr_mnist", data_dir)
+    cfg = _base_cfg()
+    cfg["strategy"] = "DGA"
+    # global DP requires client-side clipping => local DP on (reference
+    # privacy/__init__.py:139 asserts the same)
+    cfg["dp_config"] = {
+        "enable_local_dp": True, "enable_global_dp": True,
+        "global_sigma": 0.1, "max_grad": 0.1, "eps": 100.0,
+        "delta": 1e-7, "max_weight": 1.0, "min_weight": 0.0,
+        "weight_scaler": 1.0}
+    _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
+
+
+def test_send_dicts_mode(tmp_path):
+    data_dir = str(tmp_path / "data")
+    _make_data("cv_lr_mnist", data_dir)
+    cfg = _base_cfg()
+    cfg["server_config"]["send_dicts"] = True
+    _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
+
+
+def test_privacy_metrics_indices_and_leakage(tmp_path):
+    data_dir = str(tmp_path / "data")
+    _make_data("nlg_gru", data_dir)
+    cfg = _base_cfg("nlg_gru")
+    cfg["model_config"].update(embed_dim=16, hidden_dim=32)
+    cfg["privacy_metrics_config"] = {
+        "apply_metrics": True, "apply_indices_extraction": True,
+        "allowed_word_rank": 9000,
+        "apply_leakage_metric": True, "max_leakage": 30.0,
+        "max_allowed_leakage": 3.0,
+        "adaptive_leakage_threshold": 0.95,
+        "is_leakage_weighted": True,
+        "attacker_optimizer_config": {"type": "adamax", "lr": 0.03,
+                                      "amsgrad": False}}
+    _run_task("nlg_gru", cfg, tmp_path, data_dir)
