@@ -357,6 +357,22 @@ class Trainer(TrainerBase):
 
         cache.set_lr(get_lr(self.optimizer) if self.optimizer is not None
                      else float(cache.lr_t[0]))
+
+        # whole-epoch fast path: device-resident uniform shard ⇒ the entire
+        # local epoch is ONE graph replay (ops/graphs.py GraphedClientEpoch)
+        ds = getattr(self.train_dataloader, "dataset", None)
+        bs = getattr(self.train_dataloader, "batch_size", 0)
+        if (ds is not None and bs
+                and torch.is_tensor(getattr(ds, "x", None)) and ds.x.is_cuda
+                and torch.is_tensor(getattr(ds, "y", None))
+                and getattr(self.train_dataloader, "shuffle", False)
+                and (desired_max_samples is None
+                     or desired_max_samples >= len(ds.x))):
+            from ..ops.graphs import epoch_graph_for
+            eg = epoch_graph_for(cache, ds.x, ds.y, bs)
+            if eg is not None:
+                return self._run_epoch_one_graph(eg, ds, bs, cache)
+
         train_loader = self.train_dataloader.create_loader()
         for batch in train_loader:
             if desired_max_samples is not None and num_samples >= desired_max_samples:
@@ -403,6 +419,46 @@ class Trainer(TrainerBase):
             loss_total = 0.0
         self._finalize_sufficient_stats()
         return num_samples, loss_total
+
+    def _run_epoch_one_graph(self, eg, ds, bs, cache):
+        """Run one client epoch as a single graph replay + eager ragged
+        tail.  Shuffle order comes from the host torch RNG (per-client seed
+        discipline preserved)."""
+        n = len(ds.x)
+        order = torch.randperm(n)
+        eg.reset_client()
+        eg.run_epoch(ds.x, ds.y, order)
+        n_batches = eg.n_batches
+        tail = order[n_batches * bs:]
+        if len(tail):
+            idx = tail.to(ds.x.device, non_blocking=True)
+            self.arena.grad.zero_()
+            loss = self.model.loss({"x": ds.x.index_select(0, idx),
+                                    "y": ds.y.index_select(0, idx)})
+            loss.backward()
+            ops.clip_stats_accumulate(
+                self.arena.grad,
+                float(self.max_grad_norm) if self.max_grad_norm else -1.0,
+                eg.stats_acc)
+            ops.sgd_step_devlr(self.arena.data, self.arena.grad,
+                               eg.momentum_buf, cache.lr_t,
+                               momentum=cache.momentum, dampening=0.0,
+                               weight_decay=cache.weight_decay,
+                               nesterov=cache.nesterov, first_step=False)
+            eg.loss_acc += loss.detach()
+            n_batches += 1
+        self.step += n_batches
+        if self.lr_scheduler is not None:
+            self.lr_scheduler.step()
+        self._stats_acc += eg.stats_acc
+        self.counter += n_batches * self.arena.total
+        if self.lazy_stats:
+            self.loss_dev = eg.loss_acc.clone()
+            self.stats_dev = self._stats_acc.clone()
+            return n, None
+        loss_total = float(eg.loss_acc)
+        self._finalize_sufficient_stats()
+        return n, loss_total
 
     def run_train_epoch_fedprox(self, desired_max_samples=None,
                                 apply_privacy_metrics=False, algo_payload=None):

@@ -133,3 +133,101 @@ class GraphCache:
                                   tuple(x.shape), tuple(y.shape), y.dtype)
             self._graphs[key] = g
         return g
+
+
+class GraphedClientEpoch:
+    """Whole-epoch capture: every full-size batch of a client's local epoch
+    replays as ONE hipGraph.
+
+    The client's device-resident shard is copied (D2D) into a static
+    buffer, a fresh shuffle order is staged through a pinned host buffer,
+    and the graph gathers each batch with ``index_select`` before the
+    captured fwd/bwd/clip+stats/SGD sequence — so per client the host does
+    two tiny copies and one graph launch, regardless of batch count.
+    Keyed by (n_samples, n_batches, batch shape); the ragged tail (n %
+    batch_size) stays on the per-batch path.
+    """
+
+    def __init__(self, model, arena: ParameterArena, lr_t: torch.Tensor,
+                 max_grad_norm, momentum: float, weight_decay: float,
+                 nesterov: bool, shard_x_shape, y_dtype, n: int, bs: int):
+        self.model = model
+        self.arena = arena
+        self.lr_t = lr_t
+        self.momentum = momentum
+        self.n = n
+        self.bs = bs
+        self.n_batches = n // bs
+        dev = arena.device
+        self.static_x = torch.zeros((n, *shard_x_shape), device=dev)
+        self.static_y = torch.zeros((n,), dtype=y_dtype, device=dev)
+        self.static_idx = torch.zeros((n,), dtype=torch.int64, device=dev)
+        self.host_idx = torch.zeros((n,), dtype=torch.int64).pin_memory()
+        self.stats_acc = torch.zeros(2, device=dev)
+        self.loss_acc = torch.zeros((), device=dev)
+        self.momentum_buf = arena.new_buffer() if momentum != 0.0 else None
+        mn = float(max_grad_norm) if max_grad_norm is not None else -1.0
+
+        def epoch_body():
+            for b in range(self.n_batches):
+                idx = self.static_idx[b * bs:(b + 1) * bs]
+                x = self.static_x.index_select(0, idx)
+                y = self.static_y.index_select(0, idx)
+                self.arena.grad.zero_()
+                loss = self.model.loss({"x": x, "y": y})
+                loss.backward()
+                ops.clip_stats_accumulate(self.arena.grad, mn, self.stats_acc)
+                ops.sgd_step_devlr(self.arena.data, self.arena.grad,
+                                   self.momentum_buf, self.lr_t,
+                                   momentum=momentum, dampening=0.0,
+                                   weight_decay=weight_decay,
+                                   nesterov=nesterov, first_step=False)
+                self.loss_acc += loss.detach()
+
+        saved = arena.data.clone()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                epoch_body()
+        torch.cuda.current_stream().wait_stream(s)
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            epoch_body()
+
+        arena.data.copy_(saved)
+        arena.grad.zero_()
+        self.reset_client()
+        torch.cuda.synchronize()
+
+    def reset_client(self):
+        self.stats_acc.zero_()
+        self.loss_acc.zero_()
+        if self.momentum_buf is not None:
+            self.momentum_buf.zero_()
+
+    def run_epoch(self, x_shard, y_shard, order_cpu):
+        """order_cpu: CPU int64 tensor of length n (fresh shuffle)."""
+        self.static_x.copy_(x_shard, non_blocking=True)
+        self.static_y.copy_(y_shard, non_blocking=True)
+        self.host_idx.copy_(order_cpu)
+        self.static_idx.copy_(self.host_idx, non_blocking=True)
+        self.graph.replay()
+
+
+def epoch_graph_for(cache: "GraphCache", x_shard, y_shard, bs: int):
+    """Fetch/capture the whole-epoch graph for this shard geometry."""
+    n = x_shard.shape[0]
+    if n < bs:  # single ragged batch — per-batch path handles it
+        return None
+    key = ("epoch", n, bs, tuple(x_shard.shape[1:]), y_shard.dtype)
+    g = cache._graphs.get(key)
+    if g is None:
+        g = GraphedClientEpoch(cache.model, cache.arena, cache.lr_t,
+                               cache.max_grad_norm, cache.momentum,
+                               cache.weight_decay, cache.nesterov,
+                               tuple(x_shard.shape[1:]), y_shard.dtype,
+                               n, bs)
+        cache._graphs[key] = g
+    return g

@@ -99,3 +99,46 @@ def test_graph_replay_uses_updated_lr():
     g.run_batch(x, y)
     torch.cuda.synchronize()
     assert not torch.allclose(a.data, w0)
+
+
+def test_whole_epoch_graph_matches_eager():
+    """GraphedClientEpoch (one replay per epoch) vs the eager sequence on
+    the same shuffle order, incl. a ragged tail (n % bs != 0)."""
+    from msrflute_amd.ops.arena import ParameterArena
+    from msrflute_amd.ops.graphs import GraphCache, epoch_graph_for
+    from msrflute_amd import ops
+
+    torch.manual_seed(11)
+    m1 = TinyModel().cuda()
+    m2 = TinyModel().cuda()
+    m2.load_state_dict(m1.state_dict())
+    a1, a2 = ParameterArena(m1), ParameterArena(m2)
+    lr, max_norm, bs = 0.05, 1.0, 16
+    n = 16 * 4 + 7  # ragged tail of 7
+    xs = torch.randn(n, 32).cuda()
+    ys = torch.randint(0, 5, (n,)).cuda()
+    order = torch.randperm(n)
+
+    # eager reference over the same order (full batches only)
+    full = order[: (n // bs) * bs]
+    batches = [(xs[full[i:i + bs].cuda()], ys[full[i:i + bs].cuda()])
+               for i in range(0, len(full), bs)]
+    s1, l1 = _run_eager(m1, a1, batches, lr, max_norm)
+
+    cache = GraphCache(m2, a2, {"type": "sgd", "lr": lr}, max_norm)
+    cache.set_lr(lr)
+    eg = epoch_graph_for(cache, xs, ys, bs)
+    assert eg is not None and eg.n_batches == 4
+    eg.reset_client()
+    eg.run_epoch(xs, ys, order)
+    torch.cuda.synchronize()
+
+    assert torch.allclose(a1.data, a2.data, rtol=1e-5, atol=1e-6)
+    assert torch.allclose(s1, eg.stats_acc, rtol=1e-4, atol=1e-5)
+    assert torch.allclose(l1, eg.loss_acc, rtol=1e-5, atol=1e-6)
+
+    # second client replays with a different order and stays finite
+    eg.reset_client()
+    eg.run_epoch(xs, ys, torch.randperm(n))
+    torch.cuda.synchronize()
+    assert torch.isfinite(eg.loss_acc).item()
