@@ -206,6 +206,16 @@ class OptimizationServer:
         rt = self.runtime
         is_chief = rt.rank == 0
         eval_list = []
+        profiler = None
+        if self.do_profiling:
+            # reference wraps rounds in cProfile (server.py:327-331); on a
+            # GPU additionally emit a roctx range so rocprofv3 runtime
+            # traces show round boundaries
+            import cProfile
+            profiler = cProfile.Profile()
+            profiler.enable()
+            if torch.cuda.is_available():
+                torch.cuda.nvtx.range_push(f"fl_round_{i}")
         if True:
             begin = time.time()
             metrics_payload = {}
@@ -455,6 +465,18 @@ class OptimizationServer:
             if is_chief:
                 for k, v in metrics_payload.items():
                     log_metric(k, v, step=i)
+
+        if profiler is not None:
+            if torch.cuda.is_available():
+                torch.cuda.nvtx.range_pop()
+            profiler.disable()
+            import io
+            import pstats
+            buf = io.StringIO()
+            pstats.Stats(profiler, stream=buf).sort_stats(
+                "cumulative").print_stats(20)
+            print_rank(f"round {i} profile:\n{buf.getvalue()}",
+                       loglevel=logging.DEBUG)
 
     # ------------------------------------------------------------------
     def backup_models(self, i):
