@@ -466,3 +466,75 @@ def convex_inference(model_global, model_personal, alpha):
         (1 - alpha) * model_global["probabilities"]
     preds = torch.argmax(torch.tensor(np.asarray(probs)), dim=1)
     return torch.mean((preds == targets).float()).item()
+
+
+class ClientPool:
+    """P client executors on P HIP streams: clients within a round train
+    CONCURRENTLY on the GPU (the round is a weighted sum — order-free).
+
+    Each executor owns a model replica, arena, graph cache and a round
+    accumulator; its stream serializes its own clients.  Weighted
+    pseudo-gradients accumulate stream-locally; ``flush()`` joins the
+    streams and adds the P accumulators into the server grad arena.  This
+    is the MI355X-native replacement for the reference's process-per-GPU
+    worker pool: concurrency lives on streams inside one
+    rank instead of extra processes.
+    """
+
+    def __init__(self, config, task, data_path, server_arena, model_path=None,
+                 n_parallel=4):
+        self.server_arena = server_arena
+        self.executors = [ClientExecutor(config, task, data_path,
+                                         server_arena, model_path)
+                          for _ in range(max(1, int(n_parallel)))]
+        self.streams = [torch.cuda.Stream() for _ in self.executors]
+        self.round_accums = [server_arena.new_buffer() for _ in self.executors]
+        self._rr = 0
+        self._streams_dirty = False
+
+    # properties the server reads off the executor
+    @property
+    def client_strategy(self):
+        return self.executors[0].client_strategy
+
+    @property
+    def perf_acc(self):
+        merged = {}
+        for ex in self.executors:
+            for k, v in ex.perf_acc.items():
+                merged[k] = merged.get(k, 0) + v
+        return merged
+
+    def process_round(self, client, initial_lr, iteration, round_seed=0):
+        k = self._rr % len(self.executors)
+        self._rr += 1
+        ex, st = self.executors[k], self.streams[k]
+        # share the packed shard store across replicas (one copy in HBM)
+        prim = self.executors[0]
+        if ex is not prim and prim._shard_store_tried and not ex._shard_store_tried:
+            ex._shard_store = prim._shard_store
+            ex._shard_store_tried = True
+        st.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(st):
+            out = ex.process_round(client, initial_lr, iteration, round_seed)
+            payload = out.get("pl")
+            if payload is not None and payload.get("grad") is not None \
+                    and payload["weight"] != 0.0:
+                ops.axpy(self.round_accums[k], payload["grad"], 1.0)
+                payload["grad"] = None
+                payload["pooled"] = True
+        self._streams_dirty = True
+        return out
+
+    def flush(self, target_grad):
+        """Join client streams and fold the P accumulators into the server
+        grad arena.  Must run before the round's lazy-stats finalize."""
+        if not self._streams_dirty:
+            return
+        cur = torch.cuda.current_stream()
+        for st in self.streams:
+            cur.wait_stream(st)
+        for acc in self.round_accums:
+            ops.axpy(target_grad, acc, 1.0)
+            acc.zero_()
+        self._streams_dirty = False

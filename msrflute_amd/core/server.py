@@ -130,10 +130,24 @@ class OptimizationServer:
         self.strategy = StrategyClass("server", config, model_path,
                                       runtime=runtime)
 
-        # per-rank persistent client workspace
-        self.executor = ClientExecutor(config, self.task, data_path,
+        # per-rank persistent client workspace; with a GPU and an
+        # order-free strategy, a pool of stream-parallel executors trains
+        # several clients concurrently (ClientPool)
+        n_par = int(config["client_config"].get("parallel_clients", 4))
+        if (torch.cuda.is_available() and n_par > 1
+                and config["strategy"] in ("FedAvg", "FedProx")
+                and server_config.get("type") != "personalization"
+                and not config.get("privacy_metrics_config", {}).get(
+                    "apply_metrics", False)):
+            from .client import ClientPool
+            self.executor = ClientPool(config, self.task, data_path,
                                        server_arena=arena,
-                                       model_path=model_path)
+                                       model_path=model_path,
+                                       n_parallel=n_par)
+        else:
+            self.executor = ClientExecutor(config, self.task, data_path,
+                                           server_arena=arena,
+                                           model_path=model_path)
 
         # per-client sample counts for size-aware partitioning
         from . import client as client_mod
@@ -292,6 +306,11 @@ class OptimizationServer:
                 meta["wt"] = payload["weight"] if payload is not None else 0.0
                 local_outputs.append((client_idx, meta))
                 self.run_stats["secsPerClient"][-1].append(time.time() - clients_begin)
+
+            # join client streams + fold pool accumulators into the server
+            # grad arena (no-op for the single-executor path)
+            if hasattr(self.executor, "flush"):
+                self.executor.flush(self.worker_trainer.arena.grad)
 
             # ---- batched finalize of deferred client stats ---------------
             # lazy-stats clients carried device tensors; ONE host transfer

@@ -68,6 +68,9 @@ class FedAvg(BaseStrategy):
         if payload["weight"] == 0.0:
             return False
         self.client_weights.append(payload["weight"])
+        if payload.get("pooled"):
+            # gradient already accumulated stream-locally by ClientPool
+            return True
         if self.aggregate_fast:
             accumulate_flat_grad(worker_trainer, payload["grad"])
         else:
