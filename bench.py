@@ -64,6 +64,7 @@ def build_config(args):
             "seed": 1234,
         },
         "client_config": {
+            "parallel_clients": int(os.environ.get("BENCH_PAR", "4")),
             "do_profiling": False,
             "ignore_subtask": False,
             "data_config": {
