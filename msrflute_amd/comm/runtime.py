@@ -91,6 +91,27 @@ class FedRuntime:
         dist.all_gather_object(out, obj)
         return out
 
+    def all_gather_rows(self, t: torch.Tensor,
+                        counts: Sequence[int]) -> List[torch.Tensor]:
+        """Gather variable-row 2-D float tensors without object pickling.
+        ``counts[r]`` (known identically on every rank — partitions are
+        deterministic) gives rank r's row count; rows are padded to the max
+        and truncated after one fused all_gather.  On NCCL this keeps the
+        per-round metadata exchange on the xGMI fabric as one small
+        collective instead of a pickled object broadcast chain."""
+        if self.size == 1:
+            return [t]
+        k = t.shape[1] if t.dim() == 2 else 0
+        m = max(int(c) for c in counts)
+        dev = ("cuda" if self.backend == "nccl" and torch.cuda.is_available()
+               else "cpu")
+        pad = torch.zeros((m, k), dtype=torch.float64, device=dev)
+        if t.numel():
+            pad[: t.shape[0]] = t.to(dev, torch.float64)
+        out = [torch.empty_like(pad) for _ in range(self.size)]
+        dist.all_gather(out, pad)
+        return [o[: int(c)].cpu() for o, c in zip(out, counts)]
+
     def barrier(self):
         if self.size > 1:
             if self.backend == "nccl" and torch.cuda.is_available():

@@ -270,7 +270,8 @@ class OptimizationServer:
                                                     num_clients_curr_iter, i)
             weights = ([self.client_num_samples_all[c] for c in sampled_idx_clients]
                        if self.client_num_samples_all is not None else None)
-            my_clients = rt.partition(sampled_idx_clients, weights)[rt.rank]
+            parts = rt.partition(sampled_idx_clients, weights)
+            my_clients = parts[rt.rank]
 
             clients_begin = time.time()
             apply_privacy_metrics = bool(
@@ -333,7 +334,32 @@ class OptimizationServer:
                     del meta["_lazy"]
 
             # ---- metadata exchange (one all_gather per round) ------------
-            gathered = sum(rt.all_gather_object(local_outputs), [])
+            # numeric metas travel as ONE fused tensor all_gather (no
+            # pickle on the fabric); privacy-metric dicts fall back to the
+            # object path
+            if rt.size > 1 and not apply_privacy_metrics:
+                cols = ["tl", "mg", "ng", "vg", "rg", "ns", "wt", "ts"]
+                rows = [[float(cid), 1.0 if meta["accepted"] else 0.0]
+                        + [float(meta.get(c, 0.0)) for c in cols]
+                        for cid, meta in local_outputs]
+                local_t = torch.tensor(rows, dtype=torch.float64).reshape(
+                    len(rows), 2 + len(cols))
+                per_rank = rt.all_gather_rows(local_t,
+                                              [len(p) for p in parts])
+                gathered = []
+                for r, block in enumerate(per_rank):
+                    if r == rt.rank:
+                        gathered.extend(local_outputs)  # keep local cs dicts
+                        continue
+                    for row in block.tolist():
+                        meta = {"accepted": bool(row[1]),
+                                **{c: row[2 + j] for j, c in enumerate(cols)},
+                                "ns": int(row[7]),
+                                "cs": {"full cost": 0.0, "training": 0.0,
+                                       "setup": 0.0}}
+                        gathered.append((int(row[0]), meta))
+            else:
+                gathered = sum(rt.all_gather_object(local_outputs), [])
             order = {c: k for k, c in enumerate(sampled_idx_clients)}
             gathered.sort(key=lambda t: order.get(t[0], 1 << 30))
 
