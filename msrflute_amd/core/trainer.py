@@ -173,8 +173,14 @@ class Trainer(TrainerBase):
         self.reset_gradient_power()
 
     # -- gradient sufficient statistics (K5) ------------------------------
+    def _device(self):
+        if self.arena is not None:
+            return self.arena.device
+        p = next(self.model.parameters(), None)
+        return p.device if p is not None else torch.device("cpu")
+
     def reset_gradient_power(self):
-        dev = self.arena.device if self.arena is not None else "cpu"
+        dev = self._device()
         self._stats_acc = torch.zeros(2, dtype=torch.float32, device=dev)
         self.counter = 0
         self.sum_grad = 0.0
@@ -294,7 +300,7 @@ class Trainer(TrainerBase):
         self.zero_grad()
         # loss accumulates in a device scalar: ONE host sync per epoch
         # instead of the reference's per-batch loss.item()
-        dev = self.arena.device if self.arena is not None else "cpu"
+        dev = self._device()
         loss_acc = torch.zeros((), device=dev)
 
         train_loader = self.train_dataloader.create_loader()
