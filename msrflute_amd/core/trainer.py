@@ -361,6 +361,14 @@ class Trainer(TrainerBase):
         g = None
         n_graph_batches = 0
 
+        # probe one batch: graphs need dict batches of plain (x, y) tensors
+        # (nlg/mlm/newsrec batch shapes run the generic eager path)
+        probe = next(iter(self.train_dataloader.create_loader()), None)
+        if not (isinstance(probe, dict) and torch.is_tensor(probe.get("x"))
+                and torch.is_tensor(probe.get("y"))):
+            self.graph_cache = None
+            return self.run_train_epoch(desired_max_samples)
+
         cache.set_lr(get_lr(self.optimizer) if self.optimizer is not None
                      else float(cache.lr_t[0]))
 
@@ -386,6 +394,9 @@ class Trainer(TrainerBase):
             x, y = batch["x"], batch["y"]
             if g is None:
                 g = cache.get(x, y)
+                if g is None:  # capture failed — this model runs eager
+                    self.graph_cache = None
+                    return self.run_train_epoch(desired_max_samples)
                 g.reset_client()
             if tuple(x.shape) == tuple(g.static_x.shape):
                 g.run_batch(x, y)

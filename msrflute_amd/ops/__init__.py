@@ -172,7 +172,13 @@ def quantize_dequantize(x, n_bins: int, threshold_quantile: float):
         # Stats via torch's device reductions (stay on device), fused
         # bin+mask via the HIP kernel — no host synchronization.
         min_t, max_t = torch.aminmax(x)
-        thresh_t = torch.quantile(x.abs(), threshold_quantile)
+        # torch.quantile caps input size (~2^24); for bigger arenas the
+        # sparsification threshold comes from a strided 4M-element sample
+        # (the threshold is a heuristic sparsifier — reference quant.py:53)
+        a = x.abs().reshape(-1)
+        if a.numel() > (1 << 24):
+            a = a[:: (a.numel() + (1 << 22) - 1) >> 22]
+        thresh_t = torch.quantile(a, threshold_quantile)
         _C.quant_bin_mask(x, min_t.reshape(1).contiguous(),
                           max_t.reshape(1).contiguous(),
                           thresh_t.reshape(1).float().contiguous(), int(n_bins))

@@ -39,13 +39,13 @@ class GraphedClientStep:
                  max_grad_norm: Optional[float], momentum: float,
                  weight_decay: float, nesterov: bool,
                  x_shape: Tuple[int, ...], y_shape: Tuple[int, ...],
-                 y_dtype: torch.dtype):
+                 y_dtype: torch.dtype, x_dtype: torch.dtype = torch.float32):
         self.model = model
         self.arena = arena
         self.lr_t = lr_t
         self.momentum = momentum
         dev = arena.device
-        self.static_x = torch.zeros(x_shape, device=dev)
+        self.static_x = torch.zeros(x_shape, dtype=x_dtype, device=dev)
         self.static_y = torch.zeros(y_shape, dtype=y_dtype, device=dev)
         self.stats_acc = torch.zeros(2, device=dev)
         self.loss_acc = torch.zeros((), device=dev)
@@ -123,15 +123,22 @@ class GraphCache:
     def set_lr(self, lr: float):
         self.lr_t.fill_(float(lr))
 
-    def get(self, x: torch.Tensor, y: torch.Tensor) -> GraphedClientStep:
-        key = (tuple(x.shape), tuple(y.shape), y.dtype)
-        g = self._graphs.get(key)
-        if g is None:
+    def get(self, x: torch.Tensor, y: torch.Tensor):
+        key = (tuple(x.shape), x.dtype, tuple(y.shape), y.dtype)
+        if key in self._graphs:
+            return self._graphs[key]
+        try:
             g = GraphedClientStep(self.model, self.arena, self.lr_t,
                                   self.max_grad_norm, self.momentum,
                                   self.weight_decay, self.nesterov,
-                                  tuple(x.shape), tuple(y.shape), y.dtype)
-            self._graphs[key] = g
+                                  tuple(x.shape), tuple(y.shape), y.dtype,
+                                  x_dtype=x.dtype)
+        except Exception as e:  # capture-unsafe model op — run eager
+            import logging
+            logging.getLogger().warning(
+                f"hipGraph capture failed ({e}); falling back to eager")
+            g = None
+        self._graphs[key] = g
         return g
 
 
@@ -159,8 +166,9 @@ class GraphedClientEpoch:
         self.bs = bs
         self.n_batches = n // bs
         dev = arena.device
-        self.static_x = torch.zeros((n, *shard_x_shape), device=dev)
-        self.static_y = torch.zeros((n,), dtype=y_dtype, device=dev)
+        x_shape, x_dtype, y_shape = shard_x_shape
+        self.static_x = torch.zeros((n, *x_shape), dtype=x_dtype, device=dev)
+        self.static_y = torch.zeros((n, *y_shape), dtype=y_dtype, device=dev)
         self.static_idx = torch.zeros((n,), dtype=torch.int64, device=dev)
         self.host_idx = torch.zeros((n,), dtype=torch.int64).pin_memory()
         self.stats_acc = torch.zeros(2, device=dev)
@@ -221,13 +229,20 @@ def epoch_graph_for(cache: "GraphCache", x_shard, y_shard, bs: int):
     n = x_shard.shape[0]
     if n < bs:  # single ragged batch — per-batch path handles it
         return None
-    key = ("epoch", n, bs, tuple(x_shard.shape[1:]), y_shard.dtype)
+    key = ("epoch", n, bs, tuple(x_shard.shape[1:]), x_shard.dtype,
+           tuple(y_shard.shape[1:]), y_shard.dtype)
     g = cache._graphs.get(key)
     if g is None:
-        g = GraphedClientEpoch(cache.model, cache.arena, cache.lr_t,
-                               cache.max_grad_norm, cache.momentum,
-                               cache.weight_decay, cache.nesterov,
-                               tuple(x_shard.shape[1:]), y_shard.dtype,
-                               n, bs)
+        try:
+            g = GraphedClientEpoch(
+                cache.model, cache.arena, cache.lr_t, cache.max_grad_norm,
+                cache.momentum, cache.weight_decay, cache.nesterov,
+                (tuple(x_shard.shape[1:]), x_shard.dtype,
+                 tuple(y_shard.shape[1:])), y_shard.dtype, n, bs)
+        except Exception as e:  # capture-unsafe model op — run eager
+            import logging
+            logging.getLogger().warning(
+                f"epoch-graph capture failed ({e}); falling back to eager")
+            g = None
         cache._graphs[key] = g
     return g
