@@ -137,8 +137,13 @@ class ClientExecutor:
 
         # hipGraph fast path for the per-batch client step (ops/graphs.py);
         # enabled on GPU for plain-SGD clients unless disabled by config
+        # MIOpen RNN kernels segfault under hipGraph capture (hipblaslt
+        # assert -> SIGSEGV, observed with the fedshakespeare LSTM), so
+        # models containing RNN modules always run the eager path.
+        has_rnn = any(isinstance(m, torch.nn.RNNBase)
+                      for m in self.model.modules())
         self.graph_cache = None
-        if (torch.cuda.is_available()
+        if (torch.cuda.is_available() and not has_rnn
                 and self.client_config.get("use_hip_graphs", True)):
             from ..ops.graphs import GraphCache
             cache = GraphCache(

@@ -14,6 +14,8 @@ void launch_pseudo_grad(float*, const float*, const float*, float, long long,
 void launch_axpy(float*, const float*, float, long long, hipStream_t);
 void launch_scale(float*, float, long long, hipStream_t);
 void launch_sum_sumsq(const float*, long long, double*, hipStream_t);
+void launch_sum_sumsq2(const float*, long long, double*, double*, hipStream_t);
+int sum_sumsq2_partials(long long);
 void launch_clip_apply(float*, long long, const double*, float, float, float*,
                        hipStream_t);
 void launch_add_gaussian_noise(float*, long long, float, unsigned long long,
@@ -68,20 +70,27 @@ void scale(torch::Tensor x, double alpha) {
   launch_scale(x.data_ptr<float>(), (float)alpha, x.numel(), cur_stream());
 }
 
+// two-stage reduction helper: no f64 atomics on the hot path
+static torch::Tensor sum_sumsq_acc(const torch::Tensor& x) {
+  auto acc = torch::zeros({2}, x.options().dtype(torch::kFloat64));
+  int nparts = sum_sumsq2_partials(x.numel());
+  auto partials = torch::empty({2 * (long long)nparts},
+                               x.options().dtype(torch::kFloat64));
+  launch_sum_sumsq2(x.data_ptr<float>(), x.numel(),
+                    partials.data_ptr<double>(), acc.data_ptr<double>(),
+                    cur_stream());
+  return acc;
+}
+
 torch::Tensor sum_sumsq(torch::Tensor x) {
   check_flat(x, "x");
-  auto acc = torch::zeros({2}, x.options().dtype(torch::kFloat64));
-  launch_sum_sumsq(x.data_ptr<float>(), x.numel(), acc.data_ptr<double>(),
-                   cur_stream());
-  return acc.to(torch::kFloat32);
+  return sum_sumsq_acc(x).to(torch::kFloat32);
 }
 
 torch::Tensor clip_by_norm(torch::Tensor x, double max_norm, double eps) {
   check_flat(x, "x");
-  auto acc = torch::zeros({2}, x.options().dtype(torch::kFloat64));
+  auto acc = sum_sumsq_acc(x);
   auto norm = torch::empty({}, x.options());
-  launch_sum_sumsq(x.data_ptr<float>(), x.numel(), acc.data_ptr<double>(),
-                   cur_stream());
   launch_clip_apply(x.data_ptr<float>(), x.numel(), acc.data_ptr<double>(),
                     (float)max_norm, (float)eps, norm.data_ptr<float>(),
                     cur_stream());
@@ -133,9 +142,7 @@ void clip_stats_accumulate(torch::Tensor x, double max_norm, double eps,
                            torch::Tensor stats_acc) {
   check_flat(x, "x"); check_flat(stats_acc, "stats_acc");
   TORCH_CHECK(stats_acc.numel() >= 2, "stats_acc must have 2 elements");
-  auto acc = torch::zeros({2}, x.options().dtype(torch::kFloat64));
-  launch_sum_sumsq(x.data_ptr<float>(), x.numel(), acc.data_ptr<double>(),
-                   cur_stream());
+  auto acc = sum_sumsq_acc(x);
   launch_clip_apply_stats(x.data_ptr<float>(), x.numel(),
                           acc.data_ptr<double>(), (float)max_norm, (float)eps,
                           stats_acc.data_ptr<float>(), cur_stream());

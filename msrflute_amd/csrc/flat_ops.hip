@@ -147,6 +147,53 @@ __global__ void k_sum_sumsq(const float* __restrict__ x, long long n,
   block_reduce2_atomic(s, q, out2);
 }
 
+// Two-stage variant: stage 1 writes one {Σ, Σ²} pair per block (no
+// atomics — the serialized f64 atomic chain was the bottleneck at large
+// grids); stage 2 folds the partials with one small block.
+__global__ void k_sum_sumsq_partial(const float* __restrict__ x, long long n,
+                                    double* __restrict__ partials) {
+  double s = 0.0, q = 0.0;
+  long long n4 = n >> 2;
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n4;
+       i += (long long)gridDim.x * blockDim.x) {
+    float4 a = x4[i];
+    s += (double)a.x + (double)a.y + (double)a.z + (double)a.w;
+    q += (double)a.x * a.x + (double)a.y * a.y +
+         (double)a.z * a.z + (double)a.w * a.w;
+  }
+  long long tail = n4 << 2;
+  for (long long i = tail + blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < n; i += (long long)gridDim.x * blockDim.x) {
+    double v = x[i];
+    s += v; q += v * v;
+  }
+  for (int d = WAVE / 2; d > 0; d >>= 1) {
+    s += __shfl_down(s, d, WAVE);
+    q += __shfl_down(q, d, WAVE);
+  }
+  __shared__ double lds_s[BLOCK / WAVE], lds_q[BLOCK / WAVE];
+  int wave = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+  if (lane == 0) { lds_s[wave] = s; lds_q[wave] = q; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double ts = 0, tq = 0;
+    for (int w = 0; w < BLOCK / WAVE; ++w) { ts += lds_s[w]; tq += lds_q[w]; }
+    partials[2 * blockIdx.x] = ts;
+    partials[2 * blockIdx.x + 1] = tq;
+  }
+}
+
+__global__ void k_fold_partials(const double* __restrict__ partials,
+                                int n_blocks, double* __restrict__ out2) {
+  double s = 0.0, q = 0.0;
+  for (int i = threadIdx.x; i < n_blocks; i += blockDim.x) {
+    s += partials[2 * i];
+    q += partials[2 * i + 1];
+  }
+  block_reduce2_atomic(s, q, out2);
+}
+
 // Scale x by min(1, max_norm/(sqrt(sumsq)+eps)); out2[1] holds Σx² from
 // k_sum_sumsq; norm_out receives the pre-clip norm. One extra kernel, zero
 // host synchronization (torch clip_grad_norm_ semantics).
@@ -401,6 +448,18 @@ void launch_sum_sumsq(const float* x, long long n, double* out2,
   hipLaunchKernelGGL(k_sum_sumsq, dim3(grid_for_reduce(n >> 2)), dim3(BLOCK),
                      0, s, x, n, out2);
 }
+
+// two-stage: partials buffer must hold 2*grid doubles; out2 zero-init
+void launch_sum_sumsq2(const float* x, long long n, double* partials,
+                       double* out2, hipStream_t s) {
+  int grid = grid_for(n >> 2);
+  hipLaunchKernelGGL(k_sum_sumsq_partial, dim3(grid), dim3(BLOCK), 0, s,
+                     x, n, partials);
+  hipLaunchKernelGGL(k_fold_partials, dim3(1), dim3(BLOCK), 0, s,
+                     partials, grid, out2);
+}
+
+int sum_sumsq2_partials(long long n) { return grid_for(n >> 2); }
 
 void launch_clip_apply(float* x, long long n, const double* out2,
                        float max_norm, float eps, float* norm_out,
