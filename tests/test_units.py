@@ -1,0 +1,180 @@
+"""Unit tests for parity-critical math: custom optimizers, LR schedulers,
+the RDP accountant, DGA weighting, samplers, quantization semantics."""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from msrflute_amd.extensions.privacy.analysis import (compute_rdp,
+                                                      get_privacy_spent)
+from msrflute_amd.ops import reference as ref
+from msrflute_amd.strategies.utils import filter_weight
+from msrflute_amd.utils.data_utils import DynamicBatchSampler
+from msrflute_amd.utils.misc import softmax_weights
+from msrflute_amd.utils.optimizers import LAMB, LarsSGD, make_optimizer
+from msrflute_amd.utils.schedulers import (RampupKeepExpdecayKeepLRScheduler,
+                                           make_lr_scheduler)
+
+
+# ---- optimizers -------------------------------------------------------
+
+
+def _quad_problem(opt_ctor, steps=60):
+    torch.manual_seed(0)
+    w = torch.nn.Parameter(torch.randn(8, 4))
+    target = torch.randn(8, 4)
+    opt = opt_ctor([w])
+    initial = ((w - target) ** 2).sum().item()
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = ((w - target) ** 2).sum()
+        loss.backward()
+        opt.step()
+    return ((w - target) ** 2).sum().item(), initial
+
+
+def test_lamb_decreases_loss():
+    final, initial = _quad_problem(lambda p: LAMB(p, lr=0.1))
+    assert final < 0.2 * initial
+
+
+def test_lars_decreases_loss():
+    # LARS effective step = lr * trust_coef * ||w||/||g|| — tiny by design
+    # (built for large-batch regimes), so a large base lr drives the test
+    final, initial = _quad_problem(
+        lambda p: LarsSGD(p, lr=20.0, momentum=0.9), steps=200)
+    assert final < 0.5 * initial
+
+
+def test_lamb_trust_ratio_scales_update():
+    """LAMB scales each layer's step by ‖w‖/‖update‖ — a layer with huge
+    weights must take a proportionally larger absolute step."""
+    small = torch.nn.Parameter(torch.full((10,), 0.01))
+    big = torch.nn.Parameter(torch.full((10,), 100.0))
+    opt = LAMB([small, big], lr=0.01)
+    for p in (small, big):
+        p.grad = torch.full_like(p, 0.5)
+    s0, b0 = small.detach().clone(), big.detach().clone()
+    opt.step()
+    assert (big - b0).abs().mean() > 100 * (small - s0).abs().mean()
+
+
+def test_make_optimizer_all_types():
+    for t in ["sgd", "adam", "adamax", "lars", "LarsSGD", "lamb", "adamW"]:
+        m = torch.nn.Linear(4, 2)
+        opt = make_optimizer({"type": t, "lr": 0.01}, m)
+        m(torch.randn(3, 4)).sum().backward()
+        opt.step()
+
+
+# ---- LR schedulers ----------------------------------------------------
+
+
+def test_rampup_keep_expdecay_keep_trajectory():
+    opt = torch.optim.SGD([torch.nn.Parameter(torch.zeros(1))], lr=1.0)
+    s = RampupKeepExpdecayKeepLRScheduler(opt, peak_lr=0.1, floor_lr=0.001,
+                                          sr=10, si=20, sf=40)
+    lrs = []
+    for _ in range(50):
+        s.step()
+        lrs.append(opt.param_groups[0]["lr"])
+    assert lrs[4] < lrs[9] <= 0.1              # ramp
+    assert abs(lrs[15] - 0.1) < 1e-9           # keep
+    assert lrs[25] < 0.1 and lrs[25] > 0.001   # decay
+    assert abs(lrs[45] - 0.001) < 1e-9         # floor
+
+
+def test_val_loss_scheduler_reduces_on_plateau():
+    opt = torch.optim.SGD([torch.nn.Parameter(torch.zeros(1))], lr=1.0)
+    s = make_lr_scheduler({"type": "val_loss", "gamma": 0.5,
+                           "step_interval": "epoch", "patience": 0,
+                           "step_size": 1}, opt)
+    s.step(1.0)
+    s.step(1.0)  # no improvement -> decay
+    assert opt.param_groups[0]["lr"] < 1.0
+
+
+# ---- privacy accountant ----------------------------------------------
+
+
+def test_rdp_eps_decreases_with_noise():
+    orders = [1.5, 2, 4, 8, 16, 32, 64]
+    lo = get_privacy_spent(orders, compute_rdp(0.01, 4.0, 100, orders),
+                           target_delta=1e-7)[0]
+    hi = get_privacy_spent(orders, compute_rdp(0.01, 0.5, 100, orders),
+                           target_delta=1e-7)[0]
+    assert lo < hi
+
+
+def test_rdp_eps_grows_with_steps():
+    orders = [1.5, 2, 4, 8, 16, 32, 64]
+    few = get_privacy_spent(orders, compute_rdp(0.01, 1.0, 10, orders), 1e-7)[0]
+    many = get_privacy_spent(orders, compute_rdp(0.01, 1.0, 1000, orders), 1e-7)[0]
+    assert many > few > 0
+
+
+# ---- DGA weighting -----------------------------------------------------
+
+
+def test_softmax_weights_prefers_low_loss():
+    # reference semantics: UNNORMALIZED exp(-beta*loss) — the weight sum
+    # normalizes at aggregation time (dga.py:111-129 + fedavg normalize)
+    w = torch.as_tensor(softmax_weights(torch.tensor([1.0, 2.0, 3.0]), beta=1.0))
+    assert w[0] > w[1] > w[2] > 0
+    assert abs(float(w[0]) - math.exp(-1.0)) < 1e-6
+
+
+def test_filter_weight_kills_nonfinite():
+    assert filter_weight(float("inf")) == 0.0
+    assert filter_weight(float("nan")) == 0.0
+    assert filter_weight(2.0) == 2.0
+
+
+# ---- samplers ----------------------------------------------------------
+
+
+class _FakeDs:
+    def __init__(self, durations):
+        self.utt_list = [{"duration": d} for d in durations]
+
+    def __len__(self):
+        return len(self.utt_list)
+
+
+class _Sampler:
+    def __init__(self, ds):
+        self.dataset = ds
+
+    def __iter__(self):
+        return iter(range(len(self.dataset)))
+
+
+def test_dynamic_batch_sampler_respects_frame_budget():
+    ds = _FakeDs([5, 10, 3, 8, 25, 2, 7, 1])
+    s = DynamicBatchSampler(_Sampler(ds), frames_threshold=25,
+                            max_batch_size=0, unsorted_batch=False, fps=1)
+    seen = []
+    for batch in s.batches:
+        total = sum(ds.utt_list[i]["duration"] for i in batch)
+        assert total <= 25
+        seen.extend(batch)
+    assert sorted(seen) == list(range(8))
+
+
+# ---- quantization semantics -------------------------------------------
+
+
+def test_quantize_matches_naive_binning():
+    torch.manual_seed(4)
+    x = torch.randn(5000)
+    got = ref.quantize_dequantize(x.clone(), n_bins=16, threshold_quantile=0.3)
+    thresh = torch.quantile(x.abs(), 0.3)
+    mn, mx = x.min(), x.max()
+    bins = torch.linspace(mn, mx, 16)
+    width = bins[1] - bins[0]
+    idx = torch.bucketize(x - width / 2, bins).clamp(0, 15)
+    expect = torch.where(x.abs() <= thresh, torch.zeros(()), bins[idx])
+    assert torch.allclose(got, expect, atol=float(width) * 1.01 + 1e-6)
+    assert ((got == 0) == (x.abs() <= thresh)).float().mean() > 0.999
