@@ -102,6 +102,21 @@ def main():
         float(g.loss_acc)
     res["loss_sync_ms"] = timeit(sync_loss, 50, sync=False)
 
+    # pure enqueue cost of a replay (no sync): is hipGraphLaunch the wall?
+    res["replay_enqueue_ms"] = timeit(lambda: g.graph.replay(), 400,
+                                      sync=False)
+    torch.cuda.synchronize()
+    from msrflute_amd.ops.graphs import epoch_graph_for
+    ds_dev = dl.dataset
+    eg = epoch_graph_for(cache, ds_dev.x, ds_dev.y, 20)
+    if eg is not None:
+        res["epoch_replay_enqueue_ms"] = timeit(
+            lambda: eg.graph.replay(), 200, sync=False)
+        torch.cuda.synchronize()
+        order = torch.randperm(len(ds_dev.x))
+        res["epoch_run_synced_ms"] = timeit(
+            lambda: eg.run_epoch(ds_dev.x, ds_dev.y, order), 50)
+
     # graphed epoch via trainer
     from msrflute_amd.core.trainer import Trainer
     opt = ex._make_optimizer(0.1)
