@@ -28,7 +28,8 @@ from ..strategies import select_strategy
 from ..utils import log_metric, print_rank, update_json_log
 from .client import Client, ClientExecutor
 from .evaluation import Evaluation
-from .trainer import ModelUpdater, Trainer, get_lr, set_component_wise_lr
+from .trainer import (ModelUpdater, Trainer, flush_saves, get_lr,
+                      set_component_wise_lr)
 
 
 class OptimizationServer:
@@ -157,6 +158,7 @@ class OptimizationServer:
     # ------------------------------------------------------------------
     def load_saved_status(self):
         """Resume from latest checkpoint + status log (reference: server.py:183-204)."""
+        flush_saves()
         if os.path.exists(self.last_model_path):
             print_rank(f"Resuming from checkpoint model {self.last_model_path}")
             self.worker_trainer.load(self.last_model_path,
@@ -177,6 +179,12 @@ class OptimizationServer:
             print_rank(f"Resuming from status_log: cur_iter: {self.cur_iter_no}")
 
     def run(self):
+        try:
+            self._run_impl()
+        finally:
+            flush_saves()  # all queued checkpoint writes hit disk
+
+    def _run_impl(self):
         print_rank("server started")
         self.train()
         print_rank("server terminated")
@@ -526,6 +534,8 @@ class OptimizationServer:
     # ------------------------------------------------------------------
     def backup_models(self, i):
         """Reference: server.py:530-559."""
+        if (i % self.model_backup_freq) == 0:
+            flush_saves()  # epoch snapshots copy the best_* files
         self.worker_trainer.save(model_path=self.model_path, token="latest",
                                  config=self.config["server_config"])
         if (i % self.model_backup_freq) == 0:
@@ -540,6 +550,7 @@ class OptimizationServer:
                     shutil.copyfile(src, dst)
 
     def fall_back_to_prev_best_status(self):
+        flush_saves()
         """Reference: server.py:561-578."""
         if not self.fall_back_to_best_model:
             return
