@@ -154,5 +154,70 @@ def main():
     print(json.dumps(res, indent=1))
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and not os.environ.get("MB_THREADS"):
     main()
+
+
+def thread_test():
+    """Does hipGraphLaunch release the GIL? 8 epoch replays serial vs on
+    4 threads."""
+    import json
+    from concurrent.futures import ThreadPoolExecutor
+    main_res = {}
+    # rebuild minimal state
+    import argparse
+    from bench import build_config
+    from msrflute_amd.core import client as client_mod
+    from msrflute_amd.core.client import Client, ClientExecutor
+    from msrflute_amd.models import make_model
+    from msrflute_amd.models.generic_data import ArrayDataset
+    from msrflute_amd.ops.arena import ParameterArena
+    from msrflute_amd.ops.graphs import epoch_graph_for
+    from tools.create_data import make_femnist_blob
+
+    args = argparse.Namespace(warmup=1, steps=1, clients_per_round=10)
+    config = build_config(args)
+    config["model_path"] = "/tmp/mb_models"
+    os.makedirs(config["model_path"], exist_ok=True)
+    blob = make_femnist_blob(n_users=8, samples_per_user=100, seed=7)
+    ds = ArrayDataset(blob, test_only=False, user_idx=-1, args={},
+                      x_shape=(28, 28))
+    ds.user_data = blob["user_data"]
+    ds.user_data_label = blob["user_data_label"]
+    client_mod.train_dataset = ds
+
+    torch.manual_seed(1)
+    server_arena = ParameterArena(make_model(config["model_config"]),
+                                  bind_grads=True)
+    exs = [ClientExecutor(config, "cv_cnn_femnist", None, server_arena)
+           for _ in range(4)]
+    streams = [torch.cuda.Stream() for _ in exs]
+    egs = []
+    for ex in exs:
+        x = torch.randn(100, 28, 28, device="cuda")
+        y = torch.randint(0, 62, (100,), device="cuda")
+        eg = epoch_graph_for(ex.graph_cache, x, y, 20)
+        ex.graph_cache.set_lr(0.1)
+        egs.append((eg, x, y))
+
+    def one(k):
+        eg, x, y = egs[k % 4]
+        with torch.cuda.stream(streams[k % 4]):
+            eg.run_epoch(x, y, torch.randperm(100))
+
+    def serial8():
+        for k in range(8):
+            one(k)
+    main_res["serial8_ms"] = timeit(serial8, 20)
+
+    pool = ThreadPoolExecutor(max_workers=4)
+
+    def threaded8():
+        list(pool.map(one, range(8)))
+    main_res["threaded8_ms"] = timeit(threaded8, 20)
+    print(json.dumps(main_res))
+
+
+if __name__ == "__main__" and os.environ.get("MB_THREADS"):
+    thread_test()
+    sys.exit(0)
