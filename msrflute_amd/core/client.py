@@ -137,6 +137,30 @@ class ClientExecutor:
 
         # hipGraph fast path for the per-batch client step (ops/graphs.py);
         # enabled on GPU for plain-SGD clients unless disabled by config
+        # Fully-fused hand-written kernel path for the flagship CNN
+        # (ops/fused_cnn.py): engaged when the arena layout matches and the
+        # client optimizer is plain SGD (momentum/wd 0).
+        self.fused_cnn = None
+        opt_c = dict(self.client_config["optimizer_config"])
+        if (torch.cuda.is_available() and ops.HAS_EXT
+                and self.client_config.get("use_fused_cnn", True)
+                and opt_c.get("type", "sgd") == "sgd"
+                and float(opt_c.get("momentum", 0.0)) == 0.0
+                and float(opt_c.get("weight_decay", 0.0)) == 0.0):
+            from ..ops.fused_cnn import FusedCNNEpoch, matches_cnn_femnist
+            C = matches_cnn_femnist(self.arena)
+            if C is not None:
+                drops = [m.p for m in self.model.modules()
+                         if isinstance(m, torch.nn.Dropout)]
+                self.fused_cnn = FusedCNNEpoch(
+                    self.arena, C,
+                    bs=int(self.client_config["data_config"]["train"]
+                           .get("batch_size", 20)),
+                    p1=drops[0] if len(drops) > 0 else 0.25,
+                    p2=drops[1] if len(drops) > 1 else 0.5,
+                    max_grad_norm=self.client_config["data_config"]["train"]
+                    .get("max_grad_norm"))
+
         # MIOpen RNN kernels segfault under hipGraph capture (hipblaslt
         # assert -> SIGSEGV, observed with the fedshakespeare LSTM), so
         # models containing RNN modules always run the eager path.
@@ -281,6 +305,8 @@ class ClientExecutor:
             arena=self.arena,
         )
         trainer.graph_cache = self.graph_cache
+        trainer.fused_cnn = self.fused_cnn
+        trainer.round_seed = round_seed & 0x7FFFFFFFFFFF
 
         desired_max_samples = data_config.get("desired_max_samples", None)
         apply_privacy_metrics = bool(privacy_metrics_config

@@ -153,8 +153,11 @@ class Trainer(TrainerBase):
         super().__init__(model=model, train_dataloader=train_dataloader,
                          optimizer=optimizer, max_grad_norm=max_grad_norm,
                          ignore_subtask=ignore_subtask, arena=arena)
-        # optional hipGraph fast path, injected by ClientExecutor
+        # optional hipGraph / fused-kernel fast paths, injected by
+        # ClientExecutor
         self.graph_cache = None
+        self.fused_cnn = None
+        self.round_seed = 0
         self.server_replay_config = server_replay_config
         self.anneal_config = anneal_config
         self.lr_scheduler = None
@@ -291,8 +294,17 @@ class Trainer(TrainerBase):
                         apply_privacy_metrics=False, prox=None):
         """Reference: trainer.py:341-414 (and 416-501 when ``prox`` is set:
         prox = (mu, w_global_flat))."""
-        if (self.graph_cache is not None and prox is None
-                and not apply_privacy_metrics
+        fast_ok = prox is None and not apply_privacy_metrics
+        if fast_ok and self.fused_cnn is not None:
+            ds = getattr(self.train_dataloader, "dataset", None)
+            if (ds is not None and torch.is_tensor(getattr(ds, "x", None))
+                    and ds.x.is_cuda and torch.is_tensor(getattr(ds, "y", None))
+                    and ds.y.dim() == 1 and ds.x[0].numel() == 784
+                    and getattr(self.train_dataloader, "shuffle", False)
+                    and (desired_max_samples is None
+                         or desired_max_samples >= len(ds.x))):
+                return self._run_epoch_fused_cnn(ds)
+        if (self.graph_cache is not None and fast_ok
                 and self.graph_cache.supports()):
             return self._run_train_epoch_graphed(desired_max_samples)
         num_samples = 0
@@ -436,6 +448,28 @@ class Trainer(TrainerBase):
             loss_total = 0.0
         self._finalize_sufficient_stats()
         return num_samples, loss_total
+
+    def _run_epoch_fused_cnn(self, ds):
+        """One local epoch via the hand-written fused CNN kernels
+        (ops/fused_cnn.py) — no autograd, no graphs."""
+        self.reset_gradient_power()
+        fc = self.fused_cnn
+        fc.lr_t  # noqa: B018 — touch to assert constructed
+        lr = get_lr(self.optimizer) if self.optimizer is not None else 0.0
+        order = torch.randperm(len(ds.x))
+        n, n_batches = fc.run_epoch(ds.x, ds.y, order, lr, self.round_seed)
+        self.step += n_batches
+        if self.lr_scheduler is not None:
+            self.lr_scheduler.step()
+        self._stats_acc += fc.stats_acc
+        self.counter += n_batches * self.arena.total
+        if self.lazy_stats:
+            self.loss_dev = fc.loss_acc.reshape(()).clone()
+            self.stats_dev = self._stats_acc.clone()
+            return n, None
+        loss_total = float(fc.loss_acc)
+        self._finalize_sufficient_stats()
+        return n, loss_total
 
     def _run_epoch_one_graph(self, eg, ds, bs, cache):
         """Run one client epoch as a single graph replay + eager ragged

@@ -37,6 +37,20 @@ void launch_gru_gates(const float*, const float*, const float*, float*, int,
                       int, hipStream_t);
 }
 
+struct CnnWorkspace {
+  float *xb, *a1, *r2, *a2, *z3, *a3, *dlogits, *dz3, *da2, *dz2, *dz1;
+  int *yb;
+  unsigned char *pidx, *m2, *m3;
+  double *red_partials, *red_acc;
+};
+
+extern "C" void launch_cnn_epoch(
+    const float* shard_x, const long long* shard_y, const long long* order,
+    long long n, int bs, int C, float* params, float* grads,
+    CnnWorkspace ws, const float* lr_t, float max_norm, float p1, float p2,
+    float* stats_acc, float* loss_acc, unsigned long long seed,
+    hipStream_t s);
+
 namespace {
 
 hipStream_t cur_stream() {
@@ -199,6 +213,69 @@ void quant_bin_mask(torch::Tensor x, torch::Tensor min_t, torch::Tensor max_t,
                         thresh_t.data_ptr<float>(), (int)n_bins, cur_stream());
 }
 
+// fully-fused CNN-FEMNIST client epoch (fused_cnn.hip): one host call
+// trains a whole local epoch — fwd, bwd, clip+stats, SGD per batch —
+// bypassing autograd and hip graphs entirely.
+void cnn_epoch(torch::Tensor shard_x, torch::Tensor shard_y,
+               torch::Tensor order, int64_t bs, int64_t C,
+               torch::Tensor params, torch::Tensor grads,
+               torch::Tensor work_f, torch::Tensor work_i,
+               torch::Tensor work_b, torch::Tensor work_d,
+               torch::Tensor lr_t, double max_norm, double p1, double p2,
+               torch::Tensor stats_acc, torch::Tensor loss_acc,
+               int64_t seed) {
+  check_flat(params, "params"); check_flat(grads, "grads");
+  check_flat(lr_t, "lr_t"); check_flat(stats_acc, "stats_acc");
+  check_flat(loss_acc, "loss_acc"); check_flat(work_f, "work_f");
+  TORCH_CHECK(shard_x.is_cuda() && shard_x.is_contiguous() &&
+              shard_x.scalar_type() == torch::kFloat32, "shard_x");
+  TORCH_CHECK(shard_y.is_cuda() && shard_y.is_contiguous() &&
+              shard_y.scalar_type() == torch::kInt64, "shard_y");
+  TORCH_CHECK(order.is_cuda() && order.scalar_type() == torch::kInt64,
+              "order must be int64 on device");
+  long long n = shard_y.numel();
+  TORCH_CHECK(shard_x.numel() == n * 784, "shard_x must be [n,784]");
+  int B = (int)bs;
+  // slice the float workspace
+  float* f = work_f.data_ptr<float>();
+  CnnWorkspace ws;
+  ws.xb = f;            f += (long long)B * 784;
+  ws.a1 = f;            f += (long long)B * 21632;
+  ws.r2 = f;            f += (long long)B * 36864;
+  ws.a2 = f;            f += (long long)B * 9216;
+  ws.z3 = f;            f += (long long)B * 128;
+  ws.a3 = f;            f += (long long)B * 128;
+  ws.dlogits = f;       f += (long long)B * C;
+  ws.dz3 = f;           f += (long long)B * 128;
+  ws.da2 = f;           f += (long long)B * 9216;
+  ws.dz2 = f;           f += (long long)B * 36864;
+  ws.dz1 = f;           f += (long long)B * 21632;
+  TORCH_CHECK(f - work_f.data_ptr<float>() <= work_f.numel(),
+              "float workspace too small");
+  ws.yb = work_i.data_ptr<int>();
+  TORCH_CHECK(work_i.numel() >= B, "int workspace too small");
+  unsigned char* u = work_b.data_ptr<unsigned char>();
+  ws.pidx = u;          u += (long long)B * 9216;
+  ws.m2 = u;            u += (long long)B * 9216;
+  ws.m3 = u;            u += (long long)B * 128;
+  TORCH_CHECK(u - work_b.data_ptr<unsigned char>() <= work_b.numel(),
+              "byte workspace too small");
+  ws.red_partials = work_d.data_ptr<double>();
+  ws.red_acc = work_d.data_ptr<double>() + work_d.numel() - 2;
+  TORCH_CHECK(work_d.numel() >= 2 * 2048 + 2, "double workspace too small");
+  launch_cnn_epoch(shard_x.data_ptr<float>(),
+                   reinterpret_cast<const long long*>(
+                       shard_y.data_ptr<int64_t>()),
+                   reinterpret_cast<const long long*>(
+                       order.data_ptr<int64_t>()),
+                   n, B, (int)C,
+                   params.data_ptr<float>(), grads.data_ptr<float>(), ws,
+                   lr_t.data_ptr<float>(), (float)max_norm, (float)p1,
+                   (float)p2, stats_acc.data_ptr<float>(),
+                   loss_acc.data_ptr<float>(), (unsigned long long)seed,
+                   cur_stream());
+}
+
 // fused GRU gate math (no-grad eval path of the nlg_gru recurrence)
 torch::Tensor gru_gates(torch::Tensor g_i, torch::Tensor g_h,
                         torch::Tensor h) {
@@ -231,4 +308,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segmented_sqnorm", &segmented_sqnorm);
   m.def("quant_bin_mask", &quant_bin_mask);
   m.def("gru_gates", &gru_gates);
+  m.def("cnn_epoch", &cnn_epoch);
 }

@@ -1,0 +1,529 @@
+// Fully-fused CNN-FEMNIST client step for gfx950 (BASELINE north star:
+// "Client.process_round() local-SGD ... as hand-written CDNA4 HIP
+// kernels").  The flagship benchmark model (reference
+// experiments/cv_cnn_femnist/model.py: conv1 1->32 3x3, relu, conv2
+// 32->64 3x3, relu, maxpool2, dropout .25, fc1 9216->128, relu,
+// dropout .5, fc2 128->C, CE) trained WITHOUT torch autograd: forward,
+// backward, clip+stats and the SGD step are ~17 kernel launches per
+// batch driven by ONE host call per epoch (launch_cnn_epoch).
+//
+// Why: hipGraphLaunch costs ~4.6us/node on this stack, so the ~230-node
+// autograd capture costs ~1.07 ms host per client epoch (PERF.md).  This
+// path is ~85 raw launches (~2us each) per epoch and owns its numerics:
+// gradients are written (not accumulated) straight into the flat arena
+// at the parameters' fixed offsets, so no zero_grad is needed either.
+//
+// Shapes are the task's (28x28 in, spatial 26/24/12 fixed); batch B and
+// class count C are runtime args.  Dropout uses Philox keyed by
+// (seed, batch_index) — deterministic per client epoch.
+
+#include <hip/hip_runtime.h>
+#include <hiprand/hiprand_kernel.h>
+
+#define FBLK 256
+
+// arena offsets (fp32 indices) for C classes
+struct CnnOffsets {
+  long long w1, b1, w2, b2, w3, b3, w4, b4, total;
+};
+
+static CnnOffsets cnn_offsets(int C) {
+  CnnOffsets o;
+  o.w1 = 0;               // [32,1,3,3]
+  o.b1 = o.w1 + 288;      // [32]
+  o.w2 = o.b1 + 32;       // [64,32,3,3]
+  o.b2 = o.w2 + 18432;    // [64]
+  o.w3 = o.b2 + 64;       // [128,9216]
+  o.b3 = o.w3 + 1179648;  // [128]
+  o.w4 = o.b3 + 128;      // [C,128]
+  o.b4 = o.w4 + (long long)C * 128;  // [C]
+  o.total = o.b4 + C;
+  return o;
+}
+
+// ---------------------------------------------------------------------------
+// gather a shuffled batch from the device-resident shard
+// ---------------------------------------------------------------------------
+__global__ void k_gather_batch(const float* __restrict__ shard_x,
+                               const long long* __restrict__ shard_y,
+                               const long long* __restrict__ order,
+                               long long start, int B,
+                               float* __restrict__ xb,
+                               int* __restrict__ yb) {
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < B * 784;
+       i += gridDim.x * blockDim.x) {
+    int b = i / 784, j = i % 784;
+    long long src = order[start + b];
+    xb[i] = shard_x[src * 784 + j];
+    if (j == 0) yb[b] = (int)shard_y[src];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// forward
+// ---------------------------------------------------------------------------
+// a1[b,co,y,x] = relu(b1 + sum_{kh,kw} w1[co,kh,kw] * x[b, y+kh, x+kw])
+__global__ void k_conv1_fwd(const float* __restrict__ x,
+                            const float* __restrict__ w1,
+                            const float* __restrict__ b1, int B,
+                            float* __restrict__ a1) {
+  int total = B * 32 * 26 * 26;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    int xx = i % 26, yy = (i / 26) % 26, co = (i / 676) % 32, b = i / 21632;
+    const float* xp = x + b * 784 + yy * 28 + xx;
+    const float* wp = w1 + co * 9;
+    float acc = b1[co];
+    #pragma unroll
+    for (int kh = 0; kh < 3; ++kh)
+      #pragma unroll
+      for (int kw = 0; kw < 3; ++kw)
+        acc = fmaf(wp[kh * 3 + kw], xp[kh * 28 + kw], acc);
+    a1[i] = acc > 0.f ? acc : 0.f;
+  }
+}
+
+// r2[b,co,y,x] = relu(b2 + sum_ci sum_k w2[co,ci,k]*a1[b,ci,y+kh,x+kw])
+// One block per (b, co); a1's 32 input channels are staged through LDS in
+// two 16-channel tiles (16*676*4 B = 42.25 KB, under the 64 KB static
+// LDS limit) and reused by all 576 outputs.
+__global__ void k_conv2_fwd(const float* __restrict__ a1,
+                            const float* __restrict__ w2,
+                            const float* __restrict__ b2, int B,
+                            float* __restrict__ r2) {
+  __shared__ float lds[16 * 676];
+  int b = blockIdx.x / 64, co = blockIdx.x % 64;
+  const float* wp = w2 + co * 288;  // [32,3,3]
+  float acc[3];  // 576 outputs / 256 threads -> up to 3 per thread
+  #pragma unroll
+  for (int r = 0; r < 3; ++r) acc[r] = b2[co];
+  for (int half = 0; half < 2; ++half) {
+    const float* src = a1 + ((long long)b * 32 + half * 16) * 676;
+    __syncthreads();
+    for (int i = threadIdx.x; i < 16 * 676; i += blockDim.x)
+      lds[i] = src[i];
+    __syncthreads();
+    for (int r = 0; r < 3; ++r) {
+      int o = threadIdx.x + r * FBLK;
+      if (o >= 576) break;
+      int xx = o % 24, yy = o / 24;
+      float s = acc[r];
+      for (int ci = 0; ci < 16; ++ci) {
+        const float* ap = lds + ci * 676 + yy * 26 + xx;
+        const float* wq = wp + (half * 16 + ci) * 9;
+        #pragma unroll
+        for (int kh = 0; kh < 3; ++kh)
+          #pragma unroll
+          for (int kw = 0; kw < 3; ++kw)
+            s = fmaf(wq[kh * 3 + kw], ap[kh * 26 + kw], s);
+      }
+      acc[r] = s;
+    }
+  }
+  for (int r = 0; r < 3; ++r) {
+    int o = threadIdx.x + r * FBLK;
+    if (o < 576)
+      r2[((long long)b * 64 + co) * 576 + o] = acc[r] > 0.f ? acc[r] : 0.f;
+  }
+}
+
+// maxpool 2x2 + dropout(p1): a2 = keep ? max/(1-p1) : 0; save argmax+mask
+__global__ void k_pool_drop_fwd(const float* __restrict__ r2, int B,
+                                float p1, unsigned long long seed,
+                                unsigned long long offset,
+                                float* __restrict__ a2,
+                                unsigned char* __restrict__ pidx,
+                                unsigned char* __restrict__ m2) {
+  int total = B * 64 * 144;
+  float inv_keep = (p1 < 1.f) ? 1.f / (1.f - p1) : 0.f;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    int px = i % 12, py = (i / 12) % 12, c = (i / 144) % 64, b = i / 9216;
+    const float* base = r2 + (((long long)b * 64 + c) * 24 + 2 * py) * 24 + 2 * px;
+    float v0 = base[0], v1 = base[1], v2 = base[24], v3 = base[25];
+    float m = v0; int idx = 0;
+    if (v1 > m) { m = v1; idx = 1; }
+    if (v2 > m) { m = v2; idx = 2; }
+    if (v3 > m) { m = v3; idx = 3; }
+    unsigned char keep = 1;
+    if (p1 > 0.f) {
+      hiprandStatePhilox4_32_10_t st;
+      hiprand_init(seed, (unsigned long long)i, offset, &st);
+      keep = hiprand_uniform(&st) >= p1;
+    }
+    pidx[i] = (unsigned char)idx;
+    m2[i] = keep;
+    a2[i] = keep ? m * inv_keep : 0.f;
+  }
+}
+
+// z3 = W3 @ a2flat + b3 ; a3 = dropout(relu(z3), p2)
+// one block per (b, j): 256-thread parallel dot of length 9216
+__global__ void k_fc1_fwd(const float* __restrict__ a2,
+                          const float* __restrict__ w3,
+                          const float* __restrict__ b3, int B, float p2,
+                          unsigned long long seed, unsigned long long offset,
+                          float* __restrict__ z3, float* __restrict__ a3,
+                          unsigned char* __restrict__ m3) {
+  int b = blockIdx.x / 128, j = blockIdx.x % 128;
+  const float* ap = a2 + (long long)b * 9216;
+  const float* wp = w3 + (long long)j * 9216;
+  float s = 0.f;
+  for (int k = threadIdx.x; k < 9216; k += blockDim.x)
+    s = fmaf(wp[k], ap[k], s);
+  for (int d = 32; d > 0; d >>= 1) s += __shfl_down(s, d, 64);
+  __shared__ float lds[FBLK / 64];
+  if ((threadIdx.x & 63) == 0) lds[threadIdx.x >> 6] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = b3[j];
+    for (int w = 0; w < FBLK / 64; ++w) t += lds[w];
+    int i = b * 128 + j;
+    z3[i] = t;
+    float r = t > 0.f ? t : 0.f;
+    unsigned char keep = 1;
+    if (p2 > 0.f) {
+      hiprandStatePhilox4_32_10_t st;
+      hiprand_init(seed ^ 0x9e3779b97f4a7c15ull, (unsigned long long)i,
+                   offset, &st);
+      keep = hiprand_uniform(&st) >= p2;
+    }
+    m3[i] = keep;
+    a3[i] = keep ? r / (1.f - p2) : 0.f;
+  }
+}
+
+// logits = W4 @ a3 + b4 ; softmax ; loss += -log p[target]/B ;
+// dlogit = (p - onehot)/B.  One block per sample row.
+__global__ void k_fc2_loss_fwd(const float* __restrict__ a3,
+                               const float* __restrict__ w4,
+                               const float* __restrict__ b4,
+                               const int* __restrict__ yb, int B, int C,
+                               float* __restrict__ dlogits,
+                               float* __restrict__ loss_acc) {
+  extern __shared__ float sm[];  // C floats
+  int b = blockIdx.x;
+  const float* ap = a3 + (long long)b * 128;
+  for (int j = threadIdx.x; j < C; j += blockDim.x) {
+    const float* wp = w4 + (long long)j * 128;
+    float s = b4[j];
+    #pragma unroll 4
+    for (int k = 0; k < 128; ++k) s = fmaf(wp[k], ap[k], s);
+    sm[j] = s;
+  }
+  __syncthreads();
+  // softmax over C (single wave handles it: C <= 1024 assumed small)
+  if (threadIdx.x == 0) {
+    float mx = sm[0];
+    for (int j = 1; j < C; ++j) mx = fmaxf(mx, sm[j]);
+    float z = 0.f;
+    for (int j = 0; j < C; ++j) { sm[j] = __expf(sm[j] - mx); z += sm[j]; }
+    float inv = 1.f / z;
+    int t = yb[b];
+    for (int j = 0; j < C; ++j) {
+      float p = sm[j] * inv;
+      dlogits[(long long)b * C + j] = (p - (j == t ? 1.f : 0.f)) / (float)B;
+    }
+    atomicAdd(loss_acc, -__logf(fmaxf(sm[t] * inv, 1e-30f)) / (float)B);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward
+// ---------------------------------------------------------------------------
+// dW4[j,k] = sum_b dlogits[b,j]*a3[b,k]; db4[j] = sum_b dlogits[b,j]
+__global__ void k_fc2_bwd_w(const float* __restrict__ dlogits,
+                            const float* __restrict__ a3, int B, int C,
+                            float* __restrict__ dw4, float* __restrict__ db4) {
+  int total = C * 128;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    int k = i % 128, j = i / 128;
+    float s = 0.f, sb = 0.f;
+    for (int b = 0; b < B; ++b) {
+      float d = dlogits[(long long)b * C + j];
+      s = fmaf(d, a3[(long long)b * 128 + k], s);
+      if (k == 0) sb += d;
+    }
+    dw4[i] = s;
+    if (k == 0) db4[j] = sb;
+  }
+}
+
+// dz3[b,k] = (sum_j dlogits[b,j]*W4[j,k]) * m3/(1-p2) * (z3>0)
+__global__ void k_fc2_bwd_x(const float* __restrict__ dlogits,
+                            const float* __restrict__ w4,
+                            const float* __restrict__ z3,
+                            const unsigned char* __restrict__ m3, int B,
+                            int C, float p2, float* __restrict__ dz3) {
+  int total = B * 128;
+  float inv_keep = 1.f / (1.f - p2);
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    int k = i % 128, b = i / 128;
+    float s = 0.f;
+    for (int j = 0; j < C; ++j)
+      s = fmaf(dlogits[(long long)b * C + j], w4[(long long)j * 128 + k], s);
+    float g = (p2 > 0.f) ? (m3[i] ? s * inv_keep : 0.f) : s;
+    dz3[i] = z3[i] > 0.f ? g : 0.f;
+  }
+}
+
+// dW3[j,k] = sum_b dz3[b,j]*a2[b,k]; db3[j] = sum_b dz3[b,j]
+__global__ void k_fc1_bwd_w(const float* __restrict__ dz3,
+                            const float* __restrict__ a2, int B,
+                            float* __restrict__ dw3, float* __restrict__ db3) {
+  long long total = 128LL * 9216;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    int k = (int)(i % 9216), j = (int)(i / 9216);
+    float s = 0.f, sb = 0.f;
+    for (int b = 0; b < B; ++b) {
+      float d = dz3[b * 128 + j];
+      s = fmaf(d, a2[(long long)b * 9216 + k], s);
+      if (k == 0) sb += d;
+    }
+    dw3[i] = s;
+    if (k == 0) db3[j] = sb;
+  }
+}
+
+// d_a2[b,k] = sum_j dz3[b,j] * W3[j,k]  (then dropout1 bwd is fused in
+// the pool scatter)
+__global__ void k_fc1_bwd_x(const float* __restrict__ dz3,
+                            const float* __restrict__ w3, int B,
+                            float* __restrict__ da2) {
+  long long total = (long long)B * 9216;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    int k = (int)(i % 9216), b = (int)(i / 9216);
+    const float* dp = dz3 + b * 128;
+    float s = 0.f;
+    #pragma unroll 4
+    for (int j = 0; j < 128; ++j)
+      s = fmaf(dp[j], w3[(long long)j * 9216 + k], s);
+    da2[i] = s;
+  }
+}
+
+// dropout1 bwd + maxpool scatter + relu bwd: one thread per pool cell
+// writes its 2x2 input cells (exactly one gets the grad)
+__global__ void k_pool_drop_bwd(const float* __restrict__ da2,
+                                const unsigned char* __restrict__ pidx,
+                                const unsigned char* __restrict__ m2,
+                                const float* __restrict__ r2, int B, float p1,
+                                float* __restrict__ dz2) {
+  int total = B * 64 * 144;
+  float inv_keep = (p1 < 1.f) ? 1.f / (1.f - p1) : 0.f;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    int px = i % 12, py = (i / 12) % 12, c = (i / 144) % 64, b = i / 9216;
+    float g = (p1 > 0.f) ? (m2[i] ? da2[i] * inv_keep : 0.f) : da2[i];
+    int idx = pidx[i];
+    long long base = (((long long)b * 64 + c) * 24 + 2 * py) * 24 + 2 * px;
+    #pragma unroll
+    for (int d = 0; d < 4; ++d) {
+      long long o = base + (d >> 1) * 24 + (d & 1);
+      float v = (d == idx && r2[o] > 0.f) ? g : 0.f;
+      dz2[o] = v;
+    }
+  }
+}
+
+// dW2[co,ci,kh,kw] = sum_{b,y,x} a1[b,ci,y+kh,x+kw] * dz2[b,co,y,x]
+// one block per (co,ci) pair: 9 kernel taps reduced across 256 threads
+__global__ void k_conv2_bwd_w(const float* __restrict__ a1,
+                              const float* __restrict__ dz2, int B,
+                              float* __restrict__ dw2,
+                              float* __restrict__ db2) {
+  int co = blockIdx.x / 32, ci = blockIdx.x % 32;
+  float acc[9] = {0, 0, 0, 0, 0, 0, 0, 0, 0};
+  float accb = 0.f;
+  for (int t = threadIdx.x; t < B * 576; t += blockDim.x) {
+    int o = t % 576, b = t / 576;
+    int xx = o % 24, yy = o / 24;
+    float d = dz2[((long long)b * 64 + co) * 576 + o];
+    const float* ap = a1 + ((long long)b * 32 + ci) * 676 + yy * 26 + xx;
+    #pragma unroll
+    for (int kh = 0; kh < 3; ++kh)
+      #pragma unroll
+      for (int kw = 0; kw < 3; ++kw)
+        acc[kh * 3 + kw] = fmaf(ap[kh * 26 + kw], d, acc[kh * 3 + kw]);
+    if (ci == 0) accb += d;
+  }
+  __shared__ float lds[FBLK / 64 * 10];
+  int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  #pragma unroll
+  for (int k = 0; k < 9; ++k) {
+    float s = acc[k];
+    for (int d = 32; d > 0; d >>= 1) s += __shfl_down(s, d, 64);
+    if (lane == 0) lds[wave * 10 + k] = s;
+  }
+  float sb = accb;
+  for (int d = 32; d > 0; d >>= 1) sb += __shfl_down(sb, d, 64);
+  if (lane == 0) lds[wave * 10 + 9] = sb;
+  __syncthreads();
+  if (threadIdx.x < 9) {
+    float s = 0.f;
+    for (int w = 0; w < FBLK / 64; ++w) s += lds[w * 10 + threadIdx.x];
+    dw2[(long long)(co * 32 + ci) * 9 + threadIdx.x] = s;
+  }
+  if (threadIdx.x == 9 && ci == 0) {
+    float s = 0.f;
+    for (int w = 0; w < FBLK / 64; ++w) s += lds[w * 10 + 9];
+    db2[co] = s;
+  }
+}
+
+// dz1[b,ci,p,q] = relu'(a1) * sum_{co,kh,kw valid} W2[co,ci,kh,kw] *
+//                 dz2[b,co,p-kh,q-kw]
+__global__ void k_conv2_bwd_x(const float* __restrict__ dz2,
+                              const float* __restrict__ w2,
+                              const float* __restrict__ a1, int B,
+                              float* __restrict__ dz1) {
+  int total = B * 32 * 676;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    int q = i % 26, p = (i / 26) % 26, ci = (i / 676) % 32, b = i / 21632;
+    if (a1[i] <= 0.f) { dz1[i] = 0.f; continue; }
+    float s = 0.f;
+    #pragma unroll
+    for (int kh = 0; kh < 3; ++kh) {
+      int y = p - kh;
+      if (y < 0 || y >= 24) continue;
+      #pragma unroll
+      for (int kw = 0; kw < 3; ++kw) {
+        int x = q - kw;
+        if (x < 0 || x >= 24) continue;
+        for (int co = 0; co < 64; ++co)
+          s = fmaf(w2[((long long)co * 32 + ci) * 9 + kh * 3 + kw],
+                   dz2[((long long)b * 64 + co) * 576 + y * 24 + x], s);
+      }
+    }
+    dz1[i] = s;
+  }
+}
+
+// dW1[co,kh,kw] = sum_{b,y,x} x[b,y+kh,x+kw] * dz1[b,co,y,x]; db1 likewise
+__global__ void k_conv1_bwd_w(const float* __restrict__ x,
+                              const float* __restrict__ dz1, int B,
+                              float* __restrict__ dw1,
+                              float* __restrict__ db1) {
+  int co = blockIdx.x;  // 32 blocks
+  float acc[9] = {0, 0, 0, 0, 0, 0, 0, 0, 0};
+  float accb = 0.f;
+  for (int t = threadIdx.x; t < B * 676; t += blockDim.x) {
+    int o = t % 676, b = t / 676;
+    int xx = o % 26, yy = o / 26;
+    float d = dz1[((long long)b * 32 + co) * 676 + o];
+    const float* xp = x + b * 784 + yy * 28 + xx;
+    #pragma unroll
+    for (int kh = 0; kh < 3; ++kh)
+      #pragma unroll
+      for (int kw = 0; kw < 3; ++kw)
+        acc[kh * 3 + kw] = fmaf(xp[kh * 28 + kw], d, acc[kh * 3 + kw]);
+    accb += d;
+  }
+  __shared__ float lds[FBLK / 64 * 10];
+  int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  #pragma unroll
+  for (int k = 0; k < 9; ++k) {
+    float s = acc[k];
+    for (int d = 32; d > 0; d >>= 1) s += __shfl_down(s, d, 64);
+    if (lane == 0) lds[wave * 10 + k] = s;
+  }
+  float sb = accb;
+  for (int d = 32; d > 0; d >>= 1) sb += __shfl_down(sb, d, 64);
+  if (lane == 0) lds[wave * 10 + 9] = sb;
+  __syncthreads();
+  if (threadIdx.x < 9) {
+    float s = 0.f;
+    for (int w = 0; w < FBLK / 64; ++w) s += lds[w * 10 + threadIdx.x];
+    dw1[co * 9 + threadIdx.x] = s;
+  }
+  if (threadIdx.x == 9) {
+    float s = 0.f;
+    for (int w = 0; w < FBLK / 64; ++w) s += lds[w * 10 + 9];
+    db1[co] = s;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// epoch driver: the ONLY host entry — loops every batch of one client's
+// local epoch, launching fwd/bwd + fused clip-stats + SGD per batch.
+// ---------------------------------------------------------------------------
+
+extern "C" {
+void launch_sum_sumsq2(const float*, long long, double*, double*, hipStream_t);
+void launch_clip_apply_stats(float*, long long, const double*, float, float,
+                             float*, hipStream_t);
+void launch_sgd_step(float*, const float*, float*, float, const float*, float,
+                     float, float, int, int, long long, hipStream_t);
+}
+
+struct CnnWorkspace {
+  // laid out inside one float buffer by the binding (sizes for B rows)
+  float *xb, *a1, *r2, *a2, *z3, *a3, *dlogits, *dz3, *da2, *dz2, *dz1;
+  int *yb;
+  unsigned char *pidx, *m2, *m3;
+  double *red_partials, *red_acc;
+};
+
+extern "C" void launch_cnn_epoch(
+    const float* shard_x, const long long* shard_y, const long long* order,
+    long long n, int bs, int C, float* params, float* grads,
+    CnnWorkspace ws, const float* lr_t, float max_norm, float p1, float p2,
+    float* stats_acc, float* loss_acc, unsigned long long seed,
+    hipStream_t s) {
+  CnnOffsets o = cnn_offsets(C);
+  int n_batches = (int)((n + bs - 1) / bs);
+  for (int it = 0; it < n_batches; ++it) {
+    long long start = (long long)it * bs;
+    int B = (int)((start + bs <= n) ? bs : (n - start));
+    unsigned long long off = (unsigned long long)it;
+    int g1 = (B * 784 + FBLK - 1) / FBLK;
+    hipLaunchKernelGGL(k_gather_batch, dim3(g1), dim3(FBLK), 0, s,
+                       shard_x, shard_y, order, start, B, ws.xb, ws.yb);
+    hipLaunchKernelGGL(k_conv1_fwd, dim3((B * 21632 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.xb, params + o.w1, params + o.b1,
+                       B, ws.a1);
+    hipLaunchKernelGGL(k_conv2_fwd, dim3(B * 64), dim3(FBLK), 0, s,
+                       ws.a1, params + o.w2, params + o.b2, B, ws.r2);
+    hipLaunchKernelGGL(k_pool_drop_fwd, dim3((B * 9216 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.r2, B, p1, seed, off, ws.a2,
+                       ws.pidx, ws.m2);
+    hipLaunchKernelGGL(k_fc1_fwd, dim3(B * 128), dim3(FBLK), 0, s,
+                       ws.a2, params + o.w3, params + o.b3, B, p2, seed, off,
+                       ws.z3, ws.a3, ws.m3);
+    hipLaunchKernelGGL(k_fc2_loss_fwd, dim3(B), dim3(FBLK),
+                       C * (int)sizeof(float), s, ws.a3, params + o.w4,
+                       params + o.b4, ws.yb, B, C, ws.dlogits, loss_acc);
+    hipLaunchKernelGGL(k_fc2_bwd_w, dim3((C * 128 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.dlogits, ws.a3, B, C,
+                       grads + o.w4, grads + o.b4);
+    hipLaunchKernelGGL(k_fc2_bwd_x, dim3((B * 128 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.dlogits, params + o.w4, ws.z3,
+                       ws.m3, B, C, p2, ws.dz3);
+    hipLaunchKernelGGL(k_fc1_bwd_w, dim3(2048), dim3(FBLK), 0, s,
+                       ws.dz3, ws.a2, B, grads + o.w3, grads + o.b3);
+    hipLaunchKernelGGL(k_fc1_bwd_x, dim3((B * 9216 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.dz3, params + o.w3, B, ws.da2);
+    hipLaunchKernelGGL(k_pool_drop_bwd, dim3((B * 9216 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.da2, ws.pidx, ws.m2, ws.r2, B,
+                       p1, ws.dz2);
+    hipLaunchKernelGGL(k_conv2_bwd_w, dim3(64 * 32), dim3(FBLK), 0, s,
+                       ws.a1, ws.dz2, B, grads + o.w2, grads + o.b2);
+    hipLaunchKernelGGL(k_conv2_bwd_x, dim3((B * 21632 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.dz2, params + o.w2, ws.a1, B,
+                       ws.dz1);
+    hipLaunchKernelGGL(k_conv1_bwd_w, dim3(32), dim3(FBLK), 0, s,
+                       ws.xb, ws.dz1, B, grads + o.w1, grads + o.b1);
+    // fused clip + sufficient stats + SGD on the whole arena
+    hipMemsetAsync(ws.red_acc, 0, 2 * sizeof(double), s);
+    launch_sum_sumsq2(grads, o.total, ws.red_partials, ws.red_acc, s);
+    launch_clip_apply_stats(grads, o.total, ws.red_acc, max_norm, 1e-6f,
+                            stats_acc, s);
+    launch_sgd_step(params, grads, nullptr, 0.f, lr_t, 0.f, 0.f, 0.f, 0, 0,
+                    o.total, s);
+  }
+}

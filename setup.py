@@ -21,6 +21,7 @@ ext = CUDAExtension(
     sources=[
         "msrflute_amd/csrc/bindings.cpp",
         "msrflute_amd/csrc/flat_ops.hip",
+        "msrflute_amd/csrc/fused_cnn.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
