@@ -65,6 +65,7 @@ def build_config(args):
         },
         "client_config": {
             "parallel_clients": int(os.environ.get("BENCH_PAR", "4")),
+            "use_fused_cnn": os.environ.get("BENCH_FUSED", "1") == "1",
             "do_profiling": False,
             "ignore_subtask": False,
             "data_config": {
