@@ -149,7 +149,9 @@ class ClientExecutor:
                 and float(opt_c.get("weight_decay", 0.0)) == 0.0):
             from ..ops.fused_cnn import FusedCNNEpoch, matches_cnn_femnist
             C = matches_cnn_femnist(self.arena)
-            if C is not None:
+            bs_cfg = int(self.client_config["data_config"]["train"]
+                         .get("batch_size", 20))
+            if C is not None and bs_cfg <= 32:
                 drops = [m.p for m in self.model.modules()
                          if isinstance(m, torch.nn.Dropout)]
                 self.fused_cnn = FusedCNNEpoch(

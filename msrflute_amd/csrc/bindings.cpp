@@ -235,6 +235,7 @@ void cnn_epoch(torch::Tensor shard_x, torch::Tensor shard_y,
               "order must be int64 on device");
   long long n = shard_y.numel();
   TORCH_CHECK(shard_x.numel() == n * 784, "shard_x must be [n,784]");
+  TORCH_CHECK(bs >= 1 && bs <= 32, "fused CNN epoch supports batch <= 32");
   int B = (int)bs;
   // slice the float workspace
   float* f = work_f.data_ptr<float>();

@@ -84,19 +84,24 @@ __global__ void k_conv1_fwd(const float* __restrict__ x,
 }
 
 // r2[b,co,y,x] = relu(b2 + sum_ci sum_k w2[co,ci,k]*a1[b,ci,y+kh,x+kw])
-// One block per (b, co); a1's 32 input channels are staged through LDS in
-// two 16-channel tiles (16*676*4 B = 42.25 KB, under the 64 KB static
-// LDS limit) and reused by all 576 outputs.
+// One block per (b, group of 8 co); a1's 32 input channels are staged
+// through LDS in two 16-channel tiles (42.25 KB) and reused by all 8
+// output channels — 8x less LDS-staging traffic than one-co-per-block.
+// W2 taps for the tile live in registers (8 co x 9 = 72 floats/thread is
+// too many, so taps reload per ci from L1 — W2 is 72 KB, L2-resident).
+#define CONV2_COG 8
 __global__ void k_conv2_fwd(const float* __restrict__ a1,
                             const float* __restrict__ w2,
                             const float* __restrict__ b2, int B,
                             float* __restrict__ r2) {
   __shared__ float lds[16 * 676];
-  int b = blockIdx.x / 64, co = blockIdx.x % 64;
-  const float* wp = w2 + co * 288;  // [32,3,3]
-  float acc[3];  // 576 outputs / 256 threads -> up to 3 per thread
+  int b = blockIdx.x / (64 / CONV2_COG);
+  int co0 = (blockIdx.x % (64 / CONV2_COG)) * CONV2_COG;
+  float acc[3][CONV2_COG];
   #pragma unroll
-  for (int r = 0; r < 3; ++r) acc[r] = b2[co];
+  for (int r = 0; r < 3; ++r)
+    #pragma unroll
+    for (int g = 0; g < CONV2_COG; ++g) acc[r][g] = b2[co0 + g];
   for (int half = 0; half < 2; ++half) {
     const float* src = a1 + ((long long)b * 32 + half * 16) * 676;
     __syncthreads();
@@ -107,23 +112,34 @@ __global__ void k_conv2_fwd(const float* __restrict__ a1,
       int o = threadIdx.x + r * FBLK;
       if (o >= 576) break;
       int xx = o % 24, yy = o / 24;
-      float s = acc[r];
       for (int ci = 0; ci < 16; ++ci) {
         const float* ap = lds + ci * 676 + yy * 26 + xx;
-        const float* wq = wp + (half * 16 + ci) * 9;
+        float a[9];
         #pragma unroll
         for (int kh = 0; kh < 3; ++kh)
           #pragma unroll
           for (int kw = 0; kw < 3; ++kw)
-            s = fmaf(wq[kh * 3 + kw], ap[kh * 26 + kw], s);
+            a[kh * 3 + kw] = ap[kh * 26 + kw];
+        #pragma unroll
+        for (int g = 0; g < CONV2_COG; ++g) {
+          const float* wq = w2 + ((long long)(co0 + g) * 32
+                                  + half * 16 + ci) * 9;
+          float s = acc[r][g];
+          #pragma unroll
+          for (int k = 0; k < 9; ++k) s = fmaf(wq[k], a[k], s);
+          acc[r][g] = s;
+        }
       }
-      acc[r] = s;
     }
   }
   for (int r = 0; r < 3; ++r) {
     int o = threadIdx.x + r * FBLK;
     if (o < 576)
-      r2[((long long)b * 64 + co) * 576 + o] = acc[r] > 0.f ? acc[r] : 0.f;
+      #pragma unroll
+      for (int g = 0; g < CONV2_COG; ++g) {
+        float v = acc[r][g];
+        r2[((long long)b * 64 + co0 + g) * 576 + o] = v > 0.f ? v : 0.f;
+      }
   }
 }
 
@@ -158,26 +174,37 @@ __global__ void k_pool_drop_fwd(const float* __restrict__ r2, int B,
 }
 
 // z3 = W3 @ a2flat + b3 ; a3 = dropout(relu(z3), p2)
-// one block per (b, j): 256-thread parallel dot of length 9216
+// one block per output j: the W3 row is read ONCE and dotted against all
+// B sample rows simultaneously (W3 is 4.7 MB — the one-block-per-(b,j)
+// form re-read it B times and thrashed L2).
+#define FC1_BMAX 32
 __global__ void k_fc1_fwd(const float* __restrict__ a2,
                           const float* __restrict__ w3,
                           const float* __restrict__ b3, int B, float p2,
                           unsigned long long seed, unsigned long long offset,
                           float* __restrict__ z3, float* __restrict__ a3,
                           unsigned char* __restrict__ m3) {
-  int b = blockIdx.x / 128, j = blockIdx.x % 128;
-  const float* ap = a2 + (long long)b * 9216;
+  int j = blockIdx.x;
   const float* wp = w3 + (long long)j * 9216;
-  float s = 0.f;
-  for (int k = threadIdx.x; k < 9216; k += blockDim.x)
-    s = fmaf(wp[k], ap[k], s);
-  for (int d = 32; d > 0; d >>= 1) s += __shfl_down(s, d, 64);
-  __shared__ float lds[FBLK / 64];
-  if ((threadIdx.x & 63) == 0) lds[threadIdx.x >> 6] = s;
+  float acc[FC1_BMAX];
+  #pragma unroll
+  for (int b = 0; b < FC1_BMAX; ++b) acc[b] = 0.f;
+  for (int k = threadIdx.x; k < 9216; k += blockDim.x) {
+    float w = wp[k];
+    for (int b = 0; b < B; ++b)
+      acc[b] = fmaf(w, a2[(long long)b * 9216 + k], acc[b]);
+  }
+  __shared__ float lds[FBLK / 64][FC1_BMAX];
+  int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  for (int b = 0; b < B; ++b) {
+    float s = acc[b];
+    for (int d = 32; d > 0; d >>= 1) s += __shfl_down(s, d, 64);
+    if (lane == 0) lds[wave][b] = s;
+  }
   __syncthreads();
-  if (threadIdx.x == 0) {
+  for (int b = threadIdx.x; b < B; b += blockDim.x) {
     float t = b3[j];
-    for (int w = 0; w < FBLK / 64; ++w) t += lds[w];
+    for (int w = 0; w < FBLK / 64; ++w) t += lds[w][b];
     int i = b * 128 + j;
     z3[i] = t;
     float r = t > 0.f ? t : 0.f;
@@ -487,12 +514,12 @@ extern "C" void launch_cnn_epoch(
     hipLaunchKernelGGL(k_conv1_fwd, dim3((B * 21632 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.xb, params + o.w1, params + o.b1,
                        B, ws.a1);
-    hipLaunchKernelGGL(k_conv2_fwd, dim3(B * 64), dim3(FBLK), 0, s,
-                       ws.a1, params + o.w2, params + o.b2, B, ws.r2);
+    hipLaunchKernelGGL(k_conv2_fwd, dim3(B * (64 / CONV2_COG)), dim3(FBLK),
+                       0, s, ws.a1, params + o.w2, params + o.b2, B, ws.r2);
     hipLaunchKernelGGL(k_pool_drop_fwd, dim3((B * 9216 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.r2, B, p1, seed, off, ws.a2,
                        ws.pidx, ws.m2);
-    hipLaunchKernelGGL(k_fc1_fwd, dim3(B * 128), dim3(FBLK), 0, s,
+    hipLaunchKernelGGL(k_fc1_fwd, dim3(128), dim3(FBLK), 0, s,
                        ws.a2, params + o.w3, params + o.b3, B, p2, seed, off,
                        ws.z3, ws.a3, ws.m3);
     hipLaunchKernelGGL(k_fc2_loss_fwd, dim3(B), dim3(FBLK),
