@@ -163,9 +163,13 @@ __global__ void k_pool_drop_fwd(const float* __restrict__ r2, int B,
     if (v3 > m) { m = v3; idx = 3; }
     unsigned char keep = 1;
     if (p1 > 0.f) {
+      // one Philox draw covers 4 cells (init is ~10 rounds — the cost)
       hiprandStatePhilox4_32_10_t st;
-      hiprand_init(seed, (unsigned long long)i, offset, &st);
-      keep = hiprand_uniform(&st) >= p1;
+      hiprand_init(seed, (unsigned long long)(i >> 2), offset, &st);
+      float4 u = hiprand_uniform4(&st);
+      float uu = (i & 3) == 0 ? u.x : (i & 3) == 1 ? u.y
+                 : (i & 3) == 2 ? u.z : u.w;
+      keep = uu >= p1;
     }
     pidx[i] = (unsigned char)idx;
     m2[i] = keep;
@@ -404,30 +408,62 @@ __global__ void k_conv2_bwd_w(const float* __restrict__ a1,
 
 // dz1[b,ci,p,q] = relu'(a1) * sum_{co,kh,kw valid} W2[co,ci,kh,kw] *
 //                 dz2[b,co,p-kh,q-kw]
+// Block per (b, quad of ci).  dz2 is consumed in 16-channel LDS tiles
+// (36 KB) with the W2 slice alongside, so the co-reduction runs out of
+// LDS instead of thrashing L1 with 2.3 KB strides.
 __global__ void k_conv2_bwd_x(const float* __restrict__ dz2,
                               const float* __restrict__ w2,
                               const float* __restrict__ a1, int B,
                               float* __restrict__ dz1) {
-  int total = B * 32 * 676;
-  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += gridDim.x * blockDim.x) {
-    int q = i % 26, p = (i / 26) % 26, ci = (i / 676) % 32, b = i / 21632;
-    if (a1[i] <= 0.f) { dz1[i] = 0.f; continue; }
-    float s = 0.f;
+  __shared__ float lds_dz[16 * 576];   // 36 KB
+  __shared__ float lds_w[16 * 4 * 9];  // W2[co-tile, ci-quad, taps]
+  int b = blockIdx.x / 8, ci0 = (blockIdx.x % 8) * 4;
+  float acc[4][3];  // 4 ci x up to 3 outputs per thread (676/256)
+  #pragma unroll
+  for (int c = 0; c < 4; ++c)
     #pragma unroll
-    for (int kh = 0; kh < 3; ++kh) {
-      int y = p - kh;
-      if (y < 0 || y >= 24) continue;
+    for (int r = 0; r < 3; ++r) acc[c][r] = 0.f;
+  for (int cot = 0; cot < 4; ++cot) {
+    __syncthreads();
+    const float* src = dz2 + ((long long)b * 64 + cot * 16) * 576;
+    for (int i = threadIdx.x; i < 16 * 576; i += blockDim.x)
+      lds_dz[i] = src[i];
+    for (int i = threadIdx.x; i < 16 * 4 * 9; i += blockDim.x) {
+      int t = i % 9, c = (i / 9) % 4, co = i / 36;
+      lds_w[i] = w2[((long long)(cot * 16 + co) * 32 + ci0 + c) * 9 + t];
+    }
+    __syncthreads();
+    for (int r = 0; r < 3; ++r) {
+      int o = threadIdx.x + r * FBLK;
+      if (o >= 676) break;
+      int q = o % 26, p = o / 26;
       #pragma unroll
-      for (int kw = 0; kw < 3; ++kw) {
-        int x = q - kw;
-        if (x < 0 || x >= 24) continue;
-        for (int co = 0; co < 64; ++co)
-          s = fmaf(w2[((long long)co * 32 + ci) * 9 + kh * 3 + kw],
-                   dz2[((long long)b * 64 + co) * 576 + y * 24 + x], s);
+      for (int kh = 0; kh < 3; ++kh) {
+        int y = p - kh;
+        if (y < 0 || y >= 24) continue;
+        #pragma unroll
+        for (int kw = 0; kw < 3; ++kw) {
+          int x = q - kw;
+          if (x < 0 || x >= 24) continue;
+          for (int co = 0; co < 16; ++co) {
+            float d = lds_dz[co * 576 + y * 24 + x];
+            #pragma unroll
+            for (int c = 0; c < 4; ++c)
+              acc[c][r] = fmaf(lds_w[(co * 4 + c) * 9 + kh * 3 + kw], d,
+                               acc[c][r]);
+          }
+        }
       }
     }
-    dz1[i] = s;
+  }
+  for (int r = 0; r < 3; ++r) {
+    int o = threadIdx.x + r * FBLK;
+    if (o >= 676) break;
+    #pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      long long i = ((long long)b * 32 + ci0 + c) * 676 + o;
+      dz1[i] = a1[i] > 0.f ? acc[c][r] : 0.f;
+    }
   }
 }
 
@@ -451,7 +487,8 @@ __global__ void k_conv1_bwd_w(const float* __restrict__ x,
         acc[kh * 3 + kw] = fmaf(xp[kh * 28 + kw], d, acc[kh * 3 + kw]);
     accb += d;
   }
-  __shared__ float lds[FBLK / 64 * 10];
+  __shared__ float lds[16 * 10];  // up to 16 waves (1024-thread blocks)
+  int n_waves = blockDim.x / 64;
   int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
   #pragma unroll
   for (int k = 0; k < 9; ++k) {
@@ -465,12 +502,12 @@ __global__ void k_conv1_bwd_w(const float* __restrict__ x,
   __syncthreads();
   if (threadIdx.x < 9) {
     float s = 0.f;
-    for (int w = 0; w < FBLK / 64; ++w) s += lds[w * 10 + threadIdx.x];
+    for (int w = 0; w < n_waves; ++w) s += lds[w * 10 + threadIdx.x];
     dw1[co * 9 + threadIdx.x] = s;
   }
   if (threadIdx.x == 9) {
     float s = 0.f;
-    for (int w = 0; w < FBLK / 64; ++w) s += lds[w * 10 + 9];
+    for (int w = 0; w < n_waves; ++w) s += lds[w * 10 + 9];
     db1[co] = s;
   }
 }
@@ -540,10 +577,9 @@ extern "C" void launch_cnn_epoch(
                        p1, ws.dz2);
     hipLaunchKernelGGL(k_conv2_bwd_w, dim3(64 * 32), dim3(FBLK), 0, s,
                        ws.a1, ws.dz2, B, grads + o.w2, grads + o.b2);
-    hipLaunchKernelGGL(k_conv2_bwd_x, dim3((B * 21632 + FBLK - 1) / FBLK),
-                       dim3(FBLK), 0, s, ws.dz2, params + o.w2, ws.a1, B,
-                       ws.dz1);
-    hipLaunchKernelGGL(k_conv1_bwd_w, dim3(32), dim3(FBLK), 0, s,
+    hipLaunchKernelGGL(k_conv2_bwd_x, dim3(B * 8), dim3(FBLK), 0, s,
+                       ws.dz2, params + o.w2, ws.a1, B, ws.dz1);
+    hipLaunchKernelGGL(k_conv1_bwd_w, dim3(32), dim3(1024), 0, s,
                        ws.xb, ws.dz1, B, grads + o.w1, grads + o.b1);
     // fused clip + sufficient stats + SGD on the whole arena
     hipMemsetAsync(ws.red_acc, 0, 2 * sizeof(double), s);
