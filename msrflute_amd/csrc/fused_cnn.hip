@@ -108,9 +108,10 @@ __global__ void k_conv2_fwd(const float* __restrict__ a1,
     for (int i = threadIdx.x; i < 16 * 676; i += blockDim.x)
       lds[i] = src[i];
     __syncthreads();
+    #pragma unroll
     for (int r = 0; r < 3; ++r) {
       int o = threadIdx.x + r * FBLK;
-      if (o >= 576) break;
+      if (o >= 576) continue;
       int xx = o % 24, yy = o / 24;
       for (int ci = 0; ci < 16; ++ci) {
         const float* ap = lds + ci * 676 + yy * 26 + xx;
@@ -132,6 +133,7 @@ __global__ void k_conv2_fwd(const float* __restrict__ a1,
       }
     }
   }
+  #pragma unroll
   for (int r = 0; r < 3; ++r) {
     int o = threadIdx.x + r * FBLK;
     if (o < 576)
@@ -195,12 +197,16 @@ __global__ void k_fc1_fwd(const float* __restrict__ a2,
   for (int b = 0; b < FC1_BMAX; ++b) acc[b] = 0.f;
   for (int k = threadIdx.x; k < 9216; k += blockDim.x) {
     float w = wp[k];
-    for (int b = 0; b < B; ++b)
-      acc[b] = fmaf(w, a2[(long long)b * 9216 + k], acc[b]);
+    #pragma unroll
+    for (int b = 0; b < FC1_BMAX; ++b)
+      if (b < B)
+        acc[b] = fmaf(w, a2[(long long)b * 9216 + k], acc[b]);
   }
   __shared__ float lds[FBLK / 64][FC1_BMAX];
   int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-  for (int b = 0; b < B; ++b) {
+  #pragma unroll
+  for (int b = 0; b < FC1_BMAX; ++b) {
+    if (b >= B) break;  // uniform across the wave: no divergence
     float s = acc[b];
     for (int d = 32; d > 0; d >>= 1) s += __shfl_down(s, d, 64);
     if (lane == 0) lds[wave][b] = s;
@@ -433,9 +439,10 @@ __global__ void k_conv2_bwd_x(const float* __restrict__ dz2,
       lds_w[i] = w2[((long long)(cot * 16 + co) * 32 + ci0 + c) * 9 + t];
     }
     __syncthreads();
+    #pragma unroll
     for (int r = 0; r < 3; ++r) {
       int o = threadIdx.x + r * FBLK;
-      if (o >= 676) break;
+      if (o >= 676) continue;
       int q = o % 26, p = o / 26;
       #pragma unroll
       for (int kh = 0; kh < 3; ++kh) {
@@ -456,9 +463,10 @@ __global__ void k_conv2_bwd_x(const float* __restrict__ dz2,
       }
     }
   }
+  #pragma unroll
   for (int r = 0; r < 3; ++r) {
     int o = threadIdx.x + r * FBLK;
-    if (o >= 676) break;
+    if (o >= 676) continue;
     #pragma unroll
     for (int c = 0; c < 4; ++c) {
       long long i = ((long long)b * 32 + ci0 + c) * 676 + o;
