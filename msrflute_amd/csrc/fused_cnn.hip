@@ -414,64 +414,33 @@ __global__ void k_conv2_bwd_w(const float* __restrict__ a1,
 
 // dz1[b,ci,p,q] = relu'(a1) * sum_{co,kh,kw valid} W2[co,ci,kh,kw] *
 //                 dz2[b,co,p-kh,q-kw]
-// Block per (b, quad of ci).  dz2 is consumed in 16-channel LDS tiles
-// (36 KB) with the W2 slice alongside, so the co-reduction runs out of
-// LDS instead of thrashing L1 with 2.3 KB strides.
+// Flat one-thread-per-output form: the LDS-tiled variant measured SLOWER
+// (160 blocks underfill the 256-CU chip and serialize on barriers); here
+// W2 (72 KB) and the dz2 rows stream through L2 instead.
 __global__ void k_conv2_bwd_x(const float* __restrict__ dz2,
                               const float* __restrict__ w2,
                               const float* __restrict__ a1, int B,
                               float* __restrict__ dz1) {
-  __shared__ float lds_dz[16 * 576];   // 36 KB
-  __shared__ float lds_w[16 * 4 * 9];  // W2[co-tile, ci-quad, taps]
-  int b = blockIdx.x / 8, ci0 = (blockIdx.x % 8) * 4;
-  float acc[4][3];  // 4 ci x up to 3 outputs per thread (676/256)
-  #pragma unroll
-  for (int c = 0; c < 4; ++c)
+  int total = B * 32 * 676;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    int q = i % 26, p = (i / 26) % 26, ci = (i / 676) % 32, b = i / 21632;
+    if (a1[i] <= 0.f) { dz1[i] = 0.f; continue; }
+    float s = 0.f;
     #pragma unroll
-    for (int r = 0; r < 3; ++r) acc[c][r] = 0.f;
-  for (int cot = 0; cot < 4; ++cot) {
-    __syncthreads();
-    const float* src = dz2 + ((long long)b * 64 + cot * 16) * 576;
-    for (int i = threadIdx.x; i < 16 * 576; i += blockDim.x)
-      lds_dz[i] = src[i];
-    for (int i = threadIdx.x; i < 16 * 4 * 9; i += blockDim.x) {
-      int t = i % 9, c = (i / 9) % 4, co = i / 36;
-      lds_w[i] = w2[((long long)(cot * 16 + co) * 32 + ci0 + c) * 9 + t];
-    }
-    __syncthreads();
-    #pragma unroll
-    for (int r = 0; r < 3; ++r) {
-      int o = threadIdx.x + r * FBLK;
-      if (o >= 676) continue;
-      int q = o % 26, p = o / 26;
+    for (int kh = 0; kh < 3; ++kh) {
+      int y = p - kh;
+      if (y < 0 || y >= 24) continue;
       #pragma unroll
-      for (int kh = 0; kh < 3; ++kh) {
-        int y = p - kh;
-        if (y < 0 || y >= 24) continue;
-        #pragma unroll
-        for (int kw = 0; kw < 3; ++kw) {
-          int x = q - kw;
-          if (x < 0 || x >= 24) continue;
-          for (int co = 0; co < 16; ++co) {
-            float d = lds_dz[co * 576 + y * 24 + x];
-            #pragma unroll
-            for (int c = 0; c < 4; ++c)
-              acc[c][r] = fmaf(lds_w[(co * 4 + c) * 9 + kh * 3 + kw], d,
-                               acc[c][r]);
-          }
-        }
+      for (int kw = 0; kw < 3; ++kw) {
+        int x = q - kw;
+        if (x < 0 || x >= 24) continue;
+        for (int co = 0; co < 64; ++co)
+          s = fmaf(w2[((long long)co * 32 + ci) * 9 + kh * 3 + kw],
+                   dz2[((long long)b * 64 + co) * 576 + y * 24 + x], s);
       }
     }
-  }
-  #pragma unroll
-  for (int r = 0; r < 3; ++r) {
-    int o = threadIdx.x + r * FBLK;
-    if (o >= 676) continue;
-    #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      long long i = ((long long)b * 32 + ci0 + c) * 676 + o;
-      dz1[i] = a1[i] > 0.f ? acc[c][r] : 0.f;
-    }
+    dz1[i] = s;
   }
 }
 
@@ -585,8 +554,9 @@ extern "C" void launch_cnn_epoch(
                        p1, ws.dz2);
     hipLaunchKernelGGL(k_conv2_bwd_w, dim3(64 * 32), dim3(FBLK), 0, s,
                        ws.a1, ws.dz2, B, grads + o.w2, grads + o.b2);
-    hipLaunchKernelGGL(k_conv2_bwd_x, dim3(B * 8), dim3(FBLK), 0, s,
-                       ws.dz2, params + o.w2, ws.a1, B, ws.dz1);
+    hipLaunchKernelGGL(k_conv2_bwd_x, dim3((B * 21632 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.dz2, params + o.w2, ws.a1, B,
+                       ws.dz1);
     hipLaunchKernelGGL(k_conv1_bwd_w, dim3(32), dim3(1024), 0, s,
                        ws.xb, ws.dz1, B, grads + o.w1, grads + o.b1);
     // fused clip + sufficient stats + SGD on the whole arena
