@@ -180,41 +180,29 @@ __global__ void k_pool_drop_fwd(const float* __restrict__ r2, int B,
 }
 
 // z3 = W3 @ a2flat + b3 ; a3 = dropout(relu(z3), p2)
-// one block per output j: the W3 row is read ONCE and dotted against all
-// B sample rows simultaneously (W3 is 4.7 MB — the one-block-per-(b,j)
-// form re-read it B times and thrashed L2).
-#define FC1_BMAX 32
+// one block per (b, j): 256-thread coalesced dot of length 9216.  (A
+// one-block-per-j variant amortizing W3 reads across B measured 3x
+// slower: the per-b a2 reads stride 9216 and kill coalescing; here both
+// streams are contiguous and L2 absorbs the W3 re-reads.)
 __global__ void k_fc1_fwd(const float* __restrict__ a2,
                           const float* __restrict__ w3,
                           const float* __restrict__ b3, int B, float p2,
                           unsigned long long seed, unsigned long long offset,
                           float* __restrict__ z3, float* __restrict__ a3,
                           unsigned char* __restrict__ m3) {
-  int j = blockIdx.x;
+  int b = blockIdx.x / 128, j = blockIdx.x % 128;
+  const float* ap = a2 + (long long)b * 9216;
   const float* wp = w3 + (long long)j * 9216;
-  float acc[FC1_BMAX];
-  #pragma unroll
-  for (int b = 0; b < FC1_BMAX; ++b) acc[b] = 0.f;
-  for (int k = threadIdx.x; k < 9216; k += blockDim.x) {
-    float w = wp[k];
-    #pragma unroll
-    for (int b = 0; b < FC1_BMAX; ++b)
-      if (b < B)
-        acc[b] = fmaf(w, a2[(long long)b * 9216 + k], acc[b]);
-  }
-  __shared__ float lds[FBLK / 64][FC1_BMAX];
-  int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-  #pragma unroll
-  for (int b = 0; b < FC1_BMAX; ++b) {
-    if (b >= B) break;  // uniform across the wave: no divergence
-    float s = acc[b];
-    for (int d = 32; d > 0; d >>= 1) s += __shfl_down(s, d, 64);
-    if (lane == 0) lds[wave][b] = s;
-  }
+  float s = 0.f;
+  for (int k = threadIdx.x; k < 9216; k += blockDim.x)
+    s = fmaf(wp[k], ap[k], s);
+  for (int d = 32; d > 0; d >>= 1) s += __shfl_down(s, d, 64);
+  __shared__ float lds[FBLK / 64];
+  if ((threadIdx.x & 63) == 0) lds[threadIdx.x >> 6] = s;
   __syncthreads();
-  for (int b = threadIdx.x; b < B; b += blockDim.x) {
+  if (threadIdx.x == 0) {
     float t = b3[j];
-    for (int w = 0; w < FBLK / 64; ++w) t += lds[w][b];
+    for (int w = 0; w < FBLK / 64; ++w) t += lds[w];
     int i = b * 128 + j;
     z3[i] = t;
     float r = t > 0.f ? t : 0.f;
@@ -533,7 +521,7 @@ extern "C" void launch_cnn_epoch(
     hipLaunchKernelGGL(k_pool_drop_fwd, dim3((B * 9216 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.r2, B, p1, seed, off, ws.a2,
                        ws.pidx, ws.m2);
-    hipLaunchKernelGGL(k_fc1_fwd, dim3(128), dim3(FBLK), 0, s,
+    hipLaunchKernelGGL(k_fc1_fwd, dim3(B * 128), dim3(FBLK), 0, s,
                        ws.a2, params + o.w3, params + o.b3, B, p2, seed, off,
                        ws.z3, ws.a3, ws.m3);
     hipLaunchKernelGGL(k_fc2_loss_fwd, dim3(B), dim3(FBLK),
