@@ -64,7 +64,7 @@ def build_config(args):
             "seed": 1234,
         },
         "client_config": {
-            "parallel_clients": int(os.environ.get("BENCH_PAR", "4")),
+            "parallel_clients": int(os.environ.get("BENCH_PAR", "8")),
             "use_fused_cnn": os.environ.get("BENCH_FUSED", "1") == "1",
             "do_profiling": False,
             "ignore_subtask": False,

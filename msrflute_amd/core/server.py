@@ -134,7 +134,7 @@ class OptimizationServer:
         # per-rank persistent client workspace; with a GPU and an
         # order-free strategy, a pool of stream-parallel executors trains
         # several clients concurrently (ClientPool)
-        n_par = int(config["client_config"].get("parallel_clients", 4))
+        n_par = int(config["client_config"].get("parallel_clients", 8))
         if (torch.cuda.is_available() and n_par > 1
                 and config["strategy"] in ("FedAvg", "FedProx")
                 and server_config.get("type") != "personalization"
