@@ -170,6 +170,7 @@ class ClientExecutor:
                       for m in self.model.modules())
         self.graph_cache = None
         if (torch.cuda.is_available() and not has_rnn
+                and self.fused_cnn is None  # fused path supersedes graphs
                 and self.client_config.get("use_hip_graphs", True)):
             from ..ops.graphs import GraphCache
             cache = GraphCache(
