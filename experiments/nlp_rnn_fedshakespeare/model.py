@@ -10,6 +10,7 @@ from torch import nn
 from torch.nn import functional as F
 
 from msrflute_amd.core.model import BaseModel
+from msrflute_amd.ops.lstm import FusedLSTM
 from msrflute_amd.utils import to_device
 
 
@@ -18,8 +19,10 @@ class CharLSTM(nn.Module):
         super().__init__()
         self.embeddings = nn.Embedding(vocab_size, embedding_dim,
                                        padding_idx=0)
-        self.lstm = nn.LSTM(embedding_dim, hidden_size, num_layers=2,
-                            batch_first=True)
+        # FusedLSTM: nn.LSTM param names/semantics, gfx950 sequence kernels
+        # on GPU (MIOpen RNN is slow at FL batch sizes and capture-unsafe)
+        self.lstm = FusedLSTM(embedding_dim, hidden_size, num_layers=2,
+                              batch_first=True)
         self.fc = nn.Linear(hidden_size, vocab_size)
 
     def forward(self, x):

@@ -44,6 +44,12 @@ struct CnnWorkspace {
   double *red_partials, *red_acc;
 };
 
+extern "C" void launch_lstm_seq_fwd(const float*, const float*, float*,
+                                    float*, float*, int, int, int,
+                                    hipStream_t);
+extern "C" void launch_lstm_seq_bwd(const float*, const float*, const float*,
+                                    const float*, float*, int, int, int,
+                                    hipStream_t);
 extern "C" void launch_cnn_epoch(
     const float* shard_x, const long long* shard_y, const long long* order,
     long long n, int bs, int C, float* params, float* grads,
@@ -277,6 +283,38 @@ void cnn_epoch(torch::Tensor shard_x, torch::Tensor shard_y,
                    cur_stream());
 }
 
+// fused LSTM sequence recurrence (lstm_seq.hip): forward over all T
+// steps in one launch; returns (h_seq, activated gates, cell states)
+std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xp,
+                                        torch::Tensor w_hh) {
+  check_flat(xp, "xp"); check_flat(w_hh, "w_hh");
+  TORCH_CHECK(xp.dim() == 3 && xp.size(2) == 4 * 256 &&
+              w_hh.size(0) == 4 * 256 && w_hh.size(1) == 256,
+              "fused LSTM supports hidden size 256");
+  long long B = xp.size(0), T = xp.size(1);
+  auto h_seq = torch::empty({B, T, 256}, xp.options());
+  auto gates = torch::empty({B, T, 4 * 256}, xp.options());
+  auto c_seq = torch::empty({B, T, 256}, xp.options());
+  launch_lstm_seq_fwd(xp.data_ptr<float>(), w_hh.data_ptr<float>(),
+                      h_seq.data_ptr<float>(), gates.data_ptr<float>(),
+                      c_seq.data_ptr<float>(), (int)B, (int)T, 256,
+                      cur_stream());
+  return {h_seq, gates, c_seq};
+}
+
+torch::Tensor lstm_seq_bwd(torch::Tensor gates, torch::Tensor c_seq,
+                           torch::Tensor w_hh, torch::Tensor dh_out) {
+  check_flat(gates, "gates"); check_flat(c_seq, "c_seq");
+  check_flat(w_hh, "w_hh"); check_flat(dh_out, "dh_out");
+  long long B = gates.size(0), T = gates.size(1);
+  auto dg = torch::empty_like(gates);
+  launch_lstm_seq_bwd(gates.data_ptr<float>(), c_seq.data_ptr<float>(),
+                      w_hh.data_ptr<float>(), dh_out.data_ptr<float>(),
+                      dg.data_ptr<float>(), (int)B, (int)T, 256,
+                      cur_stream());
+  return dg;
+}
+
 // fused GRU gate math (no-grad eval path of the nlg_gru recurrence)
 torch::Tensor gru_gates(torch::Tensor g_i, torch::Tensor g_h,
                         torch::Tensor h) {
@@ -310,4 +348,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quant_bin_mask", &quant_bin_mask);
   m.def("gru_gates", &gru_gates);
   m.def("cnn_epoch", &cnn_epoch);
+  m.def("lstm_seq_fwd", &lstm_seq_fwd);
+  m.def("lstm_seq_bwd", &lstm_seq_bwd);
 }

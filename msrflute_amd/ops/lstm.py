@@ -1,0 +1,76 @@
+"""Fused LSTM (csrc/lstm_seq.hip) — drop-in for ``nn.LSTM(batch_first=True)``.
+
+Why: MIOpen's RNN path runs a batch-4 T=80 sequence in ~9 ms AND
+segfaults under hipGraph capture, which left the Shakespeare benchmark
+task at reference speed (PERF.md).  Here the input projection for ALL
+timesteps is one hipBLASLt GEMM, the T-step recurrence is ONE kernel
+launch per layer (forward and BPTT backward), and the weight gradients
+are two GEMMs — ~25 graph-capturable nodes per batch.
+
+Parameter names match ``nn.LSTM`` (``weight_ih_l{k}`` …) so checkpoints
+interchange; on CPU (or hidden sizes other than 256) an unregistered
+``nn.LSTM`` sharing the SAME Parameter objects runs the reference path.
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import nn
+
+from . import HAS_EXT, _C
+
+
+class _LSTMSeq(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, xp, w_hh):
+        h_seq, gates, c_seq = _C.lstm_seq_fwd(xp.contiguous(),
+                                              w_hh.contiguous())
+        ctx.save_for_backward(gates, c_seq, w_hh, h_seq)
+        return h_seq
+
+    @staticmethod
+    def backward(ctx, dh):
+        gates, c_seq, w_hh, h_seq = ctx.saved_tensors
+        dg = _C.lstm_seq_bwd(gates, c_seq, w_hh, dh.contiguous())
+        B, T, H = h_seq.shape
+        h_prev = torch.cat([h_seq.new_zeros(B, 1, H), h_seq[:, :-1]], dim=1)
+        # dW_hh = sum_t dgates_t^T h_{t-1} : one GEMM over the stacked steps
+        dw_hh = dg.reshape(-1, 4 * H).t().mm(h_prev.reshape(-1, H))
+        return dg, dw_hh
+
+
+class FusedLSTM(nn.Module):
+    """Unidirectional batch-first LSTM; inter-layer dropout unsupported
+    (the benchmark models use dropout=0)."""
+
+    def __init__(self, input_size, hidden_size, num_layers=1,
+                 batch_first=True):
+        super().__init__()
+        assert batch_first
+        self.input_size = input_size
+        self.hidden_size = hidden_size
+        self.num_layers = num_layers
+        fallback = nn.LSTM(input_size, hidden_size, num_layers=num_layers,
+                           batch_first=True)
+        # register the fallback's Parameters under nn.LSTM's names; keep
+        # the fallback itself unregistered so state_dict stays clean
+        for name, p in list(fallback.named_parameters()):
+            self.register_parameter(name, p)
+        object.__setattr__(self, "_fallback", fallback)
+
+    def _use_fused(self, x):
+        return (x.is_cuda and HAS_EXT and self.hidden_size == 256
+                and x.dtype == torch.float32)
+
+    def forward(self, x, hx=None):
+        if not self._use_fused(x):
+            self._fallback.flatten_parameters = lambda: None  # shared params
+            return self._fallback(x, hx)
+        h = x
+        for k in range(self.num_layers):
+            w_ih = getattr(self, f"weight_ih_l{k}")
+            w_hh = getattr(self, f"weight_hh_l{k}")
+            b = getattr(self, f"bias_ih_l{k}") + getattr(self, f"bias_hh_l{k}")
+            xp = h.matmul(w_ih.t()) + b
+            h = _LSTMSeq.apply(xp, w_hh)
+        return h, None

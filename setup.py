@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "msrflute_amd/csrc/bindings.cpp",
         "msrflute_amd/csrc/flat_ops.hip",
         "msrflute_amd/csrc/fused_cnn.hip",
+        "msrflute_amd/csrc/lstm_seq.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
