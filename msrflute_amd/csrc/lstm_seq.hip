@@ -23,138 +23,121 @@
 #include <hip/hip_runtime.h>
 
 #define LSTM_H 256           // hidden size of the benchmark model
-#define ROWS_PER_WG 4
-#define LSTM_THREADS (ROWS_PER_WG * LSTM_H)
 
 __device__ inline float sigf(float x) { return 1.f / (1.f + __expf(-x)); }
 
+// One 256-thread workgroup per batch row (thread = hidden unit): batch
+// rows run on separate CUs, h_{t-1} is wave-broadcast from LDS, and the
+// forward reads W_hh TRANSPOSED ([H, 4H], passed by the caller) so lanes
+// with consecutive h hit consecutive addresses.
+//
 // xp:      [B, T, 4H]  (pre-activation input projection, gate order ifgo)
-// w_hh:    [4H, H]
+// w_hh_t:  [H, 4H]     (transposed W_hh)
 // h_seq:   [B, T, H]   out
 // gates:   [B, T, 4H]  out (ACTIVATED i,f,g,o — saved for backward)
 // c_seq:   [B, T, H]   out (cell states)
 __global__ void k_lstm_seq_fwd(const float* __restrict__ xp,
-                               const float* __restrict__ w_hh,
+                               const float* __restrict__ w_hh_t,
                                float* __restrict__ h_seq,
                                float* __restrict__ gates,
                                float* __restrict__ c_seq, int B, int T) {
-  __shared__ float h_prev[ROWS_PER_WG][LSTM_H];
-  int bl = threadIdx.x / LSTM_H;      // local batch row
-  int h = threadIdx.x % LSTM_H;       // hidden index
-  int b = blockIdx.x * ROWS_PER_WG + bl;
-  bool active = b < B;
-  h_prev[bl][h] = 0.f;
+  __shared__ float h_prev[LSTM_H];
+  int h = threadIdx.x;
+  int b = blockIdx.x;
+  h_prev[h] = 0.f;
   float c = 0.f;
   __syncthreads();
   for (int t = 0; t < T; ++t) {
-    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    if (active) {
-      const float* xr = xp + ((long long)b * T + t) * 4 * LSTM_H;
-      s0 = xr[h];
-      s1 = xr[LSTM_H + h];
-      s2 = xr[2 * LSTM_H + h];
-      s3 = xr[3 * LSTM_H + h];
-      const float* w0 = w_hh + (long long)h * LSTM_H;
-      const float* w1 = w_hh + (long long)(LSTM_H + h) * LSTM_H;
-      const float* w2 = w_hh + (long long)(2 * LSTM_H + h) * LSTM_H;
-      const float* w3 = w_hh + (long long)(3 * LSTM_H + h) * LSTM_H;
-      for (int k = 0; k < LSTM_H; ++k) {
-        float hv = h_prev[bl][k];   // wave-broadcast LDS read
-        s0 = fmaf(w0[k], hv, s0);
-        s1 = fmaf(w1[k], hv, s1);
-        s2 = fmaf(w2[k], hv, s2);
-        s3 = fmaf(w3[k], hv, s3);
-      }
+    const float* xr = xp + ((long long)b * T + t) * 4 * LSTM_H;
+    float s0 = xr[h];
+    float s1 = xr[LSTM_H + h];
+    float s2 = xr[2 * LSTM_H + h];
+    float s3 = xr[3 * LSTM_H + h];
+    for (int k = 0; k < LSTM_H; ++k) {
+      float hv = h_prev[k];                       // LDS broadcast
+      const float* wr = w_hh_t + (long long)k * 4 * LSTM_H;
+      s0 = fmaf(wr[h], hv, s0);                   // coalesced across lanes
+      s1 = fmaf(wr[LSTM_H + h], hv, s1);
+      s2 = fmaf(wr[2 * LSTM_H + h], hv, s2);
+      s3 = fmaf(wr[3 * LSTM_H + h], hv, s3);
     }
     float i = sigf(s0), f = sigf(s1), g = tanhf(s2), o = sigf(s3);
     c = f * c + i * g;
     float hn = o * tanhf(c);
-    if (active) {
-      long long base = ((long long)b * T + t) * 4 * LSTM_H;
-      gates[base + h] = i;
-      gates[base + LSTM_H + h] = f;
-      gates[base + 2 * LSTM_H + h] = g;
-      gates[base + 3 * LSTM_H + h] = o;
-      c_seq[((long long)b * T + t) * LSTM_H + h] = c;
-      h_seq[((long long)b * T + t) * LSTM_H + h] = hn;
-    }
+    long long base = ((long long)b * T + t) * 4 * LSTM_H;
+    gates[base + h] = i;
+    gates[base + LSTM_H + h] = f;
+    gates[base + 2 * LSTM_H + h] = g;
+    gates[base + 3 * LSTM_H + h] = o;
+    c_seq[((long long)b * T + t) * LSTM_H + h] = c;
+    h_seq[((long long)b * T + t) * LSTM_H + h] = hn;
     __syncthreads();
-    h_prev[bl][h] = hn;
+    h_prev[h] = hn;
     __syncthreads();
   }
 }
 
 // Backward: dh_out [B,T,H] upstream grads of h_seq; produces
 // dg_pre [B,T,4H] (pre-activation gate grads == grads of xp).
+// W_hh in its ORIGINAL [4H, H] layout: w_hh[j*H + h] is coalesced across
+// lanes (consecutive h).
 __global__ void k_lstm_seq_bwd(const float* __restrict__ gates,
                                const float* __restrict__ c_seq,
                                const float* __restrict__ w_hh,
                                const float* __restrict__ dh_out,
                                float* __restrict__ dg_pre, int B, int T) {
-  __shared__ float dg_l[ROWS_PER_WG][4 * LSTM_H];  // this step's dg_pre
-  int bl = threadIdx.x / LSTM_H;
-  int h = threadIdx.x % LSTM_H;
-  int b = blockIdx.x * ROWS_PER_WG + bl;
-  bool active = b < B;
+  __shared__ float dg_l[4 * LSTM_H];
+  int h = threadIdx.x;
+  int b = blockIdx.x;
   float dc = 0.f;
-  float dh_rec = 0.f;  // recurrent part of dh (from step t+1)
+  float dh_rec = 0.f;
   for (int t = T - 1; t >= 0; --t) {
     long long base = ((long long)b * T + t) * 4 * LSTM_H;
     long long cbase = ((long long)b * T + t) * LSTM_H;
-    float di = 0.f, df = 0.f, dg = 0.f, do_ = 0.f;
-    if (active) {
-      float i = gates[base + h];
-      float f = gates[base + LSTM_H + h];
-      float g = gates[base + 2 * LSTM_H + h];
-      float o = gates[base + 3 * LSTM_H + h];
-      float ct = c_seq[cbase + h];
-      float cprev = (t > 0) ? c_seq[cbase - LSTM_H + h] : 0.f;
-      float tc = tanhf(ct);
-      float dh = dh_out[cbase + h] + dh_rec;
-      do_ = dh * tc * o * (1.f - o);
-      dc = dc + dh * o * (1.f - tc * tc);
-      di = dc * g * i * (1.f - i);
-      df = dc * cprev * f * (1.f - f);
-      dg = dc * i * (1.f - g * g);
-      dc = dc * f;  // becomes next (t-1) step's carried dc
-    }
+    float i = gates[base + h];
+    float f = gates[base + LSTM_H + h];
+    float g = gates[base + 2 * LSTM_H + h];
+    float o = gates[base + 3 * LSTM_H + h];
+    float ct = c_seq[cbase + h];
+    float cprev = (t > 0) ? c_seq[cbase - LSTM_H + h] : 0.f;
+    float tc = tanhf(ct);
+    float dh = dh_out[cbase + h] + dh_rec;
+    float do_ = dh * tc * o * (1.f - o);
+    dc = dc + dh * o * (1.f - tc * tc);
+    float di = dc * g * i * (1.f - i);
+    float df = dc * cprev * f * (1.f - f);
+    float dg = dc * i * (1.f - g * g);
+    dc = dc * f;
     __syncthreads();
-    dg_l[bl][h] = di;
-    dg_l[bl][LSTM_H + h] = df;
-    dg_l[bl][2 * LSTM_H + h] = dg;
-    dg_l[bl][3 * LSTM_H + h] = do_;
+    dg_l[h] = di;
+    dg_l[LSTM_H + h] = df;
+    dg_l[2 * LSTM_H + h] = dg;
+    dg_l[3 * LSTM_H + h] = do_;
     __syncthreads();
-    if (active) {
-      dg_pre[base + h] = di;
-      dg_pre[base + LSTM_H + h] = df;
-      dg_pre[base + 2 * LSTM_H + h] = dg;
-      dg_pre[base + 3 * LSTM_H + h] = do_;
-      // dh_{t-1} = W_hh^T @ dg_pre_t : column h of W_hh, rows 0..4H
-      float s = 0.f;
-      for (int j = 0; j < 4 * LSTM_H; ++j)
-        s = fmaf(w_hh[(long long)j * LSTM_H + h], dg_l[bl][j], s);
-      dh_rec = s;
-    }
-    __syncthreads();
+    dg_pre[base + h] = di;
+    dg_pre[base + LSTM_H + h] = df;
+    dg_pre[base + 2 * LSTM_H + h] = dg;
+    dg_pre[base + 3 * LSTM_H + h] = do_;
+    float s = 0.f;
+    for (int j = 0; j < 4 * LSTM_H; ++j)
+      s = fmaf(w_hh[(long long)j * LSTM_H + h], dg_l[j], s);
+    dh_rec = s;
   }
 }
 
 extern "C" {
 
-void launch_lstm_seq_fwd(const float* xp, const float* w_hh, float* h_seq,
+void launch_lstm_seq_fwd(const float* xp, const float* w_hh_t, float* h_seq,
                          float* gates, float* c_seq, int B, int T, int H,
                          hipStream_t s) {
-  // H is compile-time LSTM_H for this model family
-  int grid = (B + ROWS_PER_WG - 1) / ROWS_PER_WG;
-  hipLaunchKernelGGL(k_lstm_seq_fwd, dim3(grid), dim3(LSTM_THREADS), 0, s,
-                     xp, w_hh, h_seq, gates, c_seq, B, T);
+  hipLaunchKernelGGL(k_lstm_seq_fwd, dim3(B), dim3(LSTM_H), 0, s,
+                     xp, w_hh_t, h_seq, gates, c_seq, B, T);
 }
 
 void launch_lstm_seq_bwd(const float* gates, const float* c_seq,
                          const float* w_hh, const float* dh_out,
                          float* dg_pre, int B, int T, int H, hipStream_t s) {
-  int grid = (B + ROWS_PER_WG - 1) / ROWS_PER_WG;
-  hipLaunchKernelGGL(k_lstm_seq_bwd, dim3(grid), dim3(LSTM_THREADS), 0, s,
+  hipLaunchKernelGGL(k_lstm_seq_bwd, dim3(B), dim3(LSTM_H), 0, s,
                      gates, c_seq, w_hh, dh_out, dg_pre, B, T);
 }
 

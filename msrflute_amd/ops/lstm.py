@@ -23,8 +23,9 @@ from . import HAS_EXT, _C
 class _LSTMSeq(torch.autograd.Function):
     @staticmethod
     def forward(ctx, xp, w_hh):
-        h_seq, gates, c_seq = _C.lstm_seq_fwd(xp.contiguous(),
-                                              w_hh.contiguous())
+        # forward wants W_hh transposed for coalesced lane access
+        h_seq, gates, c_seq = _C.lstm_seq_fwd(
+            xp.contiguous(), w_hh.t().contiguous())
         ctx.save_for_backward(gates, c_seq, w_hh, h_seq)
         return h_seq
 
