@@ -76,12 +76,23 @@ class FedRuntime:
     # -- collectives (no-ops at world_size == 1) --------------------------
     def all_reduce_(self, t: torch.Tensor):
         if self.size > 1:
-            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+            if self.backend == "gloo" and t.is_cuda:
+                # gloo cannot reduce device tensors: CPU round-trip
+                host = t.detach().cpu()
+                dist.all_reduce(host, op=dist.ReduceOp.SUM)
+                t.copy_(host)
+            else:
+                dist.all_reduce(t, op=dist.ReduceOp.SUM)
         return t
 
     def broadcast_(self, t: torch.Tensor, src: int = 0):
         if self.size > 1:
-            dist.broadcast(t, src=src)
+            if self.backend == "gloo" and t.is_cuda:
+                host = t.detach().cpu()
+                dist.broadcast(host, src=src)
+                t.copy_(host)
+            else:
+                dist.broadcast(t, src=src)
         return t
 
     def all_gather_object(self, obj: Any) -> List[Any]:

@@ -108,3 +108,18 @@ def test_privacy_metrics_indices_and_leakage(tmp_path):
         "attacker_optimizer_config": {"type": "adamax", "lr": 0.03,
                                       "amsgrad": False}}
     _run_task("nlg_gru", cfg, tmp_path, data_dir)
+
+
+def test_server_replay_training(tmp_path):
+    """Server-side replay training after aggregation (reference:
+    server.py:130-151, 430-442)."""
+    data_dir = str(tmp_path / "data")
+    _make_data("cv_lr_mnist", data_dir)
+    cfg = _base_cfg()
+    cfg["server_config"]["data_config"]["train"] = {
+        "batch_size": 16, "train_data_server": "cv_lr_mnist/train_data.pt",
+        "desired_max_samples": 200, "max_grad_norm": 10.0}
+    cfg["server_config"]["server_replay_config"] = {
+        "server_iterations": 2,
+        "optimizer_config": {"type": "sgd", "lr": 0.005}}
+    _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
