@@ -114,14 +114,24 @@ class Evaluation:
 
         local_results = []
         model = self.worker_trainer.model
-        for group in my_groups:
+        if not hasattr(self, "_dl_cache"):
+            self._dl_cache = {}
+        for gi, group in enumerate(my_groups):
             data_strct = Client.get_data(group.client_id, dataset)[0]
-            if mode == "val":
-                dataloader = make_val_dataloader(data_config, data_path=None,
-                                                 task=task, data_strct=data_strct)
-            else:
-                dataloader = make_test_dataloader(data_config, data_path=None,
-                                                  task=task, data_strct=data_strct)
+            key = (mode, gi)
+            dataloader = self._dl_cache.get(key)
+            if dataloader is None:
+                if mode == "val":
+                    dataloader = make_val_dataloader(
+                        data_config, data_path=None, task=task,
+                        data_strct=data_strct)
+                else:
+                    dataloader = make_test_dataloader(
+                        data_config, data_path=None, task=task,
+                        data_strct=data_strct)
+                if hasattr(dataloader, "to_device"):
+                    dataloader.to_device()
+                self._dl_cache[key] = dataloader  # eval sets are static
             want_logits = data_config.get("wantLogits", False)
             output, metrics = run_validation_generic(model, dataloader)
             count = sum(data_strct["num_samples"])
