@@ -35,6 +35,13 @@ class GRUCellSeq(nn.Module):
     def forward(self, emb: Tensor) -> Tuple[Tensor, Tensor]:
         B, T, _ = emb.shape
         g_i = self.w_ih(emb)  # one GEMM for every step: [B, T, 3H]
+        from msrflute_amd.ops.lstm import fused_gru_available, fused_gru_seq
+        if T > 0 and fused_gru_available(self.hidden_size, emb):
+            # whole recurrence = ONE kernel launch (csrc/gru_seq.hip)
+            h_seq = fused_gru_seq(g_i, self.w_hh.weight, self.w_hh.bias)
+            hiddens = torch.cat(
+                [emb.new_zeros(B, 1, self.hidden_size), h_seq], dim=1)
+            return hiddens, h_seq[:, -1]
         h = emb.new_zeros(B, self.hidden_size)
         hiddens = [h]
         for t in range(T):

@@ -50,6 +50,12 @@ extern "C" void launch_lstm_seq_fwd(const float*, const float*, float*,
 extern "C" void launch_lstm_seq_bwd(const float*, const float*, const float*,
                                     const float*, float*, int, int, int,
                                     hipStream_t);
+extern "C" void launch_gru_seq_fwd(const float*, const float*, const float*,
+                                   float*, float*, float*, int, int,
+                                   hipStream_t);
+extern "C" void launch_gru_seq_bwd(const float*, const float*, const float*,
+                                   const float*, const float*, float*,
+                                   float*, int, int, hipStream_t);
 extern "C" void launch_cnn_epoch(
     const float* shard_x, const long long* shard_y, const long long* order,
     long long n, int bs, int C, float* params, float* grads,
@@ -315,6 +321,41 @@ torch::Tensor lstm_seq_bwd(torch::Tensor gates, torch::Tensor c_seq,
   return dg;
 }
 
+// fused GRU sequence recurrence (gru_seq.hip)
+std::vector<torch::Tensor> gru_seq_fwd(torch::Tensor gi, torch::Tensor whh_t,
+                                       torch::Tensor b_hh) {
+  check_flat(gi, "gi"); check_flat(whh_t, "whh_t"); check_flat(b_hh, "b_hh");
+  TORCH_CHECK(gi.dim() == 3 && gi.size(2) == 3 * 512 &&
+              whh_t.size(0) == 512 && whh_t.size(1) == 3 * 512,
+              "fused GRU supports hidden size 512; pass W_hh transposed");
+  long long B = gi.size(0), T = gi.size(1);
+  auto h_seq = torch::empty({B, T, 512}, gi.options());
+  auto gates = torch::empty({B, T, 3 * 512}, gi.options());
+  auto ghn = torch::empty({B, T, 512}, gi.options());
+  launch_gru_seq_fwd(gi.data_ptr<float>(), whh_t.data_ptr<float>(),
+                     b_hh.data_ptr<float>(), h_seq.data_ptr<float>(),
+                     gates.data_ptr<float>(), ghn.data_ptr<float>(),
+                     (int)B, (int)T, cur_stream());
+  return {h_seq, gates, ghn};
+}
+
+std::vector<torch::Tensor> gru_seq_bwd(torch::Tensor gates, torch::Tensor ghn,
+                                       torch::Tensor h_seq,
+                                       torch::Tensor w_hh,
+                                       torch::Tensor dh_out) {
+  check_flat(gates, "gates"); check_flat(ghn, "ghn");
+  check_flat(h_seq, "h_seq"); check_flat(w_hh, "w_hh");
+  check_flat(dh_out, "dh_out");
+  long long B = gates.size(0), T = gates.size(1);
+  auto dgi = torch::empty_like(gates);
+  auto dgh = torch::empty_like(gates);
+  launch_gru_seq_bwd(gates.data_ptr<float>(), ghn.data_ptr<float>(),
+                     h_seq.data_ptr<float>(), w_hh.data_ptr<float>(),
+                     dh_out.data_ptr<float>(), dgi.data_ptr<float>(),
+                     dgh.data_ptr<float>(), (int)B, (int)T, cur_stream());
+  return {dgi, dgh};
+}
+
 // fused GRU gate math (no-grad eval path of the nlg_gru recurrence)
 torch::Tensor gru_gates(torch::Tensor g_i, torch::Tensor g_h,
                         torch::Tensor h) {
@@ -349,5 +390,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gru_gates", &gru_gates);
   m.def("cnn_epoch", &cnn_epoch);
   m.def("lstm_seq_fwd", &lstm_seq_fwd);
+  m.def("gru_seq_fwd", &gru_seq_fwd);
+  m.def("gru_seq_bwd", &gru_seq_bwd);
   m.def("lstm_seq_bwd", &lstm_seq_bwd);
 }

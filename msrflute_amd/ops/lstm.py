@@ -75,3 +75,34 @@ class FusedLSTM(nn.Module):
             xp = h.matmul(w_ih.t()) + b
             h = _LSTMSeq.apply(xp, w_hh)
         return h, None
+
+
+class _GRUSeq(torch.autograd.Function):
+    """Fused GRU sequence (csrc/gru_seq.hip) — reference cell semantics
+    (experiments/nlg_gru/model.py GRU2)."""
+
+    @staticmethod
+    def forward(ctx, gi, w_hh, b_hh):
+        h_seq, gates, ghn = _C.gru_seq_fwd(
+            gi.contiguous(), w_hh.t().contiguous(), b_hh.contiguous())
+        ctx.save_for_backward(gates, ghn, h_seq, w_hh)
+        return h_seq
+
+    @staticmethod
+    def backward(ctx, dh):
+        gates, ghn, h_seq, w_hh = ctx.saved_tensors
+        dgi, dgh = _C.gru_seq_bwd(gates, ghn, h_seq, w_hh, dh.contiguous())
+        B, T, H = h_seq.shape
+        h_prev = torch.cat([h_seq.new_zeros(B, 1, H), h_seq[:, :-1]], dim=1)
+        dw_hh = dgh.reshape(-1, 3 * H).t().mm(h_prev.reshape(-1, H))
+        db_hh = dgh.sum(dim=(0, 1))
+        return dgi, dw_hh, db_hh
+
+
+def fused_gru_available(hidden_size, x):
+    return HAS_EXT and x.is_cuda and hidden_size == 512 \
+        and x.dtype == torch.float32
+
+
+def fused_gru_seq(gi, w_hh_weight, w_hh_bias):
+    return _GRUSeq.apply(gi, w_hh_weight, w_hh_bias)

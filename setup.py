@@ -23,6 +23,7 @@ ext = CUDAExtension(
         "msrflute_amd/csrc/flat_ops.hip",
         "msrflute_amd/csrc/fused_cnn.hip",
         "msrflute_amd/csrc/lstm_seq.hip",
+        "msrflute_amd/csrc/gru_seq.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
