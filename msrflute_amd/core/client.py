@@ -15,11 +15,10 @@ Two pieces:
 
 from __future__ import annotations
 
-import copy
 import logging
 import os
 import time
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 import torch
 
@@ -30,10 +29,8 @@ from ..ops.fused_optim import make_arena_optimizer
 from ..strategies import select_strategy
 from ..utils import (ScheduledSamplingScheduler, alpha_update, make_optimizer,
                      print_rank, to_device)
-from ..utils.dataloaders_utils import (get_dataset, make_test_dataloader,
-                                       make_train_dataloader,
-                                       make_val_dataloader)
-from .trainer import Trainer, run_validation_generic, set_component_wise_lr
+from ..utils.dataloaders_utils import get_dataset, make_train_dataloader
+from .trainer import Trainer, set_component_wise_lr
 
 # Worker-wide dataset cache (reference: client.py:45-47, 76-99)
 train_dataset = None

@@ -35,7 +35,6 @@ import yaml
 
 from .. import ops
 from ..ops.arena import ParameterArena
-from ..ops.fused_optim import make_arena_optimizer
 from ..utils import (make_lr_scheduler, make_optimizer, print_rank, to_device)
 from .metrics import Metrics
 
