@@ -138,6 +138,8 @@ class OptimizationServer:
         if (torch.cuda.is_available() and n_par > 1
                 and config["strategy"] in ("FedAvg", "FedProx")
                 and server_config.get("type") != "personalization"
+                and server_config.get("fast_aggregation", True)
+                and not config.get("dump_norm_stats", False)
                 and not config.get("privacy_metrics_config", {}).get(
                     "apply_metrics", False)):
             from .client import ClientPool
