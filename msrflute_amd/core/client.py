@@ -569,3 +569,130 @@ class ClientPool:
             ops.axpy(target_grad, acc, 1.0)
             acc.zero_()
         self._streams_dirty = False
+
+
+def _fused_round_impl(ex, client_ids, initial_lr, seeds):
+    """One _C.cnn_round call training all of ``ex``'s clients for this
+    round (copy-in, fused epoch, weighted pseudo-grad, accumulate into
+    ``ex.round_accum_target``).  Returns outputs or None if ineligible."""
+    if ex.fused_cnn is None:
+        return None
+    data_cfg = ex.client_config["data_config"]["train"]
+    store = ex._get_shard_store(data_cfg)
+    if store is None or store.x[0].numel() != 784:
+        return None
+    import msrflute_amd.core.client as cm
+    ds = cm.train_dataset
+    if ds is None:
+        return None
+    fc = ex.fused_cnn
+    counts, row_bases, order_offs, orders, w_list = [], [], [], [], []
+    off = 0
+    for cid, seed in zip(client_ids, seeds):
+        user = ds.user_list[cid]
+        i = store.user_pos.get(user)
+        if i is None:
+            return None
+        lo, hi = store.offsets[i], store.offsets[i + 1]
+        n = hi - lo
+        if n == 0:
+            return None
+        torch.manual_seed(seed & 0x7FFFFFFFFFFF)
+        orders.append(torch.randperm(n))
+        counts.append(n)
+        row_bases.append(lo)
+        order_offs.append(off)
+        off += n
+        w_list.append(float(n))  # FedAvg weight = num_samples
+    K = len(client_ids)
+    orders_cat = torch.cat(orders) if orders else torch.empty(0, dtype=torch.int64)
+    if getattr(ex, "_order_pin", None) is None or ex._order_pin.numel() < off:
+        ex._order_pin = torch.empty(max(off, 1024),
+                                    dtype=torch.int64).pin_memory()
+    ex._order_pin[:off].copy_(orders_cat)
+    orders_dev = ex._order_pin[:off].to(ex.arena.device, non_blocking=True)
+    if getattr(ex, "_round_stats", None) is None or \
+            ex._round_stats.numel() < 2 * K:
+        ex._round_stats = torch.zeros(2 * max(K, 8), device=ex.arena.device)
+        ex._round_loss = torch.zeros(max(K, 8), device=ex.arena.device)
+    ex._round_stats[: 2 * K].zero_()
+    ex._round_loss[:K].zero_()
+    fc.lr_t.fill_(float(initial_lr))
+    ops_mod = ops
+    _ = ops_mod  # (ops imported at module top)
+    from msrflute_amd import _C
+    _C.cnn_round(
+        store.x.reshape(len(store.y), -1), store.y, orders_dev,
+        torch.tensor(row_bases, dtype=torch.int64),
+        torch.tensor(order_offs, dtype=torch.int64),
+        torch.tensor(counts, dtype=torch.int64),
+        torch.tensor(w_list, dtype=torch.float32),
+        torch.tensor([s & 0x7FFFFFFFFFFF for s in seeds], dtype=torch.int64),
+        fc.bs, fc.C, ex.server_arena.data, ex.arena.data, ex.arena.grad,
+        ex.round_accum_target, fc.work_f, fc.work_i, fc.work_b, fc.work_d,
+        fc.lr_t, fc.max_norm, fc.p1, fc.p2,
+        ex._round_stats, ex._round_loss)
+    outputs = []
+    now = time.time()
+    for k, cid in enumerate(client_ids):
+        n_batches = (counts[k] + fc.bs - 1) // fc.bs
+        outputs.append((cid, {
+            "cs": {"setup": 0.0, "training": 0.0, "full cost": 0.0,
+                   "dataloader": 0.0},
+            "ns": counts[k],
+            "pl": {"weight": w_list[k], "grad": None, "pooled": True},
+            "_lazy": (ex._round_loss[k].reshape(()),
+                      ex._round_stats[2 * k: 2 * k + 2],
+                      n_batches * ex.arena.total),
+            "ts": now,
+        }))
+        ex.perf_acc["clients"] = ex.perf_acc.get("clients", 0) + 1
+    return outputs
+
+
+def _pool_run_fused_round_batch(self, client_ids, initial_lr, iteration,
+                                seeds):
+    """ClientPool: one _C.cnn_round per executor (on its stream) covering
+    its chunk of the round's clients.  Returns outputs or None if the
+    fused path is ineligible (caller falls back to the per-client path).
+    Eligibility is validated UP FRONT so no partial state is mutated on
+    the fallback path."""
+    prim = self.executors[0]
+    if prim.fused_cnn is None:
+        return None
+    store = prim._get_shard_store(
+        prim.client_config["data_config"]["train"])
+    if store is None or store.x[0].numel() != 784:
+        return None
+    import msrflute_amd.core.client as cm
+    ds = cm.train_dataset
+    if ds is None:
+        return None
+    for cid in client_ids:
+        i = store.user_pos.get(ds.user_list[cid])
+        if i is None or store.offsets[i + 1] - store.offsets[i] == 0:
+            return None
+    for ex in self.executors[1:]:
+        if not ex._shard_store_tried:
+            ex._shard_store = store
+            ex._shard_store_tried = True
+
+    P = len(self.executors)
+    chunks = [client_ids[k::P] for k in range(P)]
+    seed_chunks = [seeds[k::P] for k in range(P)]
+    outputs = []
+    for k, (chunk, schunk) in enumerate(zip(chunks, seed_chunks)):
+        if not chunk:
+            continue
+        ex, st = self.executors[k], self.streams[k]
+        ex.round_accum_target = self.round_accums[k]
+        st.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(st):
+            out = _fused_round_impl(ex, chunk, initial_lr, schunk)
+        assert out is not None  # pre-validated above
+        outputs.extend(out)
+        self._streams_dirty = True
+    return outputs
+
+
+ClientPool.run_fused_round_batch = _pool_run_fused_round_batch

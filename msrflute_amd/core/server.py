@@ -297,12 +297,31 @@ class OptimizationServer:
                 self.run_stats[key].append([])
 
             # ---- local client training ----------------------------------
+            # whole-round fused path: ONE _C.cnn_round call per executor
+            # trains every local client (copy-in, epoch, pseudo-grad,
+            # accumulate all inside the extension)
+            fused_outputs = None
+            if (hasattr(self.executor, "run_fused_round_batch")
+                    and self.config["client_config"].get("use_fused_round",
+                                                         True)
+                    and not apply_privacy_metrics):
+                seeds = [rt.round_rng(i, salt=100 + c).getrandbits(62)
+                         for c in my_clients]
+                fused_outputs = self.executor.run_fused_round_batch(
+                    my_clients, initial_lr, i, seeds)
+
             local_outputs = []
-            for client_idx in my_clients:
-                client = Client([client_idx], self.config, True)
-                output = self.executor.process_round(
-                    client, initial_lr, i,
-                    round_seed=rt.round_rng(i, salt=100 + client_idx).getrandbits(62))
+            for client_idx, pre_output in (
+                    fused_outputs if fused_outputs is not None
+                    else ((c, None) for c in my_clients)):
+                if pre_output is not None:
+                    output = pre_output
+                else:
+                    client = Client([client_idx], self.config, True)
+                    output = self.executor.process_round(
+                        client, initial_lr, i,
+                        round_seed=rt.round_rng(
+                            i, salt=100 + client_idx).getrandbits(62))
                 payload = output["pl"]
                 if output.get("wt", None) == 0.0 and payload is not None:
                     payload["weight"] = 0.0

@@ -61,7 +61,15 @@ extern "C" void launch_cnn_epoch(
     long long n, int bs, int C, float* params, float* grads,
     CnnWorkspace ws, const float* lr_t, float max_norm, float p1, float p2,
     float* stats_acc, float* loss_acc, unsigned long long seed,
-    hipStream_t s);
+    hipStream_t s, long long row_base);
+extern "C" void launch_cnn_round(
+    const float* shard_x, const long long* shard_y, const long long* orders,
+    const long long* row_bases, const long long* order_offs,
+    const long long* counts, const float* weights,
+    const unsigned long long* seeds, int K, int bs, int C,
+    const float* server_params, float* params, float* grads,
+    float* round_accum, CnnWorkspace ws, const float* lr_t, float max_norm,
+    float p1, float p2, float* stats_out, float* loss_out, hipStream_t s);
 
 namespace {
 
@@ -286,7 +294,73 @@ void cnn_epoch(torch::Tensor shard_x, torch::Tensor shard_y,
                    lr_t.data_ptr<float>(), (float)max_norm, (float)p1,
                    (float)p2, stats_acc.data_ptr<float>(),
                    loss_acc.data_ptr<float>(), (unsigned long long)seed,
-                   cur_stream());
+                   cur_stream(), 0);
+}
+
+// helper shared with cnn_epoch: slice the reusable workspace buffers
+static CnnWorkspace slice_ws(torch::Tensor& work_f, torch::Tensor& work_i,
+                             torch::Tensor& work_b, torch::Tensor& work_d,
+                             int B, int C) {
+  float* f = work_f.data_ptr<float>();
+  CnnWorkspace ws;
+  ws.xb = f;            f += (long long)B * 784;
+  ws.a1 = f;            f += (long long)B * 21632;
+  ws.r2 = f;            f += (long long)B * 36864;
+  ws.a2 = f;            f += (long long)B * 9216;
+  ws.z3 = f;            f += (long long)B * 128;
+  ws.a3 = f;            f += (long long)B * 128;
+  ws.dlogits = f;       f += (long long)B * C;
+  ws.dz3 = f;           f += (long long)B * 128;
+  ws.da2 = f;           f += (long long)B * 9216;
+  ws.dz2 = f;           f += (long long)B * 36864;
+  ws.dz1 = f;           f += (long long)B * 21632;
+  TORCH_CHECK(f - work_f.data_ptr<float>() <= work_f.numel());
+  ws.yb = work_i.data_ptr<int>();
+  unsigned char* u = work_b.data_ptr<unsigned char>();
+  ws.pidx = u;          u += (long long)B * 9216;
+  ws.m2 = u;            u += (long long)B * 9216;
+  ws.m3 = u;            u += (long long)B * 128;
+  ws.red_partials = work_d.data_ptr<double>();
+  ws.red_acc = work_d.data_ptr<double>() + work_d.numel() - 2;
+  return ws;
+}
+
+// whole-round driver: K clients in ONE host call (copy-in, fused epoch,
+// weighted pseudo-gradient, accumulate) — per-client loss/stats in slots
+void cnn_round(torch::Tensor shard_x, torch::Tensor shard_y,
+               torch::Tensor orders_dev, torch::Tensor row_bases,
+               torch::Tensor order_offs, torch::Tensor counts,
+               torch::Tensor weights, torch::Tensor seeds, int64_t bs,
+               int64_t C, torch::Tensor server_params, torch::Tensor params,
+               torch::Tensor grads, torch::Tensor round_accum,
+               torch::Tensor work_f, torch::Tensor work_i,
+               torch::Tensor work_b, torch::Tensor work_d,
+               torch::Tensor lr_t, double max_norm, double p1, double p2,
+               torch::Tensor stats_out, torch::Tensor loss_out) {
+  check_flat(server_params, "server_params"); check_flat(params, "params");
+  check_flat(grads, "grads"); check_flat(round_accum, "round_accum");
+  TORCH_CHECK(orders_dev.is_cuda() && orders_dev.scalar_type() == torch::kInt64);
+  TORCH_CHECK(!row_bases.is_cuda() && !order_offs.is_cuda() &&
+              !counts.is_cuda() && !weights.is_cuda() && !seeds.is_cuda(),
+              "per-client metadata must be host tensors");
+  int K = (int)counts.numel();
+  TORCH_CHECK(stats_out.numel() >= 2 * K && loss_out.numel() >= K);
+  TORCH_CHECK(bs >= 1 && bs <= 32);
+  CnnWorkspace ws = slice_ws(work_f, work_i, work_b, work_d, (int)bs, (int)C);
+  launch_cnn_round(
+      shard_x.data_ptr<float>(),
+      reinterpret_cast<const long long*>(shard_y.data_ptr<int64_t>()),
+      reinterpret_cast<const long long*>(orders_dev.data_ptr<int64_t>()),
+      reinterpret_cast<const long long*>(row_bases.data_ptr<int64_t>()),
+      reinterpret_cast<const long long*>(order_offs.data_ptr<int64_t>()),
+      reinterpret_cast<const long long*>(counts.data_ptr<int64_t>()),
+      weights.data_ptr<float>(),
+      reinterpret_cast<const unsigned long long*>(seeds.data_ptr<int64_t>()),
+      K, (int)bs, (int)C, server_params.data_ptr<float>(),
+      params.data_ptr<float>(), grads.data_ptr<float>(),
+      round_accum.data_ptr<float>(), ws, lr_t.data_ptr<float>(),
+      (float)max_norm, (float)p1, (float)p2, stats_out.data_ptr<float>(),
+      loss_out.data_ptr<float>(), cur_stream());
 }
 
 // fused LSTM sequence recurrence (lstm_seq.hip): forward over all T
@@ -389,6 +463,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quant_bin_mask", &quant_bin_mask);
   m.def("gru_gates", &gru_gates);
   m.def("cnn_epoch", &cnn_epoch);
+  m.def("cnn_round", &cnn_round);
   m.def("lstm_seq_fwd", &lstm_seq_fwd);
   m.def("gru_seq_fwd", &gru_seq_fwd);
   m.def("gru_seq_bwd", &gru_seq_bwd);

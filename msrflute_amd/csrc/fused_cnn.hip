@@ -47,16 +47,40 @@ static CnnOffsets cnn_offsets(int C) {
 __global__ void k_gather_batch(const float* __restrict__ shard_x,
                                const long long* __restrict__ shard_y,
                                const long long* __restrict__ order,
-                               long long start, int B,
+                               long long start, long long row_base, int B,
                                float* __restrict__ xb,
                                int* __restrict__ yb) {
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < B * 784;
        i += gridDim.x * blockDim.x) {
     int b = i / 784, j = i % 784;
-    long long src = order[start + b];
+    long long src = row_base + order[start + b];
     xb[i] = shard_x[src * 784 + j];
     if (j == 0) yb[b] = (int)shard_y[src];
   }
+}
+
+// grad = (w_server - w_trained) * weight  (client pseudo-gradient, K1+K2)
+__global__ void k_cnn_pseudo_grad(float* __restrict__ g,
+                                  const float* __restrict__ ws,
+                                  const float* __restrict__ wt, float weight,
+                                  long long n) {
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n;
+       i += (long long)gridDim.x * blockDim.x)
+    g[i] = (ws[i] - wt[i]) * weight;
+}
+
+__global__ void k_cnn_axpy(float* __restrict__ y, const float* __restrict__ x,
+                           long long n) {
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n;
+       i += (long long)gridDim.x * blockDim.x)
+    y[i] += x[i];
+}
+
+__global__ void k_copy(float* __restrict__ dst, const float* __restrict__ src,
+                       long long n) {
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n;
+       i += (long long)gridDim.x * blockDim.x)
+    dst[i] = src[i];
 }
 
 // ---------------------------------------------------------------------------
@@ -503,7 +527,7 @@ extern "C" void launch_cnn_epoch(
     long long n, int bs, int C, float* params, float* grads,
     CnnWorkspace ws, const float* lr_t, float max_norm, float p1, float p2,
     float* stats_acc, float* loss_acc, unsigned long long seed,
-    hipStream_t s) {
+    hipStream_t s, long long row_base = 0) {
   CnnOffsets o = cnn_offsets(C);
   int n_batches = (int)((n + bs - 1) / bs);
   for (int it = 0; it < n_batches; ++it) {
@@ -512,7 +536,8 @@ extern "C" void launch_cnn_epoch(
     unsigned long long off = (unsigned long long)it;
     int g1 = (B * 784 + FBLK - 1) / FBLK;
     hipLaunchKernelGGL(k_gather_batch, dim3(g1), dim3(FBLK), 0, s,
-                       shard_x, shard_y, order, start, B, ws.xb, ws.yb);
+                       shard_x, shard_y, order, start, row_base, B,
+                       ws.xb, ws.yb);
     hipLaunchKernelGGL(k_conv1_fwd, dim3((B * 21632 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.xb, params + o.w1, params + o.b1,
                        B, ws.a1);
@@ -554,5 +579,41 @@ extern "C" void launch_cnn_epoch(
                             stats_acc, s);
     launch_sgd_step(params, grads, nullptr, 0.f, lr_t, 0.f, 0.f, 0.f, 0, 0,
                     o.total, s);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// WHOLE-ROUND driver: trains K clients back-to-back in one host call —
+// per client: copy-in server weights, run the fused epoch, write the
+// weighted pseudo-gradient, accumulate into the round buffer.  Per-client
+// losses/stats land in slots of loss_out[K] / stats_out[K*2].
+// ---------------------------------------------------------------------------
+extern "C" void launch_cnn_round(
+    const float* shard_x, const long long* shard_y,
+    const long long* orders,            // concatenated per-client shuffles
+    const long long* row_bases,         // HOST: K shard row offsets
+    const long long* order_offs,        // HOST: K offsets into `orders`
+    const long long* counts,            // HOST: K sample counts
+    const float* weights,               // HOST: K aggregation weights
+    const unsigned long long* seeds,    // HOST: K dropout seeds
+    int K, int bs, int C,
+    const float* server_params, float* params, float* grads,
+    float* round_accum, CnnWorkspace ws, const float* lr_t, float max_norm,
+    float p1, float p2, float* stats_out, float* loss_out,
+    hipStream_t s) {
+  CnnOffsets o = cnn_offsets(C);
+  int gp = (int)((o.total + FBLK - 1) / FBLK);
+  if (gp > 2048) gp = 2048;
+  for (int k = 0; k < K; ++k) {
+    hipLaunchKernelGGL(k_copy, dim3(gp), dim3(FBLK), 0, s,
+                       params, server_params, o.total);
+    launch_cnn_epoch(shard_x, shard_y, orders + order_offs[k], counts[k],
+                     bs, C, params, grads, ws, lr_t, max_norm, p1, p2,
+                     stats_out + 2 * k, loss_out + k, seeds[k], s,
+                     row_bases[k]);
+    hipLaunchKernelGGL(k_cnn_pseudo_grad, dim3(gp), dim3(FBLK), 0, s,
+                       grads, server_params, params, weights[k], o.total);
+    hipLaunchKernelGGL(k_cnn_axpy, dim3(gp), dim3(FBLK), 0, s,
+                       round_accum, grads, o.total);
   }
 }

@@ -1,0 +1,74 @@
+"""Whole-round fused driver (_C.cnn_round) vs the per-client path:
+identical aggregated results given identical per-client seeds (GPU)."""
+
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _run(use_fused_round, n_rounds=3):
+    import bench
+    from msrflute_amd.comm import runtime as rt_mod
+    from msrflute_amd.core import client as client_mod
+    from msrflute_amd.core.server import OptimizationServer
+    from msrflute_amd.models import make_model
+    from msrflute_amd.models.generic_data import ArrayDataset
+    from msrflute_amd.ops.arena import ParameterArena
+    from msrflute_amd.ops.fused_optim import make_arena_optimizer
+    from tools.create_data import make_femnist_blob
+
+    class A:
+        warmup = 0
+        steps = n_rounds
+        clients_per_round = 8
+
+    config = bench.build_config(A())
+    config["client_config"]["use_fused_round"] = use_fused_round
+    config["client_config"]["parallel_clients"] = 4
+    config["model_path"] = "/tmp/fr_models"
+    os.makedirs(config["model_path"], exist_ok=True)
+
+    blob = make_femnist_blob(n_users=30, samples_per_user=100, seed=5)
+    ds = ArrayDataset(blob, test_only=False, user_idx=-1, args={},
+                      x_shape=(28, 28))
+    ds.user_data = blob["user_data"]
+    ds.user_data_label = blob["user_data_label"]
+    client_mod.train_dataset = ds
+
+    rt = rt_mod.init_runtime(backend="gloo", seed=99)
+    torch.manual_seed(99 + 12345)
+    model = make_model(config["model_config"])
+    arena = ParameterArena(model, bind_grads=True)
+    optimizer = make_arena_optimizer(
+        dict(config["server_config"]["optimizer_config"]), arena)
+    server = OptimizationServer(
+        num_clients=30, model=model, optimizer=optimizer, ss_scheduler=None,
+        data_path=None, model_path=config["model_path"],
+        server_train_dataloader=None, config=config, idx_val_clients=[],
+        idx_test_clients=[], runtime=rt, arena=arena, task="cv_cnn_femnist")
+    server.run_stats = {k: [] for k in [
+        "secsPerClientRound", "secsPerClient", "secsPerClientTraining",
+        "secsPerClientSetup", "secsPerClientFull",
+        "secsPerRoundHousekeeping", "secsPerRoundTotal",
+        "communicationCosts"]}
+    losses = []
+    for i in range(n_rounds):
+        server.run_one_round(i, housekeeping=False)
+        losses.append(sum(server.train_loss))
+    torch.cuda.synchronize()
+    out = arena.data.clone().cpu()
+    rt.shutdown()
+    rt_mod.set_runtime(None)
+    return out, losses
+
+
+def test_fused_round_matches_per_client_path():
+    w1, l1 = _run(use_fused_round=False)
+    w2, l2 = _run(use_fused_round=True)
+    assert torch.allclose(w1, w2, rtol=1e-4, atol=1e-5), \
+        (w1 - w2).abs().max().item()
+    for a, b in zip(l1, l2):
+        assert abs(a - b) < 1e-2 * max(abs(a), 1.0), (l1, l2)
