@@ -1,15 +1,37 @@
 """Whole-round fused driver (_C.cnn_round) vs the per-client path:
-identical aggregated results given identical per-client seeds (GPU)."""
+identical aggregated results given identical per-client seeds (GPU).
+
+Each variant runs in a SUBPROCESS: the comparison is only meaningful
+from identical starting state, and in-suite CUDA/allocator state
+otherwise leaks between the two runs.
+"""
 
 import os
+import subprocess
+import sys
 
 import pytest
 import torch
 
 pytestmark = pytest.mark.gpu
 
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
 
 def _run(use_fused_round, n_rounds=3):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    out = f"/tmp/fr_{int(use_fused_round)}_{n_rounds}.pt"
+    r = subprocess.run(
+        [sys.executable, __file__, str(int(use_fused_round)),
+         str(n_rounds), out],
+        env=env, capture_output=True, text=True, timeout=400, cwd=REPO)
+    assert r.returncode == 0, r.stderr[-2000:]
+    blob = torch.load(out, weights_only=False)
+    return blob["w"], blob["losses"]
+
+
+def _run_inproc(use_fused_round, n_rounds=3):
     import bench
     from msrflute_amd.comm import runtime as rt_mod
     from msrflute_amd.core import client as client_mod
@@ -63,6 +85,12 @@ def _run(use_fused_round, n_rounds=3):
     rt.shutdown()
     rt_mod.set_runtime(None)
     return out, losses
+
+
+if __name__ == "__main__":
+    use_fused, nr, path = int(sys.argv[1]), int(sys.argv[2]), sys.argv[3]
+    w, losses = _run_inproc(bool(use_fused), nr)
+    torch.save({"w": w, "losses": losses}, path)
 
 
 def test_fused_round_matches_per_client_path_one_round():
