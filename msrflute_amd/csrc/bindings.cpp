@@ -343,6 +343,15 @@ void cnn_round(torch::Tensor shard_x, torch::Tensor shard_y,
   TORCH_CHECK(!row_bases.is_cuda() && !order_offs.is_cuda() &&
               !counts.is_cuda() && !weights.is_cuda() && !seeds.is_cuda(),
               "per-client metadata must be host tensors");
+  TORCH_CHECK(row_bases.scalar_type() == torch::kInt64 &&
+              order_offs.scalar_type() == torch::kInt64 &&
+              counts.scalar_type() == torch::kInt64 &&
+              seeds.scalar_type() == torch::kInt64 &&
+              weights.scalar_type() == torch::kFloat32,
+              "metadata dtypes: int64 (weights fp32)");
+  TORCH_CHECK(row_bases.is_contiguous() && order_offs.is_contiguous() &&
+              counts.is_contiguous() && weights.is_contiguous() &&
+              seeds.is_contiguous());
   int K = (int)counts.numel();
   TORCH_CHECK(stats_out.numel() >= 2 * K && loss_out.numel() >= K);
   TORCH_CHECK(bs >= 1 && bs <= 32);
