@@ -198,8 +198,10 @@ class DeviceShardStore:
         nbytes = flat_x.nbytes + flat_y.nbytes
         if nbytes > budget_bytes:
             raise MemoryError(f"shard store would need {nbytes >> 30} GiB")
-        hx = torch.from_numpy(flat_x).pin_memory()
-        hy = torch.from_numpy(flat_y).to(torch.int64).pin_memory()
+        hx = torch.from_numpy(flat_x)
+        hy = torch.from_numpy(flat_y).to(torch.int64)
+        if torch.device(device).type == "cuda":
+            hx, hy = hx.pin_memory(), hy.pin_memory()
         self.x = hx.to(device, non_blocking=True)
         self.y = hy.to(device, non_blocking=True)
         if x_shape:
