@@ -46,6 +46,8 @@ def test_bert_adapters_freeze_base():
         "intermediate_size": 64, "adapter": True, "adapter_dim": 8},
         "training": {"batch_size": 2, "label_smoothing_factor": 0.1,
                      "seed": 1}}})
+    if torch.cuda.is_available():  # model.loss moves inputs via to_device
+        m = m.cuda()
     trainable = [n for n, p in m.named_parameters() if p.requires_grad]
     frozen = [n for n, p in m.named_parameters() if not p.requires_grad]
     assert trainable and all("adapters" in n for n in trainable)
