@@ -123,3 +123,51 @@ def test_schema_rejects_bad_optimizer():
     cfg["server_config"]["optimizer_config"]["type"] = "nonsense"
     with pytest.raises(Exception):
         FLUTEConfig.from_dict(cfg)
+
+
+def test_scheduled_sampling_ramp():
+    from msrflute_amd.utils.schedulers import ScheduledSamplingScheduler
+
+    class M:
+        pass
+    m = M()
+    s = ScheduledSamplingScheduler(m, ramp_start=2, ramp_stop=6,
+                                   initial_rate=1.0, final_rate=0.0)
+    rates = []
+    for _ in range(9):
+        s.step()
+        rates.append(m.scheduled_sampling_rate)
+    assert rates[0] == 1.0 and rates[1] == 1.0       # before ramp
+    assert rates[3] < rates[2] <= 1.0                # ramping down
+    assert rates[8] == 0.0 and m.scheduled_sampling is False
+    # checkpoint round-trip excludes the model
+    sd = s.state_dict()
+    assert "model" not in sd
+    s2 = ScheduledSamplingScheduler(m, 0, 1, 0, 0)
+    s2.load_state_dict(sd)
+    assert s2.iter == s.iter
+
+
+def test_nbest_task_scheduler_stages():
+    from msrflute_amd.utils.schedulers import NBestTaskScheduler
+    s = NBestTaskScheduler(num_tasks=[1, 3, 5], iteration_per_task=[2, 4, 6])
+    seen = []
+    for _ in range(12):
+        s.step()
+        seen.append(s.current_num_tasks())
+    # within each 6-iteration cycle: stages advance 1 -> 3 -> 5
+    assert seen[0] == 1 and 3 in seen[:6] and 5 in seen[:6]
+    assert s.no_label_updates() == 3  # 12 iters / 6 per cycle + 1
+
+
+def test_val_loss_annealing_e2e(tmp_path):
+    """Server optimizer LR decays on val-loss plateau (reference
+    val_loss annealing, utils/utils.py:151-186 + trainer.py:139-155)."""
+    data_dir = str(tmp_path / "data")
+    _make_data("cv_lr_mnist", data_dir)
+    cfg = _base_cfg(rounds=4)
+    cfg["server_config"]["annealing_config"] = {
+        "type": "val_loss", "gamma": 0.5, "step_interval": "epoch",
+        "patience": 0, "step_size": 1}
+    cfg["server_config"]["initial_val"] = True
+    _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
