@@ -26,3 +26,24 @@ def test_tsv_json_packed_vocab_roundtrip(tmp_path):
     # the packed blob loads through the engine's generic loader
     from msrflute_amd.models.generic_data import load_blob
     assert load_blob(str(tmp_path / "b.pt"))["users"] == ["alice", "bob"]
+
+
+def test_create_data_cli_roundtrip(tmp_path):
+    """tools/create_data.py CLI writes loadable train/val/test blobs."""
+    import subprocess
+    import sys
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    r = subprocess.run(
+        [sys.executable, "tools/create_data.py", "--task", "cv_lr_mnist",
+         "--out", str(tmp_path), "--users", "6", "--samples", "4"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr[-1000:]
+    from msrflute_amd.models.generic_data import load_blob
+    for split in ["train", "val", "test"]:
+        blob = load_blob(str(tmp_path / "cv_lr_mnist" / f"{split}_data.pt"))
+        assert blob["users"] and blob["user_data"]
+    train = load_blob(str(tmp_path / "cv_lr_mnist" / "train_data.pt"))
+    assert len(train["users"]) == 6
+    assert train["num_samples"][0] == 4
