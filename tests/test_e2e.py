@@ -162,3 +162,16 @@ def test_single_vs_two_process_same_result_dga(tmp_data_dir, tmp_path):
                      map_location="cpu", weights_only=False)["model_state_dict"]
     for k in sd1:
         assert torch.allclose(sd1[k], sd2[k], rtol=1e-4, atol=1e-6), k
+
+
+def test_four_process_more_ranks_than_clients(tmp_data_dir, tmp_path):
+    """world_size=4 with 3 clients/round: one rank is idle every round —
+    the zero-row metadata gather and empty-partition paths must hold."""
+    cfg = _write_config(tmp_path, max_iteration=3, initial_val=False,
+                        num_clients_per_iteration=3)
+    out = str(tmp_path / "w4")
+    r = _run(4, cfg, tmp_data_dir, out, port=29809)
+    assert r.returncode == 0, r.stderr[-3000:]
+    m0 = _read_metrics(out, 0)
+    losses = [m["value"] for m in m0 if m["key"] == "Training loss"]
+    assert len(losses) == 3 and all(l > 0 for l in losses)
