@@ -179,6 +179,14 @@ def init_runtime(backend: str = "nccl", seed: int = 0) -> FedRuntime:
     global _RUNTIME
     if _RUNTIME is None:
         _RUNTIME = FedRuntime(backend=backend, seed=seed)
+    elif _RUNTIME.seed != seed:
+        # Reconfigure the round-derivation seed for a new run in the same
+        # process.  (A stale singleton seed silently changed client
+        # sampling for subsequent runs — observed as in-suite divergence
+        # of otherwise-deterministic training comparisons.)
+        print_rank(f"runtime already initialized; updating seed "
+                   f"{_RUNTIME.seed} -> {seed}")
+        _RUNTIME.seed = seed
     return _RUNTIME
 
 
