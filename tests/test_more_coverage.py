@@ -171,3 +171,46 @@ def test_val_loss_annealing_e2e(tmp_path):
         "patience": 0, "step_size": 1}
     cfg["server_config"]["initial_val"] = True
     _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
+
+
+def test_mlm_line_by_line_mode():
+    """mlm_bert dataset line-by-line framing (reference dataset.py:70-82):
+    one padded frame per utterance instead of concatenate-and-chunk."""
+    from importlib.machinery import SourceFileLoader
+    ds_mod = SourceFileLoader(
+        "mlm_ds", os.path.join(REPO, "experiments/mlm_bert/dataloaders/"
+                               "dataset.py")).load_module()
+    blob = {"users": ["a"], "num_samples": [2],
+            "user_data": {"a": {"x": [[5, 6, 7], [8, 9, 10, 11, 12]]}}}
+    lbl = ds_mod.Dataset(blob, args={"max_seq_length": 4,
+                                     "process_line_by_line": True},
+                         user_idx=0)
+    assert len(lbl) == 2                       # one frame per utterance
+    ids0, attn0 = lbl[0]
+    assert list(ids0) == [5, 6, 7, 0] and list(attn0) == [1, 1, 1, 0]
+    ids1, _ = lbl[1]
+    assert list(ids1) == [8, 9, 10, 11]        # truncated to max_seq_length
+
+    grouped = ds_mod.Dataset(blob, args={"max_seq_length": 4}, user_idx=0)
+    assert len(grouped) == 2                   # 8 tokens -> 2 frames of 4
+    g0, _ = grouped[0]
+    assert list(g0) == [5, 6, 7, 8]            # concatenated across utts
+
+
+def test_quant_threshold_annealing_progresses(tmp_path):
+    """quant_thresh multiplies by quant_anneal every round (reference
+    server.py:295-298)."""
+    import json
+    data_dir = str(tmp_path / "data")
+    _make_data("cv_lr_mnist", data_dir)
+    cfg = _base_cfg(rounds=3)
+    cfg["strategy"] = "DGA"
+    cfg["client_config"].update(quant_thresh=0.5, quant_bits=4,
+                                quant_anneal=0.5)
+    out = _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
+    path = os.path.join(out, "msrflute_amd", "log", "metrics_rank0.jsonl")
+    with open(path) as f:
+        vals = [json.loads(l)["value"] for l in f
+                if json.loads(l)["key"] == "Quantization Thresh."]
+    assert len(vals) == 3
+    assert vals[0] == 0.25 and vals[1] == 0.125 and vals[2] == 0.0625
