@@ -214,3 +214,55 @@ def test_quant_threshold_annealing_progresses(tmp_path):
                 if json.loads(l)["key"] == "Quantization Thresh."]
     assert len(vals) == 3
     assert vals[0] == 0.25 and vals[1] == 0.125 and vals[2] == 0.0625
+
+
+def test_client_range_sampling(tmp_path):
+    """num_clients_per_iteration as a "min,max" random range (reference
+    server.py:84-86, 284-291)."""
+    import json
+    data_dir = str(tmp_path / "data")
+    _make_data("cv_lr_mnist", data_dir)
+    cfg = _base_cfg(rounds=4)
+    cfg["server_config"]["num_clients_per_iteration"] = "2,5"
+    out = _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
+    path = os.path.join(out, "msrflute_amd", "log", "metrics_rank0.jsonl")
+    with open(path) as f:
+        ns = [json.loads(l)["value"] for l in f
+              if json.loads(l)["key"] == "Clients for round"]
+    assert len(ns) == 4 and all(2 <= n <= 5 for n in ns)
+
+
+def test_fedavg_freeze_layer_zeroes_segment():
+    """freeze_layer zeroes that parameter's pseudo-gradient before upload
+    (reference fedavg.py:81-86)."""
+    from msrflute_amd.ops.arena import ParameterArena
+    from msrflute_amd.strategies.fedavg import FedAvg
+
+    net = torch.nn.Sequential(torch.nn.Linear(4, 3), torch.nn.Linear(3, 2))
+    arena = ParameterArena(net, bind_grads=True)
+    arena.grad.fill_(1.0)
+    cfg = {"model_config": {"freeze_layer": "1.weight"},
+           "client_config": {}, "server_config": {}, "strategy": "FedAvg"}
+
+    class Tr:
+        num_samples = 7
+    t = Tr()
+    t.arena = arena
+    strat = FedAvg("client", cfg)
+    payload = strat.generate_client_payload(t)
+    assert payload["weight"] == 7.0
+    off, n = arena.segment_of("1.weight")
+    assert payload["grad"][off:off + n].abs().sum() == 0
+    other_off, other_n = arena.segment_of("0.weight")
+    assert payload["grad"][other_off:other_off + other_n].abs().sum() > 0
+
+
+def test_component_wise_lr(tmp_path):
+    """trainer_config.updatable_names freezes/relearns components
+    (reference utils/utils.py:725-751)."""
+    data_dir = str(tmp_path / "data")
+    _make_data("cv_lr_mnist", data_dir)
+    cfg = _base_cfg(rounds=2)
+    cfg["client_config"]["trainer_config"] = {
+        "updatable_names": ["net.linear.weight"]}
+    _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
