@@ -266,3 +266,17 @@ def test_component_wise_lr(tmp_path):
     cfg["client_config"]["trainer_config"] = {
         "updatable_names": ["net.linear.weight"]}
     _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
+
+
+def test_epoch_snapshots_and_best_acc_criterion(tmp_path):
+    """model_backup_freq epoch{i} snapshots + best_model_criterion=acc
+    (reference server.py:530-559, evaluation best tracking)."""
+    data_dir = str(tmp_path / "data")
+    _make_data("cv_lr_mnist", data_dir)
+    cfg = _base_cfg(rounds=4)
+    cfg["server_config"].update(model_backup_freq=2, initial_val=True,
+                                best_model_criterion="acc")
+    out = _run_task("cv_lr_mnist", cfg, tmp_path, data_dir)
+    models = os.listdir(os.path.join(out, "msrflute_amd", "models"))
+    assert any(m.startswith("epoch") for m in models), models
+    assert "best_val_acc_model.tar" in models
