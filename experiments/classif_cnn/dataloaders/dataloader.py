@@ -1,4 +1,7 @@
-"""classif_cnn dataloader over the shared array machinery."""
+"""classif_cnn dataloader over the shared array machinery.
+
+Reference: experiments/classif_cnn/dataloaders/dataloader.py.
+"""
 
 import os
 from importlib.machinery import SourceFileLoader

@@ -1,4 +1,7 @@
-"""classif_cnn blob dataset: 3x32x32 CIFAR-shaped images."""
+"""classif_cnn blob dataset: 3x32x32 CIFAR-shaped images.
+
+Reference: experiments/classif_cnn/dataloaders/dataset.py (CIFAR10 HDF5 blob; testing/create_data.py:142-152).
+"""
 
 from msrflute_amd.models.generic_data import ArrayDataset
 

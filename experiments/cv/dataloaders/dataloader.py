@@ -1,4 +1,7 @@
-"""cv dataloader over the shared array machinery."""
+"""cv dataloader over the shared array machinery.
+
+Reference: experiments/cv/dataloaders/dataloader.py.
+"""
 
 import os
 from importlib.machinery import SourceFileLoader

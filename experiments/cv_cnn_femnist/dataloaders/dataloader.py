@@ -1,4 +1,7 @@
-"""cv_cnn_femnist dataloader over the shared array machinery."""
+"""cv_cnn_femnist dataloader over the shared array machinery.
+
+Reference: experiments/cv_cnn_femnist/dataloaders/dataloader.py.
+"""
 
 import os
 from importlib.machinery import SourceFileLoader

@@ -1,4 +1,7 @@
-"""cv_lr_mnist dataloader over the shared array machinery."""
+"""cv_lr_mnist dataloader over the shared array machinery.
+
+Reference: experiments/cv_lr_mnist/dataloaders/dataloader.py.
+"""
 
 import os
 from importlib.machinery import SourceFileLoader

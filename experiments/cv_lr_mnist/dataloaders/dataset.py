@@ -1,4 +1,7 @@
-"""MNIST blob dataset (flat 784 features)."""
+"""MNIST blob dataset (flat 784 features).
+
+Reference: experiments/cv_lr_mnist/dataloaders/dataset.py (flat 784-feature MNIST rows, 1000 users).
+"""
 
 from msrflute_amd.models.generic_data import ArrayDataset
 

@@ -1,4 +1,7 @@
-"""cv_resnet_fedcifar100 dataloader over the shared array machinery."""
+"""cv_resnet_fedcifar100 dataloader over the shared array machinery.
+
+Reference: experiments/cv_resnet_fedcifar100/dataloaders/dataloader.py.
+"""
 
 import os
 from importlib.machinery import SourceFileLoader

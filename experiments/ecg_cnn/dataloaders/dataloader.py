@@ -1,4 +1,7 @@
-"""ecg_cnn dataloader over the shared array machinery."""
+"""ecg_cnn dataloader over the shared array machinery.
+
+Reference: experiments/ecg_cnn/dataloaders/dataloader.py.
+"""
 
 import os
 from importlib.machinery import SourceFileLoader

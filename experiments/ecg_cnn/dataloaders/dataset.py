@@ -1,4 +1,7 @@
-"""ecg_cnn blob dataset: 187-sample heartbeat traces, 5 classes."""
+"""ecg_cnn blob dataset: 187-sample heartbeat traces, 5 classes.
+
+Reference: experiments/ecg_cnn/dataloaders/dataset.py (heartbeat CSV -> HDF5 rows).
+"""
 
 from msrflute_amd.models.generic_data import ArrayDataset
 

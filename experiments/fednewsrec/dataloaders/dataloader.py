@@ -1,4 +1,7 @@
-"""fednewsrec dataloader: batches are {'x': (click, candidates), 'y': y}."""
+"""fednewsrec dataloader: batches are {'x': (click, candidates), 'y': y}.
+
+Reference: experiments/fednewsrec/dataloaders/dataloader.py.
+"""
 
 import os
 from importlib.machinery import SourceFileLoader

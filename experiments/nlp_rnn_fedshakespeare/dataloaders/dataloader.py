@@ -1,4 +1,7 @@
-"""nlp_rnn_fedshakespeare dataloader over the shared array machinery."""
+"""nlp_rnn_fedshakespeare dataloader over the shared array machinery.
+
+Reference: experiments/nlp_rnn_fedshakespeare/dataloaders/dataloader.py.
+"""
 
 import os
 from importlib.machinery import SourceFileLoader

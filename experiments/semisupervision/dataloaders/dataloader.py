@@ -1,6 +1,9 @@
 """semisupervision dataloader: dict batches over the labeled view (the
 unlabeled phases build their own torch DataLoaders inside the FedLabels
-trainer)."""
+trainer).
+
+Reference: experiments/semisupervision/dataloaders/dataloader.py.
+"""
 
 import os
 from importlib.machinery import SourceFileLoader
