@@ -571,6 +571,21 @@ class ClientPool:
         self._streams_dirty = False
 
 
+def _fused_round_eligible(ex, data_cfg):
+    """The fused whole-round driver trains full shards with plain SGD and
+    aggregates every layer's gradient; fall back to the per-client path
+    whenever the eager path would behave differently: a frozen layer
+    (its pseudo-grad segment must be zeroed, fedavg.py:58) or a
+    num_skips_threshold (clients can be zero-weighted mid-epoch).
+    desired_max_samples is checked per client against shard sizes by the
+    callers (the eager path truncates each client's epoch there)."""
+    if ex.model_config.get("freeze_layer", None):
+        return False
+    if int(ex.client_config.get("num_skips_threshold", -1)) >= 0:
+        return False
+    return True
+
+
 def _fused_round_impl(ex, client_ids, initial_lr, seeds):
     """One _C.cnn_round call training all of ``ex``'s clients for this
     round (copy-in, fused epoch, weighted pseudo-grad, accumulate into
@@ -578,6 +593,8 @@ def _fused_round_impl(ex, client_ids, initial_lr, seeds):
     if ex.fused_cnn is None:
         return None
     data_cfg = ex.client_config["data_config"]["train"]
+    if not _fused_round_eligible(ex, data_cfg):
+        return None
     store = ex._get_shard_store(data_cfg)
     if store is None or store.x[0].numel() != 784:
         return None
@@ -586,6 +603,7 @@ def _fused_round_impl(ex, client_ids, initial_lr, seeds):
     if ds is None:
         return None
     fc = ex.fused_cnn
+    dms = data_cfg.get("desired_max_samples", None)
     counts, row_bases, order_offs, orders, w_list = [], [], [], [], []
     off = 0
     for cid, seed in zip(client_ids, seeds):
@@ -595,7 +613,7 @@ def _fused_round_impl(ex, client_ids, initial_lr, seeds):
             return None
         lo, hi = store.offsets[i], store.offsets[i + 1]
         n = hi - lo
-        if n == 0:
+        if n == 0 or (dms is not None and n > dms):
             return None
         torch.manual_seed(seed & 0x7FFFFFFFFFFF)
         orders.append(torch.randperm(n))
@@ -609,8 +627,16 @@ def _fused_round_impl(ex, client_ids, initial_lr, seeds):
     if getattr(ex, "_order_pin", None) is None or ex._order_pin.numel() < off:
         ex._order_pin = torch.empty(max(off, 1024),
                                     dtype=torch.int64).pin_memory()
+        ex._order_pin_ev = None
+    if ex._order_pin_ev is not None:
+        # previous round's non_blocking H2D of this buffer may still be
+        # queued; overwriting the pinned staging before it lands would
+        # corrupt that round's shuffle orders
+        ex._order_pin_ev.synchronize()
     ex._order_pin[:off].copy_(orders_cat)
     orders_dev = ex._order_pin[:off].to(ex.arena.device, non_blocking=True)
+    ex._order_pin_ev = torch.cuda.Event()
+    ex._order_pin_ev.record()
     if getattr(ex, "_round_stats", None) is None or \
             ex._round_stats.numel() < 2 * K:
         ex._round_stats = torch.zeros(2 * max(K, 8), device=ex.arena.device)
@@ -660,17 +686,23 @@ def _pool_run_fused_round_batch(self, client_ids, initial_lr, iteration,
     prim = self.executors[0]
     if prim.fused_cnn is None:
         return None
-    store = prim._get_shard_store(
-        prim.client_config["data_config"]["train"])
+    data_cfg = prim.client_config["data_config"]["train"]
+    if not _fused_round_eligible(prim, data_cfg):
+        return None
+    store = prim._get_shard_store(data_cfg)
     if store is None or store.x[0].numel() != 784:
         return None
     import msrflute_amd.core.client as cm
     ds = cm.train_dataset
     if ds is None:
         return None
+    dms = data_cfg.get("desired_max_samples", None)
     for cid in client_ids:
         i = store.user_pos.get(ds.user_list[cid])
-        if i is None or store.offsets[i + 1] - store.offsets[i] == 0:
+        if i is None:
+            return None
+        n = store.offsets[i + 1] - store.offsets[i]
+        if n == 0 or (dms is not None and n > dms):
             return None
     for ex in self.executors[1:]:
         if not ex._shard_store_tried:

@@ -240,305 +240,308 @@ class OptimizationServer:
             profiler.enable()
             if torch.cuda.is_available():
                 torch.cuda.nvtx.range_push(f"fl_round_{i}")
-        if True:
-            begin = time.time()
-            metrics_payload = {}
+        begin = time.time()
+        metrics_payload = {}
 
-            def log_m(k, v):
-                metrics_payload[k] = v
+        def log_m(k, v):
+            metrics_payload[k] = v
 
-            print_rank(f"==== iteration {i}")
-            log_m("Current iteration", i)
+        print_rank(f"==== iteration {i}")
+        log_m("Current iteration", i)
 
-            initial_lr = self.initial_lr_client * self.lr_weight
-            log_m("Client learning rate", initial_lr)
+        initial_lr = self.initial_lr_client * self.lr_weight
+        log_m("Client learning rate", initial_lr)
 
-            # clear the server grad arena before accumulation
-            self.worker_trainer.arena.zero_grad()
-            self.train_loss = []
+        # clear the server grad arena before accumulation
+        self.worker_trainer.arena.zero_grad()
+        self.train_loss = []
 
-            # number of clients this round (int or random range)
-            if len(self.num_clients_per_iteration) > 1:
-                num_clients_curr_iter = rt.round_rng(i, salt=2).randint(
-                    self.num_clients_per_iteration[0],
-                    self.num_clients_per_iteration[1])
+        # number of clients this round (int or random range)
+        if len(self.num_clients_per_iteration) > 1:
+            num_clients_curr_iter = rt.round_rng(i, salt=2).randint(
+                self.num_clients_per_iteration[0],
+                self.num_clients_per_iteration[1])
+        else:
+            num_clients_curr_iter = self.num_clients_per_iteration[0]
+        log_m("Clients for round", num_clients_curr_iter)
+
+        # quantization-threshold annealing (reference: server.py:295-298)
+        if self.quant_thresh is not None:
+            cc = self.config["client_config"]
+            cc["quant_thresh"] = cc.get("quant_thresh", self.quant_thresh) * \
+                cc.get("quant_anneal", 1.0)
+            self.quant_thresh = cc["quant_thresh"]
+            self.executor.client_strategy.quant_threshold = self.quant_thresh
+            log_m("Quantization Thresh.", cc["quant_thresh"])
+
+        # deterministic sampling — identical on every rank
+        sampled_idx_clients = rt.sample_clients(self.client_idx_list,
+                                                num_clients_curr_iter, i)
+        weights = ([self.client_num_samples_all[c] for c in sampled_idx_clients]
+                   if self.client_num_samples_all is not None else None)
+        parts = rt.partition(sampled_idx_clients, weights)
+        my_clients = parts[rt.rank]
+
+        clients_begin = time.time()
+        apply_privacy_metrics = bool(
+            self.config.get("privacy_metrics_config", None)
+            and self.config["privacy_metrics_config"]["apply_metrics"])
+        adaptive_leakage = apply_privacy_metrics and \
+            self.config["privacy_metrics_config"].get("adaptive_leakage_threshold", None)
+        privacy_metrics_stats = defaultdict(list)
+
+        for key in ["secsPerClient", "secsPerClientFull",
+                    "secsPerClientTraining", "secsPerClientSetup",
+                    "communicationCosts"]:
+            self.run_stats[key].append([])
+
+        # ---- local client training ----------------------------------
+        # whole-round fused path: ONE _C.cnn_round call per executor
+        # trains every local client (copy-in, epoch, pseudo-grad,
+        # accumulate all inside the extension)
+        fused_outputs = None
+        # The fused driver trains plain SGD with no proximal term, so it
+        # is only equivalent to the eager path under FedAvg (FedProx adds
+        # (mu/2)||w-w_g||^2 every local step — reference trainer.py:463).
+        if (hasattr(self.executor, "run_fused_round_batch")
+                and self.config["strategy"] == "FedAvg"
+                and self.config["client_config"].get("use_fused_round",
+                                                     True)
+                and not apply_privacy_metrics):
+            seeds = [rt.round_rng(i, salt=100 + c).getrandbits(62)
+                     for c in my_clients]
+            fused_outputs = self.executor.run_fused_round_batch(
+                my_clients, initial_lr, i, seeds)
+
+        local_outputs = []
+        for client_idx, pre_output in (
+                fused_outputs if fused_outputs is not None
+                else ((c, None) for c in my_clients)):
+            if pre_output is not None:
+                output = pre_output
             else:
-                num_clients_curr_iter = self.num_clients_per_iteration[0]
-            log_m("Clients for round", num_clients_curr_iter)
-
-            # quantization-threshold annealing (reference: server.py:295-298)
-            if self.quant_thresh is not None:
-                cc = self.config["client_config"]
-                cc["quant_thresh"] = cc.get("quant_thresh", self.quant_thresh) * \
-                    cc.get("quant_anneal", 1.0)
-                self.quant_thresh = cc["quant_thresh"]
-                self.executor.client_strategy.quant_threshold = self.quant_thresh
-                log_m("Quantization Thresh.", cc["quant_thresh"])
-
-            # deterministic sampling — identical on every rank
-            sampled_idx_clients = rt.sample_clients(self.client_idx_list,
-                                                    num_clients_curr_iter, i)
-            weights = ([self.client_num_samples_all[c] for c in sampled_idx_clients]
-                       if self.client_num_samples_all is not None else None)
-            parts = rt.partition(sampled_idx_clients, weights)
-            my_clients = parts[rt.rank]
-
-            clients_begin = time.time()
-            apply_privacy_metrics = bool(
-                self.config.get("privacy_metrics_config", None)
-                and self.config["privacy_metrics_config"]["apply_metrics"])
-            adaptive_leakage = apply_privacy_metrics and \
-                self.config["privacy_metrics_config"].get("adaptive_leakage_threshold", None)
-            privacy_metrics_stats = defaultdict(list)
-
-            for key in ["secsPerClient", "secsPerClientFull",
-                        "secsPerClientTraining", "secsPerClientSetup",
-                        "communicationCosts"]:
-                self.run_stats[key].append([])
-
-            # ---- local client training ----------------------------------
-            # whole-round fused path: ONE _C.cnn_round call per executor
-            # trains every local client (copy-in, epoch, pseudo-grad,
-            # accumulate all inside the extension)
-            fused_outputs = None
-            if (hasattr(self.executor, "run_fused_round_batch")
-                    and self.config["client_config"].get("use_fused_round",
-                                                         True)
-                    and not apply_privacy_metrics):
-                seeds = [rt.round_rng(i, salt=100 + c).getrandbits(62)
-                         for c in my_clients]
-                fused_outputs = self.executor.run_fused_round_batch(
-                    my_clients, initial_lr, i, seeds)
-
-            local_outputs = []
-            for client_idx, pre_output in (
-                    fused_outputs if fused_outputs is not None
-                    else ((c, None) for c in my_clients)):
-                if pre_output is not None:
-                    output = pre_output
-                else:
-                    client = Client([client_idx], self.config, True)
-                    output = self.executor.process_round(
-                        client, initial_lr, i,
-                        round_seed=rt.round_rng(
-                            i, salt=100 + client_idx).getrandbits(62))
-                payload = output["pl"]
-                if output.get("wt", None) == 0.0 and payload is not None:
-                    payload["weight"] = 0.0
-                if isinstance(self.strategy, _needs_cid_cls()):
-                    ok = self.strategy.process_individual_payload(
-                        self.worker_trainer, payload, client_id=client_idx)
-                else:
-                    ok = self.strategy.process_individual_payload(
-                        self.worker_trainer, payload)
-                meta = {k: v for k, v in output.items() if k != "pl"}
-                meta["accepted"] = bool(ok)
-                meta["wt"] = payload["weight"] if payload is not None else 0.0
-                local_outputs.append((client_idx, meta))
-                self.run_stats["secsPerClient"][-1].append(time.time() - clients_begin)
-
-            # join client streams + fold pool accumulators into the server
-            # grad arena (no-op for the single-executor path)
-            if hasattr(self.executor, "flush"):
-                self.executor.flush(self.worker_trainer.arena.grad)
-
-            # ---- batched finalize of deferred client stats ---------------
-            # lazy-stats clients carried device tensors; ONE host transfer
-            # materializes every client's loss/Σg/Σg² for this rank
-            lazy = [(idx, meta) for idx, (_, meta) in enumerate(local_outputs)
-                    if "_lazy" in meta]
-            if lazy:
-                flat = torch.stack(
-                    [torch.stack([meta["_lazy"][0].reshape(()),
-                                  meta["_lazy"][1][0], meta["_lazy"][1][1]])
-                     for _, meta in lazy]).cpu().tolist()
-                for (idx, meta), (tl, s, q) in zip(lazy, flat):
-                    n = max(meta["_lazy"][2], 1)
-                    mean = s / n
-                    meta["tl"] = tl
-                    meta["ng"] = mean
-                    meta["mg"] = math.sqrt(max(q / n, 0.0))
-                    meta["vg"] = max(q / n - mean ** 2, 0.0)
-                    meta["rg"] = math.sqrt(max(q, 0.0))
-                    del meta["_lazy"]
-
-            # ---- metadata exchange (one all_gather per round) ------------
-            # numeric metas travel as ONE fused tensor all_gather (no
-            # pickle on the fabric); privacy-metric dicts fall back to the
-            # object path
-            if rt.size > 1 and not apply_privacy_metrics:
-                cols = ["tl", "mg", "ng", "vg", "rg", "ns", "wt", "ts"]
-                rows = [[float(cid), 1.0 if meta["accepted"] else 0.0]
-                        + [float(meta.get(c, 0.0)) for c in cols]
-                        for cid, meta in local_outputs]
-                local_t = torch.tensor(rows, dtype=torch.float64).reshape(
-                    len(rows), 2 + len(cols))
-                per_rank = rt.all_gather_rows(local_t,
-                                              [len(p) for p in parts])
-                gathered = []
-                for r, block in enumerate(per_rank):
-                    if r == rt.rank:
-                        gathered.extend(local_outputs)  # keep local cs dicts
-                        continue
-                    for row in block.tolist():
-                        meta = {"accepted": bool(row[1]),
-                                **{c: row[2 + j] for j, c in enumerate(cols)},
-                                "ns": int(row[7]),
-                                "cs": {"full cost": 0.0, "training": 0.0,
-                                       "setup": 0.0}}
-                        gathered.append((int(row[0]), meta))
+                client = Client([client_idx], self.config, True)
+                output = self.executor.process_round(
+                    client, initial_lr, i,
+                    round_seed=rt.round_rng(
+                        i, salt=100 + client_idx).getrandbits(62))
+            payload = output["pl"]
+            if output.get("wt", None) == 0.0 and payload is not None:
+                payload["weight"] = 0.0
+            if isinstance(self.strategy, _needs_cid_cls()):
+                ok = self.strategy.process_individual_payload(
+                    self.worker_trainer, payload, client_id=client_idx)
             else:
-                gathered = sum(rt.all_gather_object(local_outputs), [])
-            order = {c: k for k, c in enumerate(sampled_idx_clients)}
-            gathered.sort(key=lambda t: order.get(t[0], 1 << 30))
+                ok = self.strategy.process_individual_payload(
+                    self.worker_trainer, payload)
+            meta = {k: v for k, v in output.items() if k != "pl"}
+            meta["accepted"] = bool(ok)
+            meta["wt"] = payload["weight"] if payload is not None else 0.0
+            local_outputs.append((client_idx, meta))
+            self.run_stats["secsPerClient"][-1].append(time.time() - clients_begin)
 
-            client_losses, client_mag_grads = [], []
-            client_mean_grads, client_var_grads, client_norm_grads = [], [], []
-            client_weights_all = []
-            client_pos = {}
-            for pos, (cid, meta) in enumerate(gathered):
-                if not meta["accepted"]:
-                    num_clients_curr_iter -= 1
+        # join client streams + fold pool accumulators into the server
+        # grad arena (no-op for the single-executor path)
+        if hasattr(self.executor, "flush"):
+            self.executor.flush(self.worker_trainer.arena.grad)
+
+        # ---- batched finalize of deferred client stats ---------------
+        # lazy-stats clients carried device tensors; ONE host transfer
+        # materializes every client's loss/Σg/Σg² for this rank
+        lazy = [(idx, meta) for idx, (_, meta) in enumerate(local_outputs)
+                if "_lazy" in meta]
+        if lazy:
+            flat = torch.stack(
+                [torch.stack([meta["_lazy"][0].reshape(()),
+                              meta["_lazy"][1][0], meta["_lazy"][1][1]])
+                 for _, meta in lazy]).cpu().tolist()
+            for (idx, meta), (tl, s, q) in zip(lazy, flat):
+                n = max(meta["_lazy"][2], 1)
+                mean = s / n
+                meta["tl"] = tl
+                meta["ng"] = mean
+                meta["mg"] = math.sqrt(max(q / n, 0.0))
+                meta["vg"] = max(q / n - mean ** 2, 0.0)
+                meta["rg"] = math.sqrt(max(q, 0.0))
+                del meta["_lazy"]
+
+        # ---- metadata exchange (one all_gather per round) ------------
+        # numeric metas travel as ONE fused tensor all_gather (no
+        # pickle on the fabric); privacy-metric dicts fall back to the
+        # object path
+        if rt.size > 1 and not apply_privacy_metrics:
+            cols = ["tl", "mg", "ng", "vg", "rg", "ns", "wt", "ts"]
+            rows = [[float(cid), 1.0 if meta["accepted"] else 0.0]
+                    + [float(meta.get(c, 0.0)) for c in cols]
+                    for cid, meta in local_outputs]
+            local_t = torch.tensor(rows, dtype=torch.float64).reshape(
+                len(rows), 2 + len(cols))
+            per_rank = rt.all_gather_rows(local_t,
+                                          [len(p) for p in parts])
+            gathered = []
+            for r, block in enumerate(per_rank):
+                if r == rt.rank:
+                    gathered.extend(local_outputs)  # keep local cs dicts
                     continue
-                client_pos[cid] = len(client_losses)
-                self.train_loss.append(meta["tl"])
-                client_losses.append(meta["tl"])
-                client_mag_grads.append(meta["mg"])
-                client_mean_grads.append(meta["ng"])
-                client_var_grads.append(meta["vg"])
-                client_norm_grads.append(meta["rg"])
-                client_weights_all.append(meta["wt"])
-                cs = meta["cs"]
-                self.run_stats["secsPerClientFull"][-1].append(cs["full cost"])
-                self.run_stats["secsPerClientTraining"][-1].append(cs["training"])
-                self.run_stats["secsPerClientSetup"][-1].append(cs["setup"])
-                self.run_stats["communicationCosts"][-1].append(time.time() - meta["ts"])
-                if apply_privacy_metrics and "ps" in meta:
-                    for metric, value in meta["ps"].items():
-                        privacy_metrics_stats[metric].append(value)
-            self.strategy._client_pos = client_pos
+                for row in block.tolist():
+                    meta = {"accepted": bool(row[1]),
+                            **{c: row[2 + j] for j, c in enumerate(cols)},
+                            "ns": int(row[7]),
+                            "cs": {"full cost": 0.0, "training": 0.0,
+                                   "setup": 0.0}}
+                    gathered.append((int(row[0]), meta))
+        else:
+            gathered = sum(rt.all_gather_object(local_outputs), [])
+        order = {c: k for k, c in enumerate(sampled_idx_clients)}
+        gathered.sort(key=lambda t: order.get(t[0], 1 << 30))
 
-            client_mag_grads = np.array(client_mag_grads)
-            client_mean_grads = np.array(client_mean_grads)
-            client_var_grads = np.array(client_var_grads)
-            client_norm_grads = np.array(client_norm_grads)
-            client_stats = (np.array(client_weights_all), client_mag_grads,
-                            client_mean_grads, client_var_grads)
+        client_losses, client_mag_grads = [], []
+        client_mean_grads, client_var_grads, client_norm_grads = [], [], []
+        client_weights_all = []
+        client_pos = {}
+        for pos, (cid, meta) in enumerate(gathered):
+            if not meta["accepted"]:
+                num_clients_curr_iter -= 1
+                continue
+            client_pos[cid] = len(client_losses)
+            self.train_loss.append(meta["tl"])
+            client_losses.append(meta["tl"])
+            client_mag_grads.append(meta["mg"])
+            client_mean_grads.append(meta["ng"])
+            client_var_grads.append(meta["vg"])
+            client_norm_grads.append(meta["rg"])
+            client_weights_all.append(meta["wt"])
+            cs = meta["cs"]
+            self.run_stats["secsPerClientFull"][-1].append(cs["full cost"])
+            self.run_stats["secsPerClientTraining"][-1].append(cs["training"])
+            self.run_stats["secsPerClientSetup"][-1].append(cs["setup"])
+            self.run_stats["communicationCosts"][-1].append(time.time() - meta["ts"])
+            if apply_privacy_metrics and "ps" in meta:
+                for metric, value in meta["ps"].items():
+                    privacy_metrics_stats[metric].append(value)
+        self.strategy._client_pos = client_pos
 
-            dump_norm_stats = self.config.get("dump_norm_stats", False)
-            if dump_norm_stats and is_chief:
-                with open(os.path.join(self.model_path, "norm_stats.txt"), "a",
-                          encoding="utf-8") as f:
-                    f.write(f"{json.dumps(list(client_norm_grads))}\n")
+        client_mag_grads = np.array(client_mag_grads)
+        client_mean_grads = np.array(client_mean_grads)
+        client_var_grads = np.array(client_var_grads)
+        client_norm_grads = np.array(client_norm_grads)
+        client_stats = (np.array(client_weights_all), client_mag_grads,
+                        client_mean_grads, client_var_grads)
 
-            if apply_privacy_metrics:
-                for metric, values in privacy_metrics_stats.items():
-                    if metric == "Dropped clients":
-                        log_m(metric, sum(values))
-                    else:
-                        log_m(metric, max(values))
-                if isinstance(adaptive_leakage, float):
-                    values = sorted(privacy_metrics_stats["Practical epsilon (Max leakage)"])
-                    if values:
-                        new_threshold = values[int(adaptive_leakage * len(values))]
-                        print_rank(f"Updating leakage threshold to {new_threshold}")
-                        self.config["privacy_metrics_config"]["max_allowed_leakage"] = new_threshold
+        dump_norm_stats = self.config.get("dump_norm_stats", False)
+        if dump_norm_stats and is_chief:
+            with open(os.path.join(self.model_path, "norm_stats.txt"), "a",
+                      encoding="utf-8") as f:
+                f.write(f"{json.dumps(list(client_norm_grads))}\n")
 
-            end = time.time()
-            self.run_stats["secsPerClientRound"].append(end - begin)
-            begin = end
-            log_m("Training loss", sum(self.train_loss))
+        if apply_privacy_metrics:
+            for metric, values in privacy_metrics_stats.items():
+                if metric == "Dropped clients":
+                    log_m(metric, sum(values))
+                else:
+                    log_m(metric, max(values))
+            if isinstance(adaptive_leakage, float):
+                values = sorted(privacy_metrics_stats["Practical epsilon (Max leakage)"])
+                if values:
+                    new_threshold = values[int(adaptive_leakage * len(values))]
+                    print_rank(f"Updating leakage threshold to {new_threshold}")
+                    self.config["privacy_metrics_config"]["max_allowed_leakage"] = new_threshold
 
-            # ---- combine: all-reduce + replicated server update ----------
-            cs_for_strategy = (client_mag_grads, client_mean_grads, client_var_grads)
-            if hasattr(self.strategy, "want_rl") and self.strategy.want_rl:
-                cs_for_strategy = client_stats
-            self.losses = self.strategy.combine_payloads(
-                worker_trainer=self.worker_trainer, curr_iter=i,
-                num_clients_curr_iter=max(num_clients_curr_iter, 1),
-                total_clients=len(self.client_idx_list),
-                client_stats=cs_for_strategy, logger=log_m)
+        end = time.time()
+        self.run_stats["secsPerClientRound"].append(end - begin)
+        begin = end
+        log_m("Training loss", sum(self.train_loss))
 
-            # ---- server replay training (reference: server.py:430-442) ----
-            if self.server_trainer is not None:
-                print_rank("Running replay iterations on server")
-                torch.manual_seed(rt.round_rng(i, salt=3).getrandbits(62))
-                if "updatable_names" in self.server_trainer_config:
-                    set_component_wise_lr(self.worker_trainer.model,
-                                          self.server_optimizer_config,
-                                          self.server_trainer_config["updatable_names"])
-                self.server_trainer.prepare_iteration(self.worker_trainer.model)
-                self.server_trainer.train_desired_samples(self.server_replay_iterations)
-                self.worker_trainer.model.load_state_dict(
-                    self.server_trainer.model.state_dict())
+        # ---- combine: all-reduce + replicated server update ----------
+        cs_for_strategy = (client_mag_grads, client_mean_grads, client_var_grads)
+        if hasattr(self.strategy, "want_rl") and self.strategy.want_rl:
+            cs_for_strategy = client_stats
+        self.losses = self.strategy.combine_payloads(
+            worker_trainer=self.worker_trainer, curr_iter=i,
+            num_clients_curr_iter=max(num_clients_curr_iter, 1),
+            total_clients=len(self.client_idx_list),
+            client_stats=cs_for_strategy, logger=log_m)
 
-            self.worker_trainer.run_ss_scheduler()
+        # ---- server replay training (reference: server.py:430-442) ----
+        if self.server_trainer is not None:
+            print_rank("Running replay iterations on server")
+            torch.manual_seed(rt.round_rng(i, salt=3).getrandbits(62))
+            if "updatable_names" in self.server_trainer_config:
+                set_component_wise_lr(self.worker_trainer.model,
+                                      self.server_optimizer_config,
+                                      self.server_trainer_config["updatable_names"])
+            self.server_trainer.prepare_iteration(self.worker_trainer.model)
+            self.server_trainer.train_desired_samples(self.server_replay_iterations)
+            self.worker_trainer.model.load_state_dict(
+                self.server_trainer.model.state_dict())
 
-            # ---- evaluation cadence --------------------------------------
-            if housekeeping and ((i + 1) % self.val_freq) == 0:
-                eval_list.append("val")
-            if housekeeping and ((i + 1) % self.req_freq) == 0:
-                eval_list.append("test")
+        self.worker_trainer.run_ss_scheduler()
 
-            ran_val = "val" in eval_list
-            if len(eval_list) > 0:
-                print_rank(f"Running {eval_list} at itr={i+1}")
-                self.metrics["worker_trainer"] = self.worker_trainer
-                if hasattr(self.strategy, "tmp_unsup") and self.strategy.tmp_unsup is not None:
-                    self.metrics["tmp_sup"] = self.strategy.tmp_sup
-                    self.metrics["tmp_unsup"] = self.strategy.tmp_unsup
-                self.metrics = self.evaluation.run(eval_list, self.metrics,
-                                                   metric_logger=log_m)
-                self.losses = self.evaluation.losses
-                eval_list = []
+        # ---- evaluation cadence --------------------------------------
+        if housekeeping and ((i + 1) % self.val_freq) == 0:
+            eval_list.append("val")
+        if housekeeping and ((i + 1) % self.req_freq) == 0:
+            eval_list.append("test")
 
-            # client-LR decay on val plateau.  NOTE: the reference checks
-            # `'val' in eval_list` AFTER clearing the list (server.py:462-469)
-            # making the decay dead code; we implement the documented intent.
-            if ran_val and self.losses and self.losses[0] is not None:
-                log_m("LR for agg. opt.", get_lr(self.worker_trainer.optimizer))
-                if not (self.losses[0] < self.metrics.get("best_val_loss", float("inf"))):
-                    self.lr_weight *= self.lr_decay_factor
-                    print_rank(f"LOG: Client weight of learning rate {self.lr_weight}..")
+        ran_val = "val" in eval_list
+        if len(eval_list) > 0:
+            print_rank(f"Running {eval_list} at itr={i+1}")
+            self.metrics["worker_trainer"] = self.worker_trainer
+            if hasattr(self.strategy, "tmp_unsup") and self.strategy.tmp_unsup is not None:
+                self.metrics["tmp_sup"] = self.strategy.tmp_sup
+                self.metrics["tmp_unsup"] = self.strategy.tmp_unsup
+            self.metrics = self.evaluation.run(eval_list, self.metrics,
+                                               metric_logger=log_m)
+            self.losses = self.evaluation.losses
+            eval_list = []
 
-            # ---- checkpoint / backup / fallback --------------------------
-            if is_chief and housekeeping:
-                self.backup_models(i)
-            if self.fall_back_to_best_model and housekeeping:
-                rt.barrier()
-                self.fall_back_to_prev_best_status()
+        # client-LR decay on val plateau.  NOTE: the reference checks
+        # `'val' in eval_list` AFTER clearing the list (server.py:462-469)
+        # making the decay dead code; we implement the documented intent.
+        if ran_val and self.losses and self.losses[0] is not None:
+            log_m("LR for agg. opt.", get_lr(self.worker_trainer.optimizer))
+            if not (self.losses[0] < self.metrics.get("best_val_loss", float("inf"))):
+                self.lr_weight *= self.lr_decay_factor
+                print_rank(f"LOG: Client weight of learning rate {self.lr_weight}..")
 
-            if housekeeping and len(self.metrics) > 1 and is_chief:
-                update_json_log(self.log_path, {
-                    "i": i + 1,
-                    "best_val_loss": float(self.metrics.get("best_val_loss", float("inf"))),
-                    "best_val_acc": float(self.metrics.get("best_val_acc", 0)),
-                    "best_test_loss": float(self.metrics.get("best_test_loss", float("inf"))),
-                    "best_test_acc": float(self.metrics.get("best_test_acc", 0)),
-                    "weight": float(self.lr_weight),
-                    "num_label_updates": int(self.no_label_updates),
-                })
+        # ---- checkpoint / backup / fallback --------------------------
+        if is_chief and housekeeping:
+            self.backup_models(i)
+        if self.fall_back_to_best_model and housekeeping:
+            rt.barrier()
+            self.fall_back_to_prev_best_status()
 
-            end = time.time()
-            self.run_stats["secsPerRoundHousekeeping"].append(end - begin)
-            self.run_stats["secsPerRoundTotal"].append(
-                self.run_stats["secsPerClientRound"][-1]
-                + self.run_stats["secsPerRoundHousekeeping"][-1])
-            log_m("secsPerRoundTotal", self.run_stats["secsPerRoundTotal"][-1])
+        if housekeeping and len(self.metrics) > 1 and is_chief:
+            update_json_log(self.log_path, {
+                "i": i + 1,
+                "best_val_loss": float(self.metrics.get("best_val_loss", float("inf"))),
+                "best_val_acc": float(self.metrics.get("best_val_acc", 0)),
+                "best_test_loss": float(self.metrics.get("best_test_loss", float("inf"))),
+                "best_test_acc": float(self.metrics.get("best_test_acc", 0)),
+                "weight": float(self.lr_weight),
+                "num_label_updates": int(self.no_label_updates),
+            })
 
-            if self.do_profiling:
-                for metric in ["secsPerClient", "secsPerClientTraining",
-                               "secsPerClientFull", "secsPerClientSetup",
-                               "communicationCosts"]:
-                    vals = self.run_stats[metric][-1]
-                    if vals:
-                        log_m(f"{metric}Mean", float(np.mean(vals)))
-                        log_m(f"{metric}Median", float(np.median(vals)))
-                        log_m(f"{metric}Max", float(max(vals)))
+        end = time.time()
+        self.run_stats["secsPerRoundHousekeeping"].append(end - begin)
+        self.run_stats["secsPerRoundTotal"].append(
+            self.run_stats["secsPerClientRound"][-1]
+            + self.run_stats["secsPerRoundHousekeeping"][-1])
+        log_m("secsPerRoundTotal", self.run_stats["secsPerRoundTotal"][-1])
 
-            if is_chief:
-                for k, v in metrics_payload.items():
-                    log_metric(k, v, step=i)
+        if self.do_profiling:
+            for metric in ["secsPerClient", "secsPerClientTraining",
+                           "secsPerClientFull", "secsPerClientSetup",
+                           "communicationCosts"]:
+                vals = self.run_stats[metric][-1]
+                if vals:
+                    log_m(f"{metric}Mean", float(np.mean(vals)))
+                    log_m(f"{metric}Median", float(np.median(vals)))
+                    log_m(f"{metric}Max", float(max(vals)))
+
+        if is_chief:
+            for k, v in metrics_payload.items():
+                log_metric(k, v, step=i)
 
         if profiler is not None:
             if torch.cuda.is_available():
@@ -571,8 +574,8 @@ class OptimizationServer:
                     shutil.copyfile(src, dst)
 
     def fall_back_to_prev_best_status(self):
-        flush_saves()
         """Reference: server.py:561-578."""
+        flush_saves()
         if not self.fall_back_to_best_model:
             return
         if os.path.exists(self.best_model_path):

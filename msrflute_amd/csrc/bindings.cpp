@@ -355,6 +355,31 @@ void cnn_round(torch::Tensor shard_x, torch::Tensor shard_y,
   int K = (int)counts.numel();
   TORCH_CHECK(stats_out.numel() >= 2 * K && loss_out.numel() >= K);
   TORCH_CHECK(bs >= 1 && bs <= 32);
+  TORCH_CHECK(row_bases.numel() == K && order_offs.numel() == K &&
+              weights.numel() == K && seeds.numel() == K,
+              "per-client metadata lengths must all equal K");
+  // bound every per-client (offset, count) against the shard / order
+  // tensors so inconsistent host metadata raises instead of launching
+  // out-of-bounds device reads
+  {
+    long long n_rows = shard_y.numel();
+    TORCH_CHECK(shard_x.numel() == n_rows * 784,
+                "shard_x rows must match shard_y (784 features each)");
+    long long n_orders = orders_dev.numel();
+    auto rb = row_bases.accessor<int64_t, 1>();
+    auto oo = order_offs.accessor<int64_t, 1>();
+    auto ct = counts.accessor<int64_t, 1>();
+    for (int k = 0; k < K; ++k) {
+      TORCH_CHECK(ct[k] >= 0 && rb[k] >= 0 && oo[k] >= 0,
+                  "negative per-client metadata at k=", k);
+      TORCH_CHECK(rb[k] + ct[k] <= n_rows,
+                  "client ", k, ": row_base+count ", rb[k] + ct[k],
+                  " exceeds shard rows ", n_rows);
+      TORCH_CHECK(oo[k] + ct[k] <= n_orders,
+                  "client ", k, ": order_off+count ", oo[k] + ct[k],
+                  " exceeds orders length ", n_orders);
+    }
+  }
   CnnWorkspace ws = slice_ws(work_f, work_i, work_b, work_d, (int)bs, (int)C);
   launch_cnn_round(
       shard_x.data_ptr<float>(),

@@ -79,6 +79,7 @@ class FusedCNNEpoch:
         self.stats_acc = torch.zeros(2, dtype=torch.float32, device=dev)
         self.loss_acc = torch.zeros(1, dtype=torch.float32, device=dev)
         self._order_pin = None
+        self._order_ev = None
 
     def run_epoch(self, shard_x: torch.Tensor, shard_y: torch.Tensor,
                   order_cpu: torch.Tensor, lr: float, seed: int):
@@ -92,8 +93,16 @@ class FusedCNNEpoch:
         assert x.shape[1] == 784, "fused CNN epoch expects 784-feature rows"
         if self._order_pin is None or self._order_pin.numel() < n:
             self._order_pin = torch.empty(n, dtype=torch.int64).pin_memory()
+            self._order_ev = None
+        if self._order_ev is not None:
+            # in lazy-stats mode there is no host sync between clients, so
+            # the previous client's non_blocking H2D of this pinned buffer
+            # may still be queued — wait for it before overwriting
+            self._order_ev.synchronize()
         self._order_pin[:n].copy_(order_cpu)
         order_dev = self._order_pin[:n].to(shard_x.device, non_blocking=True)
+        self._order_ev = torch.cuda.Event()
+        self._order_ev.record()
         self.lr_t.fill_(float(lr))
         self.stats_acc.zero_()
         self.loss_acc.zero_()

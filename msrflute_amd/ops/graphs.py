@@ -171,6 +171,7 @@ class GraphedClientEpoch:
         self.static_y = torch.zeros((n, *y_shape), dtype=y_dtype, device=dev)
         self.static_idx = torch.zeros((n,), dtype=torch.int64, device=dev)
         self.host_idx = torch.zeros((n,), dtype=torch.int64).pin_memory()
+        self._idx_ev = None
         self.stats_acc = torch.zeros(2, device=dev)
         self.loss_acc = torch.zeros((), device=dev)
         self.momentum_buf = arena.new_buffer() if momentum != 0.0 else None
@@ -219,8 +220,14 @@ class GraphedClientEpoch:
         """order_cpu: CPU int64 tensor of length n (fresh shuffle)."""
         self.static_x.copy_(x_shard, non_blocking=True)
         self.static_y.copy_(y_shard, non_blocking=True)
+        if self._idx_ev is not None:
+            # the previous client's non_blocking H2D of host_idx may still
+            # be queued (no host sync between clients in lazy-stats mode)
+            self._idx_ev.synchronize()
         self.host_idx.copy_(order_cpu)
         self.static_idx.copy_(self.host_idx, non_blocking=True)
+        self._idx_ev = torch.cuda.Event()
+        self._idx_ev.record()
         self.graph.replay()
 
 
@@ -231,18 +238,18 @@ def epoch_graph_for(cache: "GraphCache", x_shard, y_shard, bs: int):
         return None
     key = ("epoch", n, bs, tuple(x_shard.shape[1:]), x_shard.dtype,
            tuple(y_shard.shape[1:]), y_shard.dtype)
-    g = cache._graphs.get(key)
-    if g is None:
-        try:
-            g = GraphedClientEpoch(
-                cache.model, cache.arena, cache.lr_t, cache.max_grad_norm,
-                cache.momentum, cache.weight_decay, cache.nesterov,
-                (tuple(x_shard.shape[1:]), x_shard.dtype,
-                 tuple(y_shard.shape[1:])), y_shard.dtype, n, bs)
-        except Exception as e:  # capture-unsafe model op — run eager
-            import logging
-            logging.getLogger().warning(
-                f"epoch-graph capture failed ({e}); falling back to eager")
-            g = None
-        cache._graphs[key] = g
+    if key in cache._graphs:  # a cached None remembers a failed capture
+        return cache._graphs[key]
+    try:
+        g = GraphedClientEpoch(
+            cache.model, cache.arena, cache.lr_t, cache.max_grad_norm,
+            cache.momentum, cache.weight_decay, cache.nesterov,
+            (tuple(x_shard.shape[1:]), x_shard.dtype,
+             tuple(y_shard.shape[1:])), y_shard.dtype, n, bs)
+    except Exception as e:  # capture-unsafe model op — run eager
+        import logging
+        logging.getLogger().warning(
+            f"epoch-graph capture failed ({e}); falling back to eager")
+        g = None
+    cache._graphs[key] = g
     return g
