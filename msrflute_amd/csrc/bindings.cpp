@@ -39,6 +39,8 @@ void launch_gru_gates(const float*, const float*, const float*, float*, int,
 
 struct CnnWorkspace {
   float *xb, *a1, *r2, *a2, *z3, *a3, *dlogits, *dz3, *da2, *dz2, *dz1;
+  float *wsl;  // split-K partial slab (B*18432 floats) — keep in sync
+               // with the definition in fused_cnn.hip
   int *yb;
   unsigned char *pidx, *m2, *m3;
   double *red_partials, *red_acc;
@@ -62,6 +64,22 @@ extern "C" void launch_cnn_epoch(
     CnnWorkspace ws, const float* lr_t, float max_norm, float p1, float p2,
     float* stats_acc, float* loss_acc, unsigned long long seed,
     hipStream_t s, long long row_base);
+extern "C" void launch_conv2_fwd_mfma(const float*, const float*,
+                                      const float*, int, float*, hipStream_t);
+extern "C" void launch_conv2_bwd_x_mfma(const float*, const float*,
+                                        const float*, int, float*,
+                                        hipStream_t);
+extern "C" void launch_conv2_bwd_w_mfma(const float*, const float*, int,
+                                        float*, float*, float*, hipStream_t);
+extern "C" void launch_fc1_fwd_mfma(const float*, const float*, const float*,
+                                    int, float, unsigned long long,
+                                    unsigned long long, float*, float*,
+                                    float*, unsigned char*, hipStream_t);
+extern "C" void launch_fc1_bwd_w_mfma(const float*, const float*, int,
+                                      float*, float*, hipStream_t);
+extern "C" void launch_fc1_bwd_x_mfma(const float*, const float*, int,
+                                      float*, hipStream_t);
+
 extern "C" void launch_cnn_round(
     const float* shard_x, const long long* shard_y, const long long* orders,
     const long long* row_bases, const long long* order_offs,
@@ -314,6 +332,7 @@ static CnnWorkspace slice_ws(torch::Tensor& work_f, torch::Tensor& work_i,
   ws.da2 = f;           f += (long long)B * 9216;
   ws.dz2 = f;           f += (long long)B * 36864;
   ws.dz1 = f;           f += (long long)B * 21632;
+  ws.wsl = f;           f += (long long)B * 18432;  // split-K partial slab
   TORCH_CHECK(f - work_f.data_ptr<float>() <= work_f.numel());
   ws.yb = work_i.data_ptr<int>();
   unsigned char* u = work_b.data_ptr<unsigned char>();
@@ -480,7 +499,93 @@ torch::Tensor gru_gates(torch::Tensor g_i, torch::Tensor g_h,
 
 }  // namespace
 
+// ---------------------------------------------------------------------------
+// per-kernel MFMA debug entries (numerics tests vs torch references;
+// the epoch driver calls the same kernels internally)
+// ---------------------------------------------------------------------------
+torch::Tensor dbg_conv2_fwd_mfma(torch::Tensor a1, torch::Tensor w2,
+                                 torch::Tensor b2, int64_t B) {
+  check_flat(a1, "a1"); check_flat(w2, "w2"); check_flat(b2, "b2");
+  TORCH_CHECK(a1.numel() >= B * 21632 && w2.numel() == 18432 &&
+              b2.numel() == 64);
+  auto r2 = torch::empty({B * 36864}, a1.options());
+  launch_conv2_fwd_mfma(a1.data_ptr<float>(), w2.data_ptr<float>(),
+                        b2.data_ptr<float>(), (int)B, r2.data_ptr<float>(),
+                        cur_stream());
+  return r2;
+}
+
+torch::Tensor dbg_conv2_bwd_x_mfma(torch::Tensor dz2, torch::Tensor w2,
+                                   torch::Tensor a1, int64_t B) {
+  check_flat(dz2, "dz2"); check_flat(w2, "w2"); check_flat(a1, "a1");
+  TORCH_CHECK(dz2.numel() >= B * 36864 && a1.numel() >= B * 21632);
+  auto dz1 = torch::empty({B * 21632}, dz2.options());
+  launch_conv2_bwd_x_mfma(dz2.data_ptr<float>(), w2.data_ptr<float>(),
+                          a1.data_ptr<float>(), (int)B,
+                          dz1.data_ptr<float>(), cur_stream());
+  return dz1;
+}
+
+std::vector<torch::Tensor> dbg_conv2_bwd_w_mfma(torch::Tensor dz2,
+                                                torch::Tensor a1, int64_t B) {
+  check_flat(dz2, "dz2"); check_flat(a1, "a1");
+  auto slab = torch::empty({B * 18432}, dz2.options());
+  auto dw2 = torch::empty({18432}, dz2.options());
+  auto db2 = torch::empty({64}, dz2.options());
+  launch_conv2_bwd_w_mfma(dz2.data_ptr<float>(), a1.data_ptr<float>(),
+                          (int)B, slab.data_ptr<float>(),
+                          dw2.data_ptr<float>(), db2.data_ptr<float>(),
+                          cur_stream());
+  return {dw2, db2};
+}
+
+std::vector<torch::Tensor> dbg_fc1_fwd_mfma(torch::Tensor a2,
+                                            torch::Tensor w3,
+                                            torch::Tensor b3, int64_t B,
+                                            double p2, int64_t seed,
+                                            int64_t offset) {
+  check_flat(a2, "a2"); check_flat(w3, "w3"); check_flat(b3, "b3");
+  TORCH_CHECK(a2.numel() >= B * 9216 && w3.numel() == 128 * 9216);
+  auto slab = torch::empty({16 * B * 128}, a2.options());
+  auto z3 = torch::empty({B * 128}, a2.options());
+  auto a3 = torch::empty({B * 128}, a2.options());
+  auto m3 = torch::empty({B * 128}, a2.options().dtype(torch::kUInt8));
+  launch_fc1_fwd_mfma(a2.data_ptr<float>(), w3.data_ptr<float>(),
+                      b3.data_ptr<float>(), (int)B, (float)p2,
+                      (unsigned long long)seed, (unsigned long long)offset,
+                      slab.data_ptr<float>(), z3.data_ptr<float>(),
+                      a3.data_ptr<float>(), m3.data_ptr<unsigned char>(),
+                      cur_stream());
+  return {z3, a3, m3};
+}
+
+std::vector<torch::Tensor> dbg_fc1_bwd_w_mfma(torch::Tensor dz3,
+                                              torch::Tensor a2, int64_t B) {
+  check_flat(dz3, "dz3"); check_flat(a2, "a2");
+  auto dw3 = torch::empty({128 * 9216}, dz3.options());
+  auto db3 = torch::empty({128}, dz3.options());
+  launch_fc1_bwd_w_mfma(dz3.data_ptr<float>(), a2.data_ptr<float>(), (int)B,
+                        dw3.data_ptr<float>(), db3.data_ptr<float>(),
+                        cur_stream());
+  return {dw3, db3};
+}
+
+torch::Tensor dbg_fc1_bwd_x_mfma(torch::Tensor dz3, torch::Tensor w3,
+                                 int64_t B) {
+  check_flat(dz3, "dz3"); check_flat(w3, "w3");
+  auto da2 = torch::empty({B * 9216}, dz3.options());
+  launch_fc1_bwd_x_mfma(dz3.data_ptr<float>(), w3.data_ptr<float>(), (int)B,
+                        da2.data_ptr<float>(), cur_stream());
+  return da2;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("dbg_conv2_fwd_mfma", &dbg_conv2_fwd_mfma);
+  m.def("dbg_conv2_bwd_x_mfma", &dbg_conv2_bwd_x_mfma);
+  m.def("dbg_conv2_bwd_w_mfma", &dbg_conv2_bwd_w_mfma);
+  m.def("dbg_fc1_fwd_mfma", &dbg_fc1_fwd_mfma);
+  m.def("dbg_fc1_bwd_w_mfma", &dbg_fc1_bwd_w_mfma);
+  m.def("dbg_fc1_bwd_x_mfma", &dbg_fc1_bwd_x_mfma);
   m.doc() = "msrflute_amd gfx950 flat-arena kernels";
   m.def("pseudo_grad", &pseudo_grad);
   m.def("axpy", &axpy);

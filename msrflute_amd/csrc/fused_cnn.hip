@@ -502,6 +502,323 @@ __global__ void k_conv1_bwd_w(const float* __restrict__ x,
 }
 
 // ---------------------------------------------------------------------------
+// MFMA implicit-GEMM forms of the profile-dominant kernels (round-1
+// profile: k_fc1_fwd 18.3%, k_conv2_bwd_x 18.3%, k_conv2_fwd 16.6%,
+// k_conv2_bwd_w 12.5% — profiles/r01_bench_kernel_stats_final.md).
+// gfx950's f32-input MFMA (v_mfma_f32_16x16x4_f32) is EXACT f32 — a
+// k-ordered fmaf chain at the full f32 rate — so these keep fp32
+// numerics while moving the matrix work onto the matrix pipe and leaving
+// the VALU free for address math and epilogues.
+//
+// Fragment maps (guide §3): A[i=l&15][k=l>>4], B[k=l>>4][j=l&15] one f32
+// VGPR each; D[row=(l>>4)*4+r][col=l&15], 4 f32 per lane.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+// conv2 forward as implicit GEMM: out[(b,o),co] = sum_k a1im2col[(b,o),k]
+// * W2^T[k,co], M=576/batch-row (exact 36 16-tiles), N=64, K=288.
+// One block per (b, 64-row M-tile); a1[b] (86.5 KB) staged once in LDS.
+__global__ __launch_bounds__(256)
+void k_conv2_fwd_mfma(const float* __restrict__ a1,
+                      const float* __restrict__ w2,
+                      const float* __restrict__ b2, int B,
+                      float* __restrict__ r2) {
+  __shared__ float lds[32 * 676];
+  int b = blockIdx.x / 9, mt = blockIdx.x % 9;
+  const float* src = a1 + (long long)b * 21632;
+  for (int i = threadIdx.x; i < 21632; i += 256) lds[i] = src[i];
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int m = mt * 64 + w * 16 + (lane & 15);  // output cell o in [0,576)
+  int yy = m / 24, xx = m % 24;
+  int kc = lane >> 4;
+  f32x4 acc[4] = {f32x4{0,0,0,0}, f32x4{0,0,0,0},
+                  f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  for (int k0 = 0; k0 < 288; k0 += 4) {
+    int k = k0 + kc;
+    int ci = k / 9, rem = k % 9, kh = rem / 3, kw = rem % 3;
+    float a = lds[ci * 676 + (yy + kh) * 26 + xx + kw];
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      float bv = w2[(long long)(nt * 16 + (lane & 15)) * 288 + k];
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[nt], 0, 0, 0);
+    }
+  }
+  int om = mt * 64 + w * 16 + (lane >> 4) * 4;
+  int cl = lane & 15;
+  #pragma unroll
+  for (int r = 0; r < 4; ++r)
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      int co = nt * 16 + cl;
+      float v = acc[nt][r] + b2[co];
+      r2[((long long)b * 64 + co) * 576 + om + r] = v > 0.f ? v : 0.f;
+    }
+}
+
+// conv2 backward-data as implicit GEMM: dz1[(b,p,q),ci] = relu'(a1) *
+// sum_k dz2pad[(b,p,q),k] * W2rot[k,ci], M=676/batch-row (11 64-tiles,
+// masked tail), N=32, K=576 (co,kh,kw).  dz2[b] (147 KB) staged in LDS.
+__global__ __launch_bounds__(256)
+void k_conv2_bwd_x_mfma(const float* __restrict__ dz2,
+                        const float* __restrict__ w2,
+                        const float* __restrict__ a1, int B,
+                        float* __restrict__ dz1) {
+  __shared__ float lds[64 * 576];
+  int b = blockIdx.x / 11, mt = blockIdx.x % 11;
+  const float* src = dz2 + (long long)b * 36864;
+  for (int i = threadIdx.x; i < 36864; i += 256) lds[i] = src[i];
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int m = mt * 64 + w * 16 + (lane & 15);  // cell (p,q) in [0,676)
+  int p = m / 26, q = m % 26;
+  bool mrow = m < 676;
+  int kc = lane >> 4;
+  f32x4 acc[2] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  for (int k0 = 0; k0 < 576; k0 += 4) {
+    int k = k0 + kc;
+    int co = k / 9, rem = k % 9, kh = rem / 3, kw = rem % 3;
+    int y = p - kh, x = q - kw;
+    float a = (mrow && y >= 0 && y < 24 && x >= 0 && x < 24)
+                  ? lds[co * 576 + y * 24 + x] : 0.f;
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      float bv = w2[(long long)co * 288 + (nt * 16 + (lane & 15)) * 9
+                    + kh * 3 + kw];
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[nt], 0, 0, 0);
+    }
+  }
+  int om = mt * 64 + w * 16 + (lane >> 4) * 4;
+  int cl = lane & 15;
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    if (om + r >= 676) continue;
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      int ci = nt * 16 + cl;
+      long long o = ((long long)b * 32 + ci) * 676 + om + r;
+      dz1[o] = a1[o] > 0.f ? acc[nt][r] : 0.f;
+    }
+  }
+}
+
+// conv2 weight gradient as GEMM: dW2[co,(ci,kh,kw)] = sum_{b,o}
+// dz2[(b,o),co] * a1im2col[(b,o),(ci,kh,kw)], M=64, N=288, K=B*576,
+// split over b: block (nb in 0..5, b) computes a 64x48 tile of the
+// per-b partial into ws.wsl[b]; k_conv2_bwd_w_fold sums the B slabs in
+// fixed order (deterministic — no atomics).  a1[b] staged in LDS.
+__global__ __launch_bounds__(256)
+void k_conv2_bwd_w_mfma(const float* __restrict__ dz2,
+                        const float* __restrict__ a1, int B,
+                        float* __restrict__ slab) {
+  __shared__ float lds[32 * 676];
+  int nb = blockIdx.x % 6, b = blockIdx.x / 6;
+  const float* src = a1 + (long long)b * 21632;
+  for (int i = threadIdx.x; i < 21632; i += 256) lds[i] = src[i];
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int co = w * 16 + (lane & 15);
+  int kc = lane >> 4;
+  const float* dzb = dz2 + (long long)b * 36864;
+  f32x4 acc[3] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  for (int k0 = 0; k0 < 576; k0 += 4) {
+    int o = k0 + kc;
+    int yy = o / 24, xx = o % 24;
+    float a = dzb[co * 576 + o];
+    #pragma unroll
+    for (int nt = 0; nt < 3; ++nt) {
+      int n = nb * 48 + nt * 16 + (lane & 15);
+      int ci = n / 9, rem = n % 9, kh = rem / 3, kw = rem % 3;
+      float bv = lds[ci * 676 + (yy + kh) * 26 + xx + kw];
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[nt], 0, 0, 0);
+    }
+  }
+  float* out = slab + (long long)b * 18432;
+  int orow = w * 16 + (lane >> 4) * 4;
+  int cl = lane & 15;
+  #pragma unroll
+  for (int r = 0; r < 4; ++r)
+    #pragma unroll
+    for (int nt = 0; nt < 3; ++nt)
+      out[(orow + r) * 288 + nb * 48 + nt * 16 + cl] = acc[nt][r];
+}
+
+// fold the B per-batch-row dW2 slabs (fixed order) + db2 column sums
+__global__ void k_conv2_bwd_w_fold(const float* __restrict__ slab, int B,
+                                   float* __restrict__ dw2) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= 18432) return;
+  float s = 0.f;
+  for (int b = 0; b < B; ++b) s += slab[(long long)b * 18432 + i];
+  dw2[i] = s;
+}
+
+__global__ void k_conv2_bwd_b(const float* __restrict__ dz2, int B,
+                              float* __restrict__ db2) {
+  int co = blockIdx.x;
+  float s = 0.f;
+  for (int t = threadIdx.x; t < B * 576; t += blockDim.x) {
+    int o = t % 576, b = t / 576;
+    s += dz2[((long long)b * 64 + co) * 576 + o];
+  }
+  for (int d = 32; d > 0; d >>= 1) s += __shfl_down(s, d, 64);
+  __shared__ float lds[4];
+  if ((threadIdx.x & 63) == 0) lds[threadIdx.x >> 6] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.f;
+    for (int k = 0; k < (int)blockDim.x / 64; ++k) t += lds[k];
+    db2[co] = t;
+  }
+}
+
+// fc1 forward as split-K MFMA GEMM: z3^T[j,b] = sum_k W3[j,k]*a2[b,k],
+// M=128 (j), N=32 (b, masked to B), K=9216 split into FC1_SPLIT chunks;
+// per-chunk partials land in ws.wsl and k_fc1_fwd_reduce folds them in
+// fixed order + bias + relu + dropout (same Philox stream as the scalar
+// k_fc1_fwd, so masks are bit-identical to round 1's kernel).
+#define FC1_SPLIT 16
+__global__ __launch_bounds__(256)
+void k_fc1_fwd_mfma(const float* __restrict__ a2,
+                    const float* __restrict__ w3, int B,
+                    float* __restrict__ slab) {
+  int s = blockIdx.x;                    // k-chunk
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int kc = lane >> 4, il = lane & 15;
+  int k_base = s * (9216 / FC1_SPLIT);
+  f32x4 acc[2][2] = {{f32x4{0,0,0,0}, f32x4{0,0,0,0}},
+                     {f32x4{0,0,0,0}, f32x4{0,0,0,0}}};
+  for (int k0 = 0; k0 < 9216 / FC1_SPLIT; k0 += 4) {
+    int k = k_base + k0 + kc;
+    float a0 = w3[(long long)(w * 32 + il) * 9216 + k];
+    float a1v = w3[(long long)(w * 32 + 16 + il) * 9216 + k];
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int bu = u * 16 + il;
+      float bv = bu < B ? a2[(long long)bu * 9216 + k] : 0.f;
+      acc[0][u] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, bv, acc[0][u], 0, 0, 0);
+      acc[1][u] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1v, bv, acc[1][u], 0, 0, 0);
+    }
+  }
+  #pragma unroll
+  for (int t = 0; t < 2; ++t)
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int bu = u * 16 + il;
+      if (bu >= B) continue;
+      int j = w * 32 + t * 16 + (lane >> 4) * 4;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        slab[((long long)s * B + bu) * 128 + j + r] = acc[t][u][r];
+    }
+}
+
+__global__ void k_fc1_fwd_reduce(const float* __restrict__ slab,
+                                 const float* __restrict__ b3, int B,
+                                 float p2, unsigned long long seed,
+                                 unsigned long long offset,
+                                 float* __restrict__ z3,
+                                 float* __restrict__ a3,
+                                 unsigned char* __restrict__ m3) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= B * 128) return;
+  int b = i / 128, j = i % 128;
+  float t = b3[j];
+  for (int s = 0; s < FC1_SPLIT; ++s)
+    t += slab[((long long)s * B + b) * 128 + j];
+  z3[i] = t;
+  float r = t > 0.f ? t : 0.f;
+  unsigned char keep = 1;
+  if (p2 > 0.f) {
+    hiprandStatePhilox4_32_10_t st;
+    hiprand_init(seed ^ 0x9e3779b97f4a7c15ull, (unsigned long long)i,
+                 offset, &st);
+    keep = hiprand_uniform(&st) >= p2;
+  }
+  m3[i] = keep;
+  a3[i] = keep ? r / (1.f - p2) : 0.f;
+}
+
+// dW3[j,n] = sum_b dz3[b,j]*a2[b,n]: M=128, N=9216, K=B (<=32, masked).
+// Block = 128x64 tile (wave: 2 m-tiles x 4 n-tiles), grid = 144.
+__global__ __launch_bounds__(256)
+void k_fc1_bwd_w_mfma(const float* __restrict__ dz3,
+                      const float* __restrict__ a2, int B,
+                      float* __restrict__ dw3) {
+  int nblk = blockIdx.x;                  // 64-col slab of N
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int kc = lane >> 4, il = lane & 15;
+  f32x4 acc[2][4] = {{f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}},
+                     {f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}}};
+  for (int k0 = 0; k0 < ((B + 3) & ~3); k0 += 4) {
+    int b = k0 + kc;
+    bool kv = b < B;
+    float a0 = kv ? dz3[(long long)b * 128 + w * 32 + il] : 0.f;
+    float a1v = kv ? dz3[(long long)b * 128 + w * 32 + 16 + il] : 0.f;
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      long long n = (long long)nblk * 64 + nt * 16 + il;
+      float bv = kv ? a2[(long long)b * 9216 + n] : 0.f;
+      acc[0][nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, bv, acc[0][nt], 0, 0, 0);
+      acc[1][nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1v, bv, acc[1][nt], 0, 0, 0);
+    }
+  }
+  #pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    int j = w * 32 + t * 16 + (lane >> 4) * 4;
+    #pragma unroll
+    for (int r = 0; r < 4; ++r)
+      #pragma unroll
+      for (int nt = 0; nt < 4; ++nt)
+        dw3[(long long)(j + r) * 9216 + nblk * 64 + nt * 16 + il] =
+            acc[t][nt][r];
+  }
+}
+
+__global__ void k_fc1_bwd_b(const float* __restrict__ dz3, int B,
+                            float* __restrict__ db3) {
+  int j = blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= 128) return;
+  float s = 0.f;
+  for (int b = 0; b < B; ++b) s += dz3[(long long)b * 128 + j];
+  db3[j] = s;
+}
+
+// da2[b,n] = sum_j dz3[b,j]*W3[j,n]: M=32 (b, masked to B), N=9216,
+// K=128.  Block = 32x64 tile (wave: 1 n-tile of 16 cols x 2 m-tiles),
+// grid = 144.
+__global__ __launch_bounds__(256)
+void k_fc1_bwd_x_mfma(const float* __restrict__ dz3,
+                      const float* __restrict__ w3, int B,
+                      float* __restrict__ da2) {
+  int nblk = blockIdx.x;
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int kc = lane >> 4, il = lane & 15;
+  f32x4 acc[2] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  for (int k0 = 0; k0 < 128; k0 += 4) {
+    int j = k0 + kc;
+    long long n = (long long)nblk * 64 + w * 16 + il;
+    float bv = w3[(long long)j * 9216 + n];
+    #pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      int bu = t * 16 + il;
+      float a = bu < B ? dz3[(long long)bu * 128 + j] : 0.f;
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[t], 0, 0, 0);
+    }
+  }
+  #pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    int bu = t * 16 + (lane >> 4) * 4;
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      if (bu + r >= B) continue;
+      da2[(long long)(bu + r) * 9216 + (long long)nblk * 64 + w * 16 + il] =
+          acc[t][r];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // epoch driver: the ONLY host entry — loops every batch of one client's
 // local epoch, launching fwd/bwd + fused clip-stats + SGD per batch.
 // ---------------------------------------------------------------------------
@@ -517,6 +834,7 @@ void launch_sgd_step(float*, const float*, float*, float, const float*, float,
 struct CnnWorkspace {
   // laid out inside one float buffer by the binding (sizes for B rows)
   float *xb, *a1, *r2, *a2, *z3, *a3, *dlogits, *dz3, *da2, *dz2, *dz1;
+  float *wsl;  // split-K partial slab (B*18432 floats), shared fwd/bwd
   int *yb;
   unsigned char *pidx, *m2, *m3;
   double *red_partials, *red_acc;
@@ -541,14 +859,16 @@ extern "C" void launch_cnn_epoch(
     hipLaunchKernelGGL(k_conv1_fwd, dim3((B * 21632 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.xb, params + o.w1, params + o.b1,
                        B, ws.a1);
-    hipLaunchKernelGGL(k_conv2_fwd, dim3(B * (64 / CONV2_COG)), dim3(FBLK),
+    hipLaunchKernelGGL(k_conv2_fwd_mfma, dim3(B * 9), dim3(FBLK),
                        0, s, ws.a1, params + o.w2, params + o.b2, B, ws.r2);
     hipLaunchKernelGGL(k_pool_drop_fwd, dim3((B * 9216 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.r2, B, p1, seed, off, ws.a2,
                        ws.pidx, ws.m2);
-    hipLaunchKernelGGL(k_fc1_fwd, dim3(B * 128), dim3(FBLK), 0, s,
-                       ws.a2, params + o.w3, params + o.b3, B, p2, seed, off,
-                       ws.z3, ws.a3, ws.m3);
+    hipLaunchKernelGGL(k_fc1_fwd_mfma, dim3(FC1_SPLIT), dim3(FBLK), 0, s,
+                       ws.a2, params + o.w3, B, ws.wsl);
+    hipLaunchKernelGGL(k_fc1_fwd_reduce, dim3((B * 128 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.wsl, params + o.b3, B, p2, seed,
+                       off, ws.z3, ws.a3, ws.m3);
     hipLaunchKernelGGL(k_fc2_loss_fwd, dim3(B), dim3(FBLK),
                        C * (int)sizeof(float), s, ws.a3, params + o.w4,
                        params + o.b4, ws.yb, B, C, ws.dlogits, loss_acc);
@@ -558,18 +878,23 @@ extern "C" void launch_cnn_epoch(
     hipLaunchKernelGGL(k_fc2_bwd_x, dim3((B * 128 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.dlogits, params + o.w4, ws.z3,
                        ws.m3, B, C, p2, ws.dz3);
-    hipLaunchKernelGGL(k_fc1_bwd_w, dim3(2048), dim3(FBLK), 0, s,
-                       ws.dz3, ws.a2, B, grads + o.w3, grads + o.b3);
-    hipLaunchKernelGGL(k_fc1_bwd_x, dim3((B * 9216 + FBLK - 1) / FBLK),
-                       dim3(FBLK), 0, s, ws.dz3, params + o.w3, B, ws.da2);
+    hipLaunchKernelGGL(k_fc1_bwd_w_mfma, dim3(144), dim3(FBLK), 0, s,
+                       ws.dz3, ws.a2, B, grads + o.w3);
+    hipLaunchKernelGGL(k_fc1_bwd_b, dim3(1), dim3(128), 0, s,
+                       ws.dz3, B, grads + o.b3);
+    hipLaunchKernelGGL(k_fc1_bwd_x_mfma, dim3(144), dim3(FBLK), 0, s,
+                       ws.dz3, params + o.w3, B, ws.da2);
     hipLaunchKernelGGL(k_pool_drop_bwd, dim3((B * 9216 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.da2, ws.pidx, ws.m2, ws.r2, B,
                        p1, ws.dz2);
-    hipLaunchKernelGGL(k_conv2_bwd_w, dim3(64 * 32), dim3(FBLK), 0, s,
-                       ws.a1, ws.dz2, B, grads + o.w2, grads + o.b2);
-    hipLaunchKernelGGL(k_conv2_bwd_x, dim3((B * 21632 + FBLK - 1) / FBLK),
-                       dim3(FBLK), 0, s, ws.dz2, params + o.w2, ws.a1, B,
-                       ws.dz1);
+    hipLaunchKernelGGL(k_conv2_bwd_w_mfma, dim3(6 * B), dim3(FBLK), 0, s,
+                       ws.dz2, ws.a1, B, ws.wsl);
+    hipLaunchKernelGGL(k_conv2_bwd_w_fold, dim3((18432 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.wsl, B, grads + o.w2);
+    hipLaunchKernelGGL(k_conv2_bwd_b, dim3(64), dim3(FBLK), 0, s,
+                       ws.dz2, B, grads + o.b2);
+    hipLaunchKernelGGL(k_conv2_bwd_x_mfma, dim3(B * 11), dim3(FBLK),
+                       0, s, ws.dz2, params + o.w2, ws.a1, B, ws.dz1);
     hipLaunchKernelGGL(k_conv1_bwd_w, dim3(32), dim3(1024), 0, s,
                        ws.xb, ws.dz1, B, grads + o.w1, grads + o.b1);
     // fused clip + sufficient stats + SGD on the whole arena
@@ -616,4 +941,52 @@ extern "C" void launch_cnn_round(
     hipLaunchKernelGGL(k_cnn_axpy, dim3(gp), dim3(FBLK), 0, s,
                        round_accum, grads, o.total);
   }
+}
+
+// ---------------------------------------------------------------------------
+// debug/test launchers for the MFMA kernels (numerics tests call these
+// one-by-one against torch references — tests/test_mfma_gpu.py)
+// ---------------------------------------------------------------------------
+extern "C" {
+void launch_conv2_fwd_mfma(const float* a1, const float* w2, const float* b2,
+                           int B, float* r2, hipStream_t s) {
+  hipLaunchKernelGGL(k_conv2_fwd_mfma, dim3(B * 9), dim3(FBLK), 0, s,
+                     a1, w2, b2, B, r2);
+}
+void launch_conv2_bwd_x_mfma(const float* dz2, const float* w2,
+                             const float* a1, int B, float* dz1,
+                             hipStream_t s) {
+  hipLaunchKernelGGL(k_conv2_bwd_x_mfma, dim3(B * 11), dim3(FBLK), 0, s,
+                     dz2, w2, a1, B, dz1);
+}
+void launch_conv2_bwd_w_mfma(const float* dz2, const float* a1, int B,
+                             float* slab, float* dw2, float* db2,
+                             hipStream_t s) {
+  hipLaunchKernelGGL(k_conv2_bwd_w_mfma, dim3(6 * B), dim3(FBLK), 0, s,
+                     dz2, a1, B, slab);
+  hipLaunchKernelGGL(k_conv2_bwd_w_fold, dim3((18432 + FBLK - 1) / FBLK),
+                     dim3(FBLK), 0, s, slab, B, dw2);
+  hipLaunchKernelGGL(k_conv2_bwd_b, dim3(64), dim3(FBLK), 0, s, dz2, B, db2);
+}
+void launch_fc1_fwd_mfma(const float* a2, const float* w3, const float* b3,
+                         int B, float p2, unsigned long long seed,
+                         unsigned long long offset, float* slab, float* z3,
+                         float* a3, unsigned char* m3, hipStream_t s) {
+  hipLaunchKernelGGL(k_fc1_fwd_mfma, dim3(FC1_SPLIT), dim3(FBLK), 0, s,
+                     a2, w3, B, slab);
+  hipLaunchKernelGGL(k_fc1_fwd_reduce, dim3((B * 128 + FBLK - 1) / FBLK),
+                     dim3(FBLK), 0, s, slab, b3, B, p2, seed, offset,
+                     z3, a3, m3);
+}
+void launch_fc1_bwd_w_mfma(const float* dz3, const float* a2, int B,
+                           float* dw3, float* db3, hipStream_t s) {
+  hipLaunchKernelGGL(k_fc1_bwd_w_mfma, dim3(144), dim3(FBLK), 0, s,
+                     dz3, a2, B, dw3);
+  hipLaunchKernelGGL(k_fc1_bwd_b, dim3(1), dim3(128), 0, s, dz3, B, db3);
+}
+void launch_fc1_bwd_x_mfma(const float* dz3, const float* w3, int B,
+                           float* da2, hipStream_t s) {
+  hipLaunchKernelGGL(k_fc1_bwd_x_mfma, dim3(144), dim3(FBLK), 0, s,
+                     dz3, w3, B, da2);
+}
 }
