@@ -41,6 +41,7 @@ struct CnnWorkspace {
   float *xb, *a1, *r2, *a2, *z3, *a3, *dlogits, *dz3, *da2, *dz2, *dz1;
   float *wsl;  // split-K partial slab (B*18432 floats) — keep in sync
                // with the definition in fused_cnn.hip
+  float *w2t, *w2rot;  // per-batch W2 fragment layouts (18432 floats each)
   int *yb;
   unsigned char *pidx, *m2, *m3;
   double *red_partials, *red_acc;
@@ -64,13 +65,15 @@ extern "C" void launch_cnn_epoch(
     CnnWorkspace ws, const float* lr_t, float max_norm, float p1, float p2,
     float* stats_acc, float* loss_acc, unsigned long long seed,
     hipStream_t s, long long row_base);
+extern "C" void launch_w2_layouts(const float*, float*, float*, hipStream_t);
 extern "C" void launch_conv2_fwd_mfma(const float*, const float*,
                                       const float*, int, float*, hipStream_t);
 extern "C" void launch_conv2_bwd_x_mfma(const float*, const float*,
                                         const float*, int, float*,
                                         hipStream_t);
 extern "C" void launch_conv2_bwd_w_mfma(const float*, const float*, int,
-                                        float*, float*, float*, hipStream_t);
+                                        float*, float*, float*, float*,
+                                        hipStream_t);
 extern "C" void launch_fc1_fwd_mfma(const float*, const float*, const float*,
                                     int, float, unsigned long long,
                                     unsigned long long, float*, float*,
@@ -311,6 +314,8 @@ static CnnWorkspace slice_ws(torch::Tensor& work_f, torch::Tensor& work_i,
   ws.dz2 = f;           f += (long long)B * 36864;
   ws.dz1 = f;           f += (long long)B * 21632;
   ws.wsl = f;           f += (long long)B * 18432;  // split-K partial slab
+  ws.w2t = f;           f += 18432;
+  ws.w2rot = f;         f += 18432;
   TORCH_CHECK(f - work_f.data_ptr<float>() <= work_f.numel(),
               "float workspace too small");
   TORCH_CHECK(work_i.numel() >= B, "int workspace too small");
@@ -492,7 +497,11 @@ torch::Tensor dbg_conv2_fwd_mfma(torch::Tensor a1, torch::Tensor w2,
   TORCH_CHECK(a1.numel() >= B * 21632 && w2.numel() == 18432 &&
               b2.numel() == 64);
   auto r2 = torch::empty({B * 36864}, a1.options());
-  launch_conv2_fwd_mfma(a1.data_ptr<float>(), w2.data_ptr<float>(),
+  auto w2t = torch::empty({18432}, a1.options());
+  auto w2rot = torch::empty({18432}, a1.options());
+  launch_w2_layouts(w2.data_ptr<float>(), w2t.data_ptr<float>(),
+                    w2rot.data_ptr<float>(), cur_stream());
+  launch_conv2_fwd_mfma(a1.data_ptr<float>(), w2t.data_ptr<float>(),
                         b2.data_ptr<float>(), (int)B, r2.data_ptr<float>(),
                         cur_stream());
   return r2;
@@ -503,7 +512,11 @@ torch::Tensor dbg_conv2_bwd_x_mfma(torch::Tensor dz2, torch::Tensor w2,
   check_flat(dz2, "dz2"); check_flat(w2, "w2"); check_flat(a1, "a1");
   TORCH_CHECK(dz2.numel() >= B * 36864 && a1.numel() >= B * 21632);
   auto dz1 = torch::empty({B * 21632}, dz2.options());
-  launch_conv2_bwd_x_mfma(dz2.data_ptr<float>(), w2.data_ptr<float>(),
+  auto w2t = torch::empty({18432}, dz2.options());
+  auto w2rot = torch::empty({18432}, dz2.options());
+  launch_w2_layouts(w2.data_ptr<float>(), w2t.data_ptr<float>(),
+                    w2rot.data_ptr<float>(), cur_stream());
+  launch_conv2_bwd_x_mfma(dz2.data_ptr<float>(), w2rot.data_ptr<float>(),
                           a1.data_ptr<float>(), (int)B,
                           dz1.data_ptr<float>(), cur_stream());
   return dz1;
@@ -513,10 +526,12 @@ std::vector<torch::Tensor> dbg_conv2_bwd_w_mfma(torch::Tensor dz2,
                                                 torch::Tensor a1, int64_t B) {
   check_flat(dz2, "dz2"); check_flat(a1, "a1");
   auto slab = torch::empty({B * 18432}, dz2.options());
+  auto dz2t = torch::empty({B * 36864}, dz2.options());
   auto dw2 = torch::empty({18432}, dz2.options());
   auto db2 = torch::empty({64}, dz2.options());
   launch_conv2_bwd_w_mfma(dz2.data_ptr<float>(), a1.data_ptr<float>(),
-                          (int)B, slab.data_ptr<float>(),
+                          (int)B, dz2t.data_ptr<float>(),
+                          slab.data_ptr<float>(),
                           dw2.data_ptr<float>(), db2.data_ptr<float>(),
                           cur_stream());
   return {dw2, db2};
@@ -529,7 +544,7 @@ std::vector<torch::Tensor> dbg_fc1_fwd_mfma(torch::Tensor a2,
                                             int64_t offset) {
   check_flat(a2, "a2"); check_flat(w3, "w3"); check_flat(b3, "b3");
   TORCH_CHECK(a2.numel() >= B * 9216 && w3.numel() == 128 * 9216);
-  auto slab = torch::empty({16 * B * 128}, a2.options());
+  auto slab = torch::empty({64 * B * 128}, a2.options());
   auto z3 = torch::empty({B * 128}, a2.options());
   auto a3 = torch::empty({B * 128}, a2.options());
   auto m3 = torch::empty({B * 128}, a2.options().dtype(torch::kUInt8));

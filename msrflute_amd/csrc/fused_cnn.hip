@@ -515,12 +515,39 @@ __global__ void k_conv1_bwd_w(const float* __restrict__ x,
 // ---------------------------------------------------------------------------
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
+// W2 fragment layouts, rebuilt once per batch (the SGD step changes W2):
+// w2t[k=(ci,kh,kw)][co]   — conv2-fwd B operand, lane-contiguous in co
+// w2rot[k=(co,kh,kw)][ci] — conv2-bwd-data B operand, lane-contiguous in ci
+__global__ void k_w2_layouts(const float* __restrict__ w2,
+                             float* __restrict__ w2t,
+                             float* __restrict__ w2rot) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= 18432) return;
+  int co = i / 288, k = i % 288;
+  float v = w2[i];
+  w2t[k * 64 + co] = v;
+  int ci = k / 9, rem = k % 9;
+  w2rot[(co * 9 + rem) * 32 + ci] = v;
+}
+
+// dz2 transposed to [b][o][co] (written into ws.r2, which is free after
+// pool_drop_bwd) — conv2-bwd-weight A operand, lane-contiguous in co
+__global__ void k_dz2_transpose(const float* __restrict__ dz2, int B,
+                                float* __restrict__ dz2t) {
+  int total = B * 36864;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    int b = i / 36864, r = i % 36864, o = r / 64, co = r % 64;
+    dz2t[i] = dz2[((long long)b * 64 + co) * 576 + o];
+  }
+}
+
 // conv2 forward as implicit GEMM: out[(b,o),co] = sum_k a1im2col[(b,o),k]
 // * W2^T[k,co], M=576/batch-row (exact 36 16-tiles), N=64, K=288.
 // One block per (b, 64-row M-tile); a1[b] (86.5 KB) staged once in LDS.
 __global__ __launch_bounds__(256)
 void k_conv2_fwd_mfma(const float* __restrict__ a1,
-                      const float* __restrict__ w2,
+                      const float* __restrict__ w2t,
                       const float* __restrict__ b2, int B,
                       float* __restrict__ r2) {
   __shared__ float lds[32 * 676];
@@ -538,9 +565,10 @@ void k_conv2_fwd_mfma(const float* __restrict__ a1,
     int k = k0 + kc;
     int ci = k / 9, rem = k % 9, kh = rem / 3, kw = rem % 3;
     float a = lds[ci * 676 + (yy + kh) * 26 + xx + kw];
+    const float* wrow = w2t + (long long)k * 64 + (lane & 15);
     #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
-      float bv = w2[(long long)(nt * 16 + (lane & 15)) * 288 + k];
+      float bv = wrow[nt * 16];  // lane-contiguous 64B group reads
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[nt], 0, 0, 0);
     }
   }
@@ -561,7 +589,7 @@ void k_conv2_fwd_mfma(const float* __restrict__ a1,
 // masked tail), N=32, K=576 (co,kh,kw).  dz2[b] (147 KB) staged in LDS.
 __global__ __launch_bounds__(256)
 void k_conv2_bwd_x_mfma(const float* __restrict__ dz2,
-                        const float* __restrict__ w2,
+                        const float* __restrict__ w2rot,
                         const float* __restrict__ a1, int B,
                         float* __restrict__ dz1) {
   __shared__ float lds[64 * 576];
@@ -581,10 +609,10 @@ void k_conv2_bwd_x_mfma(const float* __restrict__ dz2,
     int y = p - kh, x = q - kw;
     float a = (mrow && y >= 0 && y < 24 && x >= 0 && x < 24)
                   ? lds[co * 576 + y * 24 + x] : 0.f;
+    const float* wrow = w2rot + (long long)k * 32 + (lane & 15);
     #pragma unroll
     for (int nt = 0; nt < 2; ++nt) {
-      float bv = w2[(long long)co * 288 + (nt * 16 + (lane & 15)) * 9
-                    + kh * 3 + kw];
+      float bv = wrow[nt * 16];  // lane-contiguous 64B group reads
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[nt], 0, 0, 0);
     }
   }
@@ -608,7 +636,7 @@ void k_conv2_bwd_x_mfma(const float* __restrict__ dz2,
 // per-b partial into ws.wsl[b]; k_conv2_bwd_w_fold sums the B slabs in
 // fixed order (deterministic — no atomics).  a1[b] staged in LDS.
 __global__ __launch_bounds__(256)
-void k_conv2_bwd_w_mfma(const float* __restrict__ dz2,
+void k_conv2_bwd_w_mfma(const float* __restrict__ dz2t,
                         const float* __restrict__ a1, int B,
                         float* __restrict__ slab) {
   __shared__ float lds[32 * 676];
@@ -619,12 +647,12 @@ void k_conv2_bwd_w_mfma(const float* __restrict__ dz2,
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
   int co = w * 16 + (lane & 15);
   int kc = lane >> 4;
-  const float* dzb = dz2 + (long long)b * 36864;
+  const float* dzb = dz2t + (long long)b * 36864;  // [o][co] layout
   f32x4 acc[3] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}};
   for (int k0 = 0; k0 < 576; k0 += 4) {
     int o = k0 + kc;
     int yy = o / 24, xx = o % 24;
-    float a = dzb[co * 576 + o];
+    float a = dzb[o * 64 + co];  // lane-contiguous 64B group reads
     #pragma unroll
     for (int nt = 0; nt < 3; ++nt) {
       int n = nb * 48 + nt * 16 + (lane & 15);
@@ -677,25 +705,39 @@ __global__ void k_conv2_bwd_b(const float* __restrict__ dz2, int B,
 // per-chunk partials land in ws.wsl and k_fc1_fwd_reduce folds them in
 // fixed order + bias + relu + dropout (same Philox stream as the scalar
 // k_fc1_fwd, so masks are bit-identical to round 1's kernel).
-#define FC1_SPLIT 16
+#define FC1_SPLIT 64
+#define FC1_CH (9216 / FC1_SPLIT)  /* 144 k per chunk */
+#define FC1_LD (FC1_CH + 1)        /* LDS row stride 145: odd multiplier of
+                                      16 mod 32 -> conflict-free group reads */
 __global__ __launch_bounds__(256)
 void k_fc1_fwd_mfma(const float* __restrict__ a2,
                     const float* __restrict__ w3, int B,
                     float* __restrict__ slab) {
-  int s = blockIdx.x;                    // k-chunk
+  __shared__ float lw[128 * FC1_LD];  // W3 chunk [128][144] (72.5 KB)
+  __shared__ float la[32 * FC1_LD];   // a2 chunk [32][144]  (18.1 KB)
+  int s = blockIdx.x;                 // k-chunk
+  int k_base = s * FC1_CH;
+  for (int i = threadIdx.x; i < 128 * FC1_CH; i += 256) {
+    int row = i / FC1_CH, kk = i % FC1_CH;  // 576 B coalesced runs per row
+    lw[row * FC1_LD + kk] = w3[(long long)row * 9216 + k_base + kk];
+  }
+  for (int i = threadIdx.x; i < 32 * FC1_CH; i += 256) {
+    int bu = i / FC1_CH, kk = i % FC1_CH;
+    la[bu * FC1_LD + kk] = bu < B ? a2[(long long)bu * 9216 + k_base + kk]
+                                  : 0.f;
+  }
+  __syncthreads();
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
   int kc = lane >> 4, il = lane & 15;
-  int k_base = s * (9216 / FC1_SPLIT);
   f32x4 acc[2][2] = {{f32x4{0,0,0,0}, f32x4{0,0,0,0}},
                      {f32x4{0,0,0,0}, f32x4{0,0,0,0}}};
-  for (int k0 = 0; k0 < 9216 / FC1_SPLIT; k0 += 4) {
-    int k = k_base + k0 + kc;
-    float a0 = w3[(long long)(w * 32 + il) * 9216 + k];
-    float a1v = w3[(long long)(w * 32 + 16 + il) * 9216 + k];
+  for (int k0 = 0; k0 < FC1_CH; k0 += 4) {
+    int k = k0 + kc;
+    float a0 = lw[(w * 32 + il) * FC1_LD + k];
+    float a1v = lw[(w * 32 + 16 + il) * FC1_LD + k];
     #pragma unroll
     for (int u = 0; u < 2; ++u) {
-      int bu = u * 16 + il;
-      float bv = bu < B ? a2[(long long)bu * 9216 + k] : 0.f;
+      float bv = la[(u * 16 + il) * FC1_LD + k];
       acc[0][u] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, bv, acc[0][u], 0, 0, 0);
       acc[1][u] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1v, bv, acc[1][u], 0, 0, 0);
     }
@@ -835,6 +877,7 @@ struct CnnWorkspace {
   // laid out inside one float buffer by the binding (sizes for B rows)
   float *xb, *a1, *r2, *a2, *z3, *a3, *dlogits, *dz3, *da2, *dz2, *dz1;
   float *wsl;  // split-K partial slab (B*18432 floats), shared fwd/bwd
+  float *w2t, *w2rot;  // per-batch W2 fragment layouts (18432 floats each)
   int *yb;
   unsigned char *pidx, *m2, *m3;
   double *red_partials, *red_acc;
@@ -881,8 +924,11 @@ extern "C" void launch_cnn_epoch(
                        dim3(FBLK), 0, s, ws.xb, params + o.w1, params + o.b1,
                        B, ws.a1);
     EPOCH_CHK("k_conv1_fwd");
+    hipLaunchKernelGGL(k_w2_layouts, dim3((18432 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, params + o.w2, ws.w2t, ws.w2rot);
+    EPOCH_CHK("k_w2_layouts");
     hipLaunchKernelGGL(k_conv2_fwd_mfma, dim3(B * 9), dim3(FBLK),
-                       0, s, ws.a1, params + o.w2, params + o.b2, B, ws.r2);
+                       0, s, ws.a1, ws.w2t, params + o.b2, B, ws.r2);
     EPOCH_CHK("k_conv2_fwd_mfma");
     hipLaunchKernelGGL(k_pool_drop_fwd, dim3((B * 9216 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.r2, B, p1, seed, off, ws.a2,
@@ -920,8 +966,11 @@ extern "C" void launch_cnn_epoch(
                        dim3(FBLK), 0, s, ws.da2, ws.pidx, ws.m2, ws.r2, B,
                        p1, ws.dz2);
     EPOCH_CHK("k_pool_drop_bwd");
+    hipLaunchKernelGGL(k_dz2_transpose, dim3((B * 36864 + FBLK - 1) / FBLK),
+                       dim3(FBLK), 0, s, ws.dz2, B, ws.r2);
+    EPOCH_CHK("k_dz2_transpose");
     hipLaunchKernelGGL(k_conv2_bwd_w_mfma, dim3(6 * B), dim3(FBLK), 0, s,
-                       ws.dz2, ws.a1, B, ws.wsl);
+                       ws.r2, ws.a1, B, ws.wsl);
     EPOCH_CHK("k_conv2_bwd_w_mfma");
     hipLaunchKernelGGL(k_conv2_bwd_w_fold, dim3((18432 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.wsl, B, grads + o.w2);
@@ -930,7 +979,7 @@ extern "C" void launch_cnn_epoch(
                        ws.dz2, B, grads + o.b2);
     EPOCH_CHK("k_conv2_bwd_b");
     hipLaunchKernelGGL(k_conv2_bwd_x_mfma, dim3(B * 11), dim3(FBLK),
-                       0, s, ws.dz2, params + o.w2, ws.a1, B, ws.dz1);
+                       0, s, ws.dz2, ws.w2rot, ws.a1, B, ws.dz1);
     EPOCH_CHK("k_conv2_bwd_x_mfma");
     hipLaunchKernelGGL(k_conv1_bwd_w, dim3(32), dim3(1024), 0, s,
                        ws.xb, ws.dz1, B, grads + o.w1, grads + o.b1);
@@ -989,22 +1038,29 @@ extern "C" void launch_cnn_round(
 // one-by-one against torch references — tests/test_mfma_gpu.py)
 // ---------------------------------------------------------------------------
 extern "C" {
-void launch_conv2_fwd_mfma(const float* a1, const float* w2, const float* b2,
+void launch_w2_layouts(const float* w2, float* w2t, float* w2rot,
+                       hipStream_t s) {
+  hipLaunchKernelGGL(k_w2_layouts, dim3((18432 + FBLK - 1) / FBLK),
+                     dim3(FBLK), 0, s, w2, w2t, w2rot);
+}
+void launch_conv2_fwd_mfma(const float* a1, const float* w2t, const float* b2,
                            int B, float* r2, hipStream_t s) {
   hipLaunchKernelGGL(k_conv2_fwd_mfma, dim3(B * 9), dim3(FBLK), 0, s,
-                     a1, w2, b2, B, r2);
+                     a1, w2t, b2, B, r2);
 }
-void launch_conv2_bwd_x_mfma(const float* dz2, const float* w2,
+void launch_conv2_bwd_x_mfma(const float* dz2, const float* w2rot,
                              const float* a1, int B, float* dz1,
                              hipStream_t s) {
   hipLaunchKernelGGL(k_conv2_bwd_x_mfma, dim3(B * 11), dim3(FBLK), 0, s,
-                     dz2, w2, a1, B, dz1);
+                     dz2, w2rot, a1, B, dz1);
 }
 void launch_conv2_bwd_w_mfma(const float* dz2, const float* a1, int B,
-                             float* slab, float* dw2, float* db2,
+                             float* dz2t, float* slab, float* dw2, float* db2,
                              hipStream_t s) {
+  hipLaunchKernelGGL(k_dz2_transpose, dim3((B * 36864 + FBLK - 1) / FBLK),
+                     dim3(FBLK), 0, s, dz2, B, dz2t);
   hipLaunchKernelGGL(k_conv2_bwd_w_mfma, dim3(6 * B), dim3(FBLK), 0, s,
-                     dz2, a1, B, slab);
+                     dz2t, a1, B, slab);
   hipLaunchKernelGGL(k_conv2_bwd_w_fold, dim3((18432 + FBLK - 1) / FBLK),
                      dim3(FBLK), 0, s, slab, B, dw2);
   hipLaunchKernelGGL(k_conv2_bwd_b, dim3(64), dim3(FBLK), 0, s, dz2, B, db2);

@@ -69,7 +69,7 @@ class FusedCNNEpoch:
         B = self.bs
         n_float = B * (784 + 21632 + 36864 + 9216 + 128 + 128 + self.C
                        + 128 + 9216 + 36864 + 21632
-                       + 18432)  # + split-K partial slab (MFMA kernels)
+                       + 18432) + 2 * 18432  # + split-K slab + W2 layouts
         self.work_f = torch.empty(n_float, dtype=torch.float32, device=dev)
         self.work_i = torch.empty(B, dtype=torch.int32, device=dev)
         self.work_b = torch.empty(B * (9216 * 2 + 128), dtype=torch.uint8,
