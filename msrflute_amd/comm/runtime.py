@@ -63,6 +63,13 @@ class FedRuntime:
         self.size = size()
         self.local_rank = local_rank()
         self.initialized = False
+        # dedicated HIP stream for the round's bulk collectives (grad
+        # all-reduce overlapped with host-side round bookkeeping)
+        self._comm_stream = None
+        # test hook: run the collective code paths even at world_size 1
+        # (a 1-rank RCCL group exercises the device branches on hardware
+        # that only has one GPU — tests/test_rccl_gpu.py)
+        self._force_collectives = False
         if self.size > 1 and not dist.is_initialized():
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
             os.environ.setdefault("MASTER_PORT", "29500")
@@ -74,8 +81,12 @@ class FedRuntime:
             torch.cuda.set_device(self.local_rank % torch.cuda.device_count())
 
     # -- collectives (no-ops at world_size == 1) --------------------------
+    @property
+    def _active(self) -> bool:
+        return self.size > 1 or self._force_collectives
+
     def all_reduce_(self, t: torch.Tensor):
-        if self.size > 1:
+        if self._active:
             if self.backend == "gloo" and t.is_cuda:
                 # gloo cannot reduce device tensors: CPU round-trip
                 host = t.detach().cpu()
@@ -85,8 +96,62 @@ class FedRuntime:
                 dist.all_reduce(t, op=dist.ReduceOp.SUM)
         return t
 
+    # -- overlapped round reduce ------------------------------------------
+    # The round's bulk communication (Σ w·g arena + Σ w scalar) is started
+    # as soon as every local client's gradient is folded, on a dedicated
+    # HIP stream, so it runs UNDER the host-side round bookkeeping (lazy
+    # client-stats finalize + the per-round metadata all_gather) and the
+    # server optimizer step only waits for it at the end
+    # (BASELINE north star: reduce "overlapped with the next client
+    # batch"; reference streams per-client uploads instead —
+    # core/federated.py:344-373).
+    def begin_grad_reduce(self, grad: torch.Tensor, local_weight_sum: float):
+        """Enqueue all_reduce(grad) + all_reduce(Σw); returns a handle for
+        ``finish_grad_reduce``.  No-op handle at world_size 1."""
+        if not self._active:
+            return ("local", None, None, float(local_weight_sum))
+        if self.backend == "nccl" and grad.is_cuda:
+            if self._comm_stream is None:
+                self._comm_stream = torch.cuda.Stream()
+            wsum = torch.tensor([local_weight_sum], dtype=torch.float64,
+                                device=grad.device)
+            ready = torch.cuda.Event()
+            ready.record()  # grad fully folded on the current stream
+            self._comm_stream.wait_event(ready)
+            with torch.cuda.stream(self._comm_stream):
+                # RCCL is stream-ordered: both reduces run on the comm
+                # stream while the default stream keeps working
+                dist.all_reduce(grad, op=dist.ReduceOp.SUM)
+                dist.all_reduce(wsum, op=dist.ReduceOp.SUM)
+                done = torch.cuda.Event()
+                done.record()
+            return ("nccl", done, wsum, None)
+        # gloo: async handles on host tensors
+        host = grad.detach().cpu() if grad.is_cuda else grad
+        h1 = dist.all_reduce(host, op=dist.ReduceOp.SUM, async_op=True)
+        wsum = torch.tensor([local_weight_sum], dtype=torch.float64)
+        h2 = dist.all_reduce(wsum, op=dist.ReduceOp.SUM, async_op=True)
+        return ("gloo", (h1, h2, host if grad.is_cuda else None, grad),
+                wsum, None)
+
+    def finish_grad_reduce(self, handle) -> float:
+        """Join the round reduce; returns the global weight sum.  The
+        reduced gradient is in place in the arena afterwards."""
+        kind, a, wsum, local = handle
+        if kind == "local":
+            return local
+        if kind == "nccl":
+            torch.cuda.current_stream().wait_event(a)
+            return float(wsum.item())  # syncs the scalar only
+        h1, h2, host, grad = a
+        h1.wait()
+        h2.wait()
+        if host is not None:
+            grad.copy_(host)
+        return float(wsum.item())
+
     def broadcast_(self, t: torch.Tensor, src: int = 0):
-        if self.size > 1:
+        if self._active:
             if self.backend == "gloo" and t.is_cuda:
                 host = t.detach().cpu()
                 dist.broadcast(host, src=src)
@@ -96,7 +161,7 @@ class FedRuntime:
         return t
 
     def all_gather_object(self, obj: Any) -> List[Any]:
-        if self.size == 1:
+        if not self._active:
             return [obj]
         out: List[Any] = [None] * self.size
         dist.all_gather_object(out, obj)
@@ -110,7 +175,7 @@ class FedRuntime:
         and truncated after one fused all_gather.  On NCCL this keeps the
         per-round metadata exchange on the xGMI fabric as one small
         collective instead of a pickled object broadcast chain."""
-        if self.size == 1:
+        if not self._active:
             return [t]
         k = t.shape[1] if t.dim() == 2 else 0
         m = max(int(c) for c in counts)
@@ -124,7 +189,7 @@ class FedRuntime:
         return [o[: int(c)].cpu() for o, c in zip(out, counts)]
 
     def barrier(self):
-        if self.size > 1:
+        if self._active:
             if self.backend == "nccl" and torch.cuda.is_available():
                 dist.barrier(device_ids=[torch.cuda.current_device()])
             else:

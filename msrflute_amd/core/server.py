@@ -345,6 +345,13 @@ class OptimizationServer:
         if hasattr(self.executor, "flush"):
             self.executor.flush(self.worker_trainer.arena.grad)
 
+        # start the round's grad + Σweight all-reduce NOW on the comm
+        # stream: it overlaps the lazy-stats finalize and metadata
+        # all_gather below, and combine_payloads joins it (north-star
+        # comm/compute overlap)
+        if hasattr(self.strategy, "begin_aggregation"):
+            self.strategy.begin_aggregation(self.worker_trainer)
+
         # ---- batched finalize of deferred client stats ---------------
         # lazy-stats clients carried device tensors; ONE host transfer
         # materializes every client's loss/Σg/Σg² for this rank

@@ -194,7 +194,11 @@ class DGA(FedAvg):
             local_weight_sum = float(sum(self.client_weights))
 
         rt = self.runtime
-        if rt is not None and rt.size > 1:
+        if self._pending_reduce is not None:
+            # overlapped round reduce already in flight (fast mode)
+            weight_sum = rt.finish_grad_reduce(self._pending_reduce)
+            self._pending_reduce = None
+        elif rt is not None and rt._active:
             rt.all_reduce_(worker_trainer.arena.grad)
             t = torch.tensor([local_weight_sum], dtype=torch.float64,
                              device=worker_trainer.arena.device

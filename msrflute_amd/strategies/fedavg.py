@@ -41,6 +41,7 @@ class FedAvg(BaseStrategy):
             # rank-local accumulators
             self.client_parameters_stack = []  # stacked mode: clones of flat grads
             self.client_weights = []
+            self._pending_reduce = None  # in-flight overlapped round reduce
 
     # -- client side -------------------------------------------------------
     def generate_client_payload(self, trainer):
@@ -100,17 +101,37 @@ class FedAvg(BaseStrategy):
         losses = worker_trainer.run_lr_scheduler(force_run_val=False)
         return losses
 
+    def begin_aggregation(self, worker_trainer):
+        """Start the round's grad + Σweight all-reduce on the comm stream
+        as soon as local clients are folded (fast-aggregation mode), so it
+        overlaps the host-side round bookkeeping; ``combine_payloads``
+        joins it (comm/compute overlap — runtime.begin_grad_reduce)."""
+        if (self.mode != "server" or not self.aggregate_fast
+                or self.runtime is None or self._pending_reduce is not None):
+            return
+        local_weight_sum = float(sum(self.client_weights))
+        self._pending_reduce = self.runtime.begin_grad_reduce(
+            worker_trainer.arena.grad, local_weight_sum)
+
     def _aggregate_gradients(self, worker_trainer, num_clients_curr_iter,
                              client_weights, metric_logger=None):
         """Local stack sum (if stacked), then the round-level collectives:
-        all_reduce(grad arena) + all_reduce(Σweight)."""
+        all_reduce(grad arena) + all_reduce(Σweight) — joined from the
+        overlapped handle when ``begin_aggregation`` already started it."""
+        if self._pending_reduce is not None:
+            weight_sum = self.runtime.finish_grad_reduce(self._pending_reduce)
+            self._pending_reduce = None
+            self.client_weights = []
+            self._last_stack = self.client_parameters_stack
+            self.client_parameters_stack = []
+            return weight_sum
         if not self.aggregate_fast:
             for flat in self.client_parameters_stack:
                 accumulate_flat_grad(worker_trainer, flat)
         local_weight_sum = float(sum(client_weights))
 
         rt = self.runtime
-        if rt is not None and rt.size > 1:
+        if rt is not None and rt._active:
             rt.all_reduce_(worker_trainer.arena.grad)
             t = torch.tensor([local_weight_sum], dtype=torch.float64,
                              device=worker_trainer.arena.device
