@@ -64,6 +64,7 @@ def build_config(args):
             "seed": 1234,
         },
         "client_config": {
+            "mixed_precision": args.dtype if args.dtype != "fp32" else "",
             "parallel_clients": int(os.environ.get("BENCH_PAR", "8")),
             "use_fused_cnn": os.environ.get("BENCH_FUSED", "1") == "1",
             "do_profiling": False,
@@ -92,6 +93,11 @@ def main():
     # benchmark) — default to that so the headline number carries the
     # reference's true per-client load
     ap.add_argument("--samples-per-client", type=int, default=200)
+    ap.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32",
+                    help="bf16 = mixed precision: bf16 MFMA GEMMs (conv2 "
+                         "fwd/bwd, fc1 fwd) with fp32 master/accum "
+                         "(BASELINE config 2); fp32 matches the "
+                         "reference's precision")
     args = ap.parse_args()
 
     from msrflute_amd.comm import runtime as rt_mod
@@ -188,7 +194,7 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": value / REF_ROUNDS_PER_SEC,
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
                 "model": "cnn_femnist (2conv+2fc, 62 classes)",
@@ -203,6 +209,11 @@ def main():
                 "client_lr": 0.1,
                 "peak_gpu_mem_mb": round(peak_mb, 1),
                 "eval_and_checkpoint": "outside timed region",
+                "precision_detail": ("bf16 MFMA GEMMs (conv2 fwd/bwd-data/"
+                                     "bwd-weight, fc1 fwd), fp32 master/"
+                                     "accum/elementwise"
+                                     if args.dtype == "bf16" else
+                                     "fp32 throughout"),
             },
         }))
     rt.shutdown()

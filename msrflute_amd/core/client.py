@@ -158,7 +158,9 @@ class ClientExecutor:
                     p1=drops[0] if len(drops) > 0 else 0.25,
                     p2=drops[1] if len(drops) > 1 else 0.5,
                     max_grad_norm=self.client_config["data_config"]["train"]
-                    .get("max_grad_norm"))
+                    .get("max_grad_norm"),
+                    use_bf16=self.client_config.get(
+                        "mixed_precision", "") == "bf16")
 
         # MIOpen RNN kernels segfault under hipGraph capture (hipblaslt
         # assert -> SIGSEGV, observed with the fedshakespeare LSTM), so
@@ -657,7 +659,7 @@ def _fused_round_impl(ex, client_ids, initial_lr, seeds):
         fc.bs, fc.C, ex.server_arena.data, ex.arena.data, ex.arena.grad,
         ex.round_accum_target, fc.work_f, fc.work_i, fc.work_b, fc.work_d,
         fc.lr_t, fc.max_norm, fc.p1, fc.p2,
-        ex._round_stats, ex._round_loss)
+        ex._round_stats, ex._round_loss, fc.use_bf16)
     outputs = []
     now = time.time()
     for k, cid in enumerate(client_ids):

@@ -64,7 +64,7 @@ extern "C" void launch_cnn_epoch(
     long long n, int bs, int C, float* params, float* grads,
     CnnWorkspace ws, const float* lr_t, float max_norm, float p1, float p2,
     float* stats_acc, float* loss_acc, unsigned long long seed,
-    hipStream_t s, long long row_base);
+    hipStream_t s, long long row_base, int use_bf16);
 extern "C" void launch_w2_layouts(const float*, float*, float*, hipStream_t);
 extern "C" void launch_conv2_fwd_mfma(const float*, const float*,
                                       const float*, int, float*, hipStream_t);
@@ -82,6 +82,22 @@ extern "C" void launch_fc1_bwd_w_mfma(const float*, const float*, int,
                                       float*, float*, hipStream_t);
 extern "C" void launch_fc1_bwd_x_mfma(const float*, const float*, int,
                                       float*, hipStream_t);
+extern "C" void launch_mfma_bf16_probe(const void*, const void*, float*,
+                                       hipStream_t);
+extern "C" void launch_fc1_fwd_mfma_bf16(const float*, const float*,
+                                         const float*, int, float,
+                                         unsigned long long,
+                                         unsigned long long, float*, float*,
+                                         float*, unsigned char*, hipStream_t);
+extern "C" void launch_conv2_fwd_mfma_bf16(const float*, const float*,
+                                           const float*, int, float*,
+                                           hipStream_t);
+extern "C" void launch_conv2_bwd_x_mfma_bf16(const float*, const float*,
+                                             const float*, int, float*,
+                                             hipStream_t);
+extern "C" void launch_conv2_bwd_w_mfma_bf16(const float*, const float*, int,
+                                             float*, float*, float*,
+                                             hipStream_t);
 
 extern "C" void launch_cnn_round(
     const float* shard_x, const long long* shard_y, const long long* orders,
@@ -90,7 +106,8 @@ extern "C" void launch_cnn_round(
     const unsigned long long* seeds, int K, int bs, int C,
     const float* server_params, float* params, float* grads,
     float* round_accum, CnnWorkspace ws, const float* lr_t, float max_norm,
-    float p1, float p2, float* stats_out, float* loss_out, hipStream_t s);
+    float p1, float p2, float* stats_out, float* loss_out, hipStream_t s,
+    int use_bf16);
 
 namespace {
 
@@ -267,7 +284,7 @@ void cnn_epoch(torch::Tensor shard_x, torch::Tensor shard_y,
                torch::Tensor work_b, torch::Tensor work_d,
                torch::Tensor lr_t, double max_norm, double p1, double p2,
                torch::Tensor stats_acc, torch::Tensor loss_acc,
-               int64_t seed) {
+               int64_t seed, bool use_bf16) {
   check_flat(params, "params"); check_flat(grads, "grads");
   check_flat(lr_t, "lr_t"); check_flat(stats_acc, "stats_acc");
   check_flat(loss_acc, "loss_acc"); check_flat(work_f, "work_f");
@@ -294,7 +311,7 @@ void cnn_epoch(torch::Tensor shard_x, torch::Tensor shard_y,
                    lr_t.data_ptr<float>(), (float)max_norm, (float)p1,
                    (float)p2, stats_acc.data_ptr<float>(),
                    loss_acc.data_ptr<float>(), (unsigned long long)seed,
-                   cur_stream(), 0);
+                   cur_stream(), 0, use_bf16 ? 1 : 0);
 }
 
 static CnnWorkspace slice_ws(torch::Tensor& work_f, torch::Tensor& work_i,
@@ -343,7 +360,8 @@ void cnn_round(torch::Tensor shard_x, torch::Tensor shard_y,
                torch::Tensor work_f, torch::Tensor work_i,
                torch::Tensor work_b, torch::Tensor work_d,
                torch::Tensor lr_t, double max_norm, double p1, double p2,
-               torch::Tensor stats_out, torch::Tensor loss_out) {
+               torch::Tensor stats_out, torch::Tensor loss_out,
+               bool use_bf16) {
   check_flat(server_params, "server_params"); check_flat(params, "params");
   check_flat(grads, "grads"); check_flat(round_accum, "round_accum");
   TORCH_CHECK(orders_dev.is_cuda() && orders_dev.scalar_type() == torch::kInt64);
@@ -401,7 +419,7 @@ void cnn_round(torch::Tensor shard_x, torch::Tensor shard_y,
       params.data_ptr<float>(), grads.data_ptr<float>(),
       round_accum.data_ptr<float>(), ws, lr_t.data_ptr<float>(),
       (float)max_norm, (float)p1, (float)p2, stats_out.data_ptr<float>(),
-      loss_out.data_ptr<float>(), cur_stream());
+      loss_out.data_ptr<float>(), cur_stream(), use_bf16 ? 1 : 0);
 }
 
 // fused LSTM sequence recurrence (lstm_seq.hip): forward over all T
@@ -507,6 +525,60 @@ torch::Tensor dbg_conv2_fwd_mfma(torch::Tensor a1, torch::Tensor w2,
   return r2;
 }
 
+torch::Tensor dbg_conv2_fwd_mfma_bf16(torch::Tensor a1, torch::Tensor w2,
+                                      torch::Tensor b2, int64_t B) {
+  check_flat(a1, "a1"); check_flat(w2, "w2"); check_flat(b2, "b2");
+  auto r2 = torch::empty({B * 36864}, a1.options());
+  launch_conv2_fwd_mfma_bf16(a1.data_ptr<float>(), w2.data_ptr<float>(),
+                             b2.data_ptr<float>(), (int)B,
+                             r2.data_ptr<float>(), cur_stream());
+  return r2;
+}
+
+torch::Tensor dbg_conv2_bwd_x_mfma_bf16(torch::Tensor dz2, torch::Tensor w2,
+                                        torch::Tensor a1, int64_t B) {
+  check_flat(dz2, "dz2"); check_flat(w2, "w2"); check_flat(a1, "a1");
+  auto dz1 = torch::empty({B * 21632}, dz2.options());
+  launch_conv2_bwd_x_mfma_bf16(dz2.data_ptr<float>(), w2.data_ptr<float>(),
+                               a1.data_ptr<float>(), (int)B,
+                               dz1.data_ptr<float>(), cur_stream());
+  return dz1;
+}
+
+std::vector<torch::Tensor> dbg_conv2_bwd_w_mfma_bf16(torch::Tensor dz2,
+                                                     torch::Tensor a1,
+                                                     int64_t B) {
+  check_flat(dz2, "dz2"); check_flat(a1, "a1");
+  auto slab = torch::empty({B * 18432}, dz2.options());
+  auto dw2 = torch::empty({18432}, dz2.options());
+  auto db2 = torch::empty({64}, dz2.options());
+  launch_conv2_bwd_w_mfma_bf16(dz2.data_ptr<float>(), a1.data_ptr<float>(),
+                               (int)B, slab.data_ptr<float>(),
+                               dw2.data_ptr<float>(), db2.data_ptr<float>(),
+                               cur_stream());
+  return {dw2, db2};
+}
+
+std::vector<torch::Tensor> dbg_fc1_fwd_mfma_bf16(torch::Tensor a2,
+                                                 torch::Tensor w3,
+                                                 torch::Tensor b3, int64_t B,
+                                                 double p2, int64_t seed,
+                                                 int64_t offset) {
+  check_flat(a2, "a2"); check_flat(w3, "w3"); check_flat(b3, "b3");
+  auto slab = torch::empty({36 * B * 128}, a2.options());
+  auto z3 = torch::empty({B * 128}, a2.options());
+  auto a3 = torch::empty({B * 128}, a2.options());
+  auto m3 = torch::empty({B * 128}, a2.options().dtype(torch::kUInt8));
+  launch_fc1_fwd_mfma_bf16(a2.data_ptr<float>(), w3.data_ptr<float>(),
+                           b3.data_ptr<float>(), (int)B, (float)p2,
+                           (unsigned long long)seed,
+                           (unsigned long long)offset,
+                           slab.data_ptr<float>(), z3.data_ptr<float>(),
+                           a3.data_ptr<float>(), m3.data_ptr<unsigned char>(),
+                           cur_stream());
+  return {z3, a3, m3};
+}
+
 torch::Tensor dbg_conv2_bwd_x_mfma(torch::Tensor dz2, torch::Tensor w2,
                                    torch::Tensor a1, int64_t B) {
   check_flat(dz2, "dz2"); check_flat(w2, "w2"); check_flat(a1, "a1");
@@ -577,7 +649,24 @@ torch::Tensor dbg_fc1_bwd_x_mfma(torch::Tensor dz3, torch::Tensor w3,
   return da2;
 }
 
+torch::Tensor dbg_mfma_bf16_probe(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16 &&
+              A.is_contiguous() && A.numel() == 16 * 32);
+  TORCH_CHECK(B.is_cuda() && B.scalar_type() == torch::kBFloat16 &&
+              B.is_contiguous() && B.numel() == 32 * 16);
+  auto D = torch::empty({16, 16},
+                        A.options().dtype(torch::kFloat32));
+  launch_mfma_bf16_probe(A.data_ptr(), B.data_ptr(), D.data_ptr<float>(),
+                         cur_stream());
+  return D;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("dbg_mfma_bf16_probe", &dbg_mfma_bf16_probe);
+  m.def("dbg_conv2_fwd_mfma_bf16", &dbg_conv2_fwd_mfma_bf16);
+  m.def("dbg_conv2_bwd_x_mfma_bf16", &dbg_conv2_bwd_x_mfma_bf16);
+  m.def("dbg_conv2_bwd_w_mfma_bf16", &dbg_conv2_bwd_w_mfma_bf16);
+  m.def("dbg_fc1_fwd_mfma_bf16", &dbg_fc1_fwd_mfma_bf16);
   m.def("dbg_conv2_fwd_mfma", &dbg_conv2_fwd_mfma);
   m.def("dbg_conv2_bwd_x_mfma", &dbg_conv2_bwd_x_mfma);
   m.def("dbg_conv2_bwd_w_mfma", &dbg_conv2_bwd_w_mfma);

@@ -883,6 +883,307 @@ struct CnnWorkspace {
   double *red_partials, *red_acc;
 };
 
+// ---------------------------------------------------------------------------
+// bf16 GEMM kernels (mixed precision: bf16 MFMA inputs at ~13-16x the
+// f32 MFMA rate, fp32 accumulators, fp32 master weights/outputs — the
+// SGD step, clip/stats, DP/quant hooks all stay fp32).  The GEMM
+// operands are materialized in LDS as bf16 with k-contiguous rows so
+// each lane's 8-element fragment is ONE 16 B ds_read; row strides are
+// padded so 16-lane fragment groups hit distinct banks.
+// Conversion f32->bf16 happens during LDS staging (round-to-nearest via
+// __bf16 cast) — no global bf16 shadow copies needed.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+__device__ __forceinline__ bf16x8 ld_bf16x8(const __bf16* p) {
+  return *reinterpret_cast<const bf16x8*>(p);
+}
+
+// fc1 forward, bf16: z3^T[j,b] partials; split-K 36 (chunk 256 = 8 bf16
+// k-steps).  LDS: W3 chunk [128][256] + a2 chunk [32][256], stride 264.
+#define FC1B_SPLIT 36
+#define FC1B_CH 256
+#define FC1B_LD 264
+__global__ __launch_bounds__(256)
+void k_fc1_fwd_mfma_bf16(const float* __restrict__ a2,
+                         const float* __restrict__ w3, int B,
+                         float* __restrict__ slab) {
+  __shared__ __bf16 lw[128 * FC1B_LD];
+  __shared__ __bf16 la[32 * FC1B_LD];
+  int s = blockIdx.x;
+  int k_base = s * FC1B_CH;
+  for (int i = threadIdx.x; i < 128 * FC1B_CH; i += 256) {
+    int row = i / FC1B_CH, kk = i % FC1B_CH;
+    lw[row * FC1B_LD + kk] = (__bf16)w3[(long long)row * 9216 + k_base + kk];
+  }
+  for (int i = threadIdx.x; i < 32 * FC1B_CH; i += 256) {
+    int bu = i / FC1B_CH, kk = i % FC1B_CH;
+    la[bu * FC1B_LD + kk] =
+        (__bf16)(bu < B ? a2[(long long)bu * 9216 + k_base + kk] : 0.f);
+  }
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int il = lane & 15, kc8 = (lane >> 4) * 8;
+  f32x4 acc[2][2] = {{f32x4{0,0,0,0}, f32x4{0,0,0,0}},
+                     {f32x4{0,0,0,0}, f32x4{0,0,0,0}}};
+  for (int k0 = 0; k0 < FC1B_CH; k0 += 32) {
+    bf16x8 a0 = ld_bf16x8(&lw[(w * 32 + il) * FC1B_LD + k0 + kc8]);
+    bf16x8 a1v = ld_bf16x8(&lw[(w * 32 + 16 + il) * FC1B_LD + k0 + kc8]);
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      bf16x8 bv = ld_bf16x8(&la[(u * 16 + il) * FC1B_LD + k0 + kc8]);
+      acc[0][u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bv, acc[0][u], 0, 0, 0);
+      acc[1][u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1v, bv, acc[1][u], 0, 0, 0);
+    }
+  }
+  #pragma unroll
+  for (int t = 0; t < 2; ++t)
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int bu = u * 16 + il;
+      if (bu >= B) continue;
+      int j = w * 32 + t * 16 + (lane >> 4) * 4;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        slab[((long long)s * B + bu) * 128 + j + r] = acc[t][u][r];
+    }
+}
+
+__global__ void k_fc1_fwd_reduce_bf16(const float* __restrict__ slab,
+                                      const float* __restrict__ b3, int B,
+                                      float p2, unsigned long long seed,
+                                      unsigned long long offset,
+                                      float* __restrict__ z3,
+                                      float* __restrict__ a3,
+                                      unsigned char* __restrict__ m3) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= B * 128) return;
+  int b = i / 128, j = i % 128;
+  float t = b3[j];
+  for (int s = 0; s < FC1B_SPLIT; ++s)
+    t += slab[((long long)s * B + b) * 128 + j];
+  z3[i] = t;
+  float r = t > 0.f ? t : 0.f;
+  unsigned char keep = 1;
+  if (p2 > 0.f) {
+    hiprandStatePhilox4_32_10_t st;
+    hiprand_init(seed ^ 0x9e3779b97f4a7c15ull, (unsigned long long)i,
+                 offset, &st);
+    keep = hiprand_uniform(&st) >= p2;
+  }
+  m3[i] = keep;
+  a3[i] = keep ? r / (1.f - p2) : 0.f;
+}
+
+// conv2 forward, bf16: im2col A tile [64][288] and W2 [64][288] (both
+// k-contiguous bf16, stride 296) built in LDS per block; 36 bf16 MFMAs
+// per wave replace 288 f32 MFMA issues.
+#define C2_LD 296
+__global__ __launch_bounds__(256)
+void k_conv2_fwd_mfma_bf16(const float* __restrict__ a1,
+                           const float* __restrict__ w2,
+                           const float* __restrict__ b2, int B,
+                           float* __restrict__ r2) {
+  __shared__ __bf16 imc[64 * C2_LD];
+  __shared__ __bf16 wb[64 * C2_LD];
+  int b = blockIdx.x / 9, mt = blockIdx.x % 9;
+  const float* a1b = a1 + (long long)b * 21632;
+  for (int i = threadIdx.x; i < 64 * 288; i += 256) {
+    int mr = i / 288, k = i % 288;
+    int m = mt * 64 + mr, yy = m / 24, xx = m % 24;
+    int ci = k / 9, rem = k % 9, kh = rem / 3, kw = rem % 3;
+    imc[mr * C2_LD + k] = (__bf16)a1b[ci * 676 + (yy + kh) * 26 + xx + kw];
+    wb[mr * C2_LD + k] = (__bf16)w2[i];  // mr doubles as co here
+  }
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int il = lane & 15, kc8 = (lane >> 4) * 8;
+  f32x4 acc[4] = {f32x4{0,0,0,0}, f32x4{0,0,0,0},
+                  f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  for (int k0 = 0; k0 < 288; k0 += 32) {
+    bf16x8 a = ld_bf16x8(&imc[(w * 16 + il) * C2_LD + k0 + kc8]);
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      bf16x8 bv = ld_bf16x8(&wb[(nt * 16 + il) * C2_LD + k0 + kc8]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc[nt], 0, 0, 0);
+    }
+  }
+  int om = mt * 64 + w * 16 + (lane >> 4) * 4;
+  #pragma unroll
+  for (int r = 0; r < 4; ++r)
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      int co = nt * 16 + il;
+      float v = acc[nt][r] + b2[co];
+      r2[((long long)b * 64 + co) * 576 + om + r] = v > 0.f ? v : 0.f;
+    }
+}
+
+// conv2 backward-data, bf16: padded-im2col A tile [64][576] of dz2 plus
+// W2^T-per-ci tile [32][576] (k=(co,kh,kw) contiguous), stride 584.
+#define C2B_LD 584
+__global__ __launch_bounds__(256)
+void k_conv2_bwd_x_mfma_bf16(const float* __restrict__ dz2,
+                             const float* __restrict__ w2,
+                             const float* __restrict__ a1, int B,
+                             float* __restrict__ dz1) {
+  __shared__ __bf16 imc[64 * C2B_LD];   // 74.8 KB
+  __shared__ __bf16 wt[32 * C2B_LD];    // 37.4 KB
+  int b = blockIdx.x / 11, mt = blockIdx.x % 11;
+  const float* dzb = dz2 + (long long)b * 36864;
+  for (int i = threadIdx.x; i < 64 * 576; i += 256) {
+    int mr = i / 576, k = i % 576;
+    int m = mt * 64 + mr, p = m / 26, q = m % 26;
+    int co = k / 9, rem = k % 9, kh = rem / 3, kw = rem % 3;
+    int y = p - kh, x = q - kw;
+    imc[mr * C2B_LD + k] =
+        (__bf16)((m < 676 && y >= 0 && y < 24 && x >= 0 && x < 24)
+                     ? dzb[co * 576 + y * 24 + x] : 0.f);
+  }
+  for (int i = threadIdx.x; i < 32 * 576; i += 256) {
+    int ci = i / 576, k = i % 576;
+    int co = k / 9, rem = k % 9;
+    wt[ci * C2B_LD + k] = (__bf16)w2[(long long)co * 288 + ci * 9 + rem];
+  }
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int il = lane & 15, kc8 = (lane >> 4) * 8;
+  f32x4 acc[2] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  for (int k0 = 0; k0 < 576; k0 += 32) {
+    bf16x8 a = ld_bf16x8(&imc[(w * 16 + il) * C2B_LD + k0 + kc8]);
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      bf16x8 bv = ld_bf16x8(&wt[(nt * 16 + il) * C2B_LD + k0 + kc8]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc[nt], 0, 0, 0);
+    }
+  }
+  int om = mt * 64 + w * 16 + (lane >> 4) * 4;
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    if (om + r >= 676) continue;
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      int ci = nt * 16 + il;
+      long long o = ((long long)b * 32 + ci) * 676 + om + r;
+      dz1[o] = a1[o] > 0.f ? acc[nt][r] : 0.f;
+    }
+  }
+}
+
+// conv2 backward-weight, bf16: dz2[b] as bf16 [64][576] (A, k=o
+// contiguous) + im2col^T tile [48][576] of a1 (B); per-b slabs folded
+// deterministically as in the f32 path.
+__global__ __launch_bounds__(256)
+void k_conv2_bwd_w_mfma_bf16(const float* __restrict__ dz2,
+                             const float* __restrict__ a1, int B,
+                             float* __restrict__ slab) {
+  __shared__ __bf16 dzb16[64 * C2B_LD];  // 74.8 KB
+  __shared__ __bf16 imt[48 * C2B_LD];    // 56.1 KB
+  int nb = blockIdx.x % 6, b = blockIdx.x / 6;
+  const float* dzb = dz2 + (long long)b * 36864;
+  const float* a1b = a1 + (long long)b * 21632;
+  for (int i = threadIdx.x; i < 64 * 576; i += 256) {
+    int co = i / 576, o = i % 576;
+    dzb16[co * C2B_LD + o] = (__bf16)dzb[i];
+    (void)o;
+  }
+  for (int i = threadIdx.x; i < 48 * 576; i += 256) {
+    int nr = i / 576, o = i % 576;
+    int n = nb * 48 + nr;
+    int ci = n / 9, rem = n % 9, kh = rem / 3, kw = rem % 3;
+    int yy = o / 24, xx = o % 24;
+    imt[nr * C2B_LD + o] = (__bf16)a1b[ci * 676 + (yy + kh) * 26 + xx + kw];
+  }
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int il = lane & 15, kc8 = (lane >> 4) * 8;
+  f32x4 acc[3] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  for (int k0 = 0; k0 < 576; k0 += 32) {
+    bf16x8 a = ld_bf16x8(&dzb16[(w * 16 + il) * C2B_LD + k0 + kc8]);
+    #pragma unroll
+    for (int nt = 0; nt < 3; ++nt) {
+      bf16x8 bv = ld_bf16x8(&imt[(nt * 16 + il) * C2B_LD + k0 + kc8]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc[nt], 0, 0, 0);
+    }
+  }
+  float* out = slab + (long long)b * 18432;
+  int orow = w * 16 + (lane >> 4) * 4;
+  #pragma unroll
+  for (int r = 0; r < 4; ++r)
+    #pragma unroll
+    for (int nt = 0; nt < 3; ++nt)
+      out[(orow + r) * 288 + nb * 48 + nt * 16 + il] = acc[nt][r];
+}
+
+
+// ---------------------------------------------------------------------------
+// bf16 MFMA probe: one v_mfma_f32_16x16x32_bf16 with the documented
+// fragment maps (A[i=l&15][k=(l>>4)*8+e], B[k=(l>>4)*8+e][j=l&15],
+// D[row=(l>>4)*4+r][col=l&15]) — the numerics test pins the layout
+// before the bf16 kernels build on it.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4_;
+
+__global__ void k_mfma_bf16_probe(const __bf16* __restrict__ A,   // [16][32]
+                                  const __bf16* __restrict__ B,   // [32][16]
+                                  float* __restrict__ D) {        // [16][16]
+  int l = threadIdx.x;
+  bf16x8 a, b;
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    int k = (l >> 4) * 8 + e;
+    a[e] = A[(l & 15) * 32 + k];
+    b[e] = B[k * 16 + (l & 15)];
+  }
+  f32x4 acc = {0, 0, 0, 0};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  #pragma unroll
+  for (int r = 0; r < 4; ++r)
+    D[((l >> 4) * 4 + r) * 16 + (l & 15)] = acc[r];
+}
+
+extern "C" void launch_fc1_fwd_mfma_bf16(const float* a2, const float* w3,
+                              const float* b3, int B, float p2,
+                              unsigned long long seed,
+                              unsigned long long offset, float* slab,
+                              float* z3, float* a3, unsigned char* m3,
+                              hipStream_t s) {
+  hipLaunchKernelGGL(k_fc1_fwd_mfma_bf16, dim3(FC1B_SPLIT), dim3(FBLK), 0, s,
+                     a2, w3, B, slab);
+  hipLaunchKernelGGL(k_fc1_fwd_reduce_bf16, dim3((B * 128 + FBLK - 1) / FBLK),
+                     dim3(FBLK), 0, s, slab, b3, B, p2, seed, offset,
+                     z3, a3, m3);
+}
+extern "C" void launch_conv2_fwd_mfma_bf16(const float* a1, const float* w2,
+                                const float* b2, int B, float* r2,
+                                hipStream_t s) {
+  hipLaunchKernelGGL(k_conv2_fwd_mfma_bf16, dim3(B * 9), dim3(FBLK), 0, s,
+                     a1, w2, b2, B, r2);
+}
+extern "C" void launch_conv2_bwd_x_mfma_bf16(const float* dz2, const float* w2,
+                                  const float* a1, int B, float* dz1,
+                                  hipStream_t s) {
+  hipLaunchKernelGGL(k_conv2_bwd_x_mfma_bf16, dim3(B * 11), dim3(FBLK), 0, s,
+                     dz2, w2, a1, B, dz1);
+}
+extern "C" void launch_conv2_bwd_w_mfma_bf16(const float* dz2, const float* a1, int B,
+                                  float* slab, float* dw2, float* db2,
+                                  hipStream_t s) {
+  hipLaunchKernelGGL(k_conv2_bwd_w_mfma_bf16, dim3(6 * B), dim3(FBLK), 0, s,
+                     dz2, a1, B, slab);
+  hipLaunchKernelGGL(k_conv2_bwd_w_fold, dim3((18432 + FBLK - 1) / FBLK),
+                     dim3(FBLK), 0, s, slab, B, dw2);
+  hipLaunchKernelGGL(k_conv2_bwd_b, dim3(64), dim3(FBLK), 0, s, dz2, B, db2);
+}
+extern "C" void launch_mfma_bf16_probe(const void* A, const void* B, float* D,
+                                       hipStream_t s) {
+  hipLaunchKernelGGL(k_mfma_bf16_probe, dim3(1), dim3(64), 0, s,
+                     (const __bf16*)A, (const __bf16*)B, D);
+}
+
+
 // CNN_EPOCH_DEBUG=1: synchronize + check after every launch so a device
 // fault names the kernel that raised it (diagnostics only — the hot path
 // never syncs)
@@ -907,7 +1208,7 @@ extern "C" void launch_cnn_epoch(
     long long n, int bs, int C, float* params, float* grads,
     CnnWorkspace ws, const float* lr_t, float max_norm, float p1, float p2,
     float* stats_acc, float* loss_acc, unsigned long long seed,
-    hipStream_t s, long long row_base = 0) {
+    hipStream_t s, long long row_base = 0, int use_bf16 = 0) {
   static const bool epoch_dbg = getenv("CNN_EPOCH_DEBUG") != nullptr;
   CnnOffsets o = cnn_offsets(C);
   int n_batches = (int)((n + bs - 1) / bs);
@@ -924,22 +1225,38 @@ extern "C" void launch_cnn_epoch(
                        dim3(FBLK), 0, s, ws.xb, params + o.w1, params + o.b1,
                        B, ws.a1);
     EPOCH_CHK("k_conv1_fwd");
-    hipLaunchKernelGGL(k_w2_layouts, dim3((18432 + FBLK - 1) / FBLK),
-                       dim3(FBLK), 0, s, params + o.w2, ws.w2t, ws.w2rot);
-    EPOCH_CHK("k_w2_layouts");
-    hipLaunchKernelGGL(k_conv2_fwd_mfma, dim3(B * 9), dim3(FBLK),
-                       0, s, ws.a1, ws.w2t, params + o.b2, B, ws.r2);
+    if (use_bf16) {
+      // bf16 kernels read w2 directly (LDS staging converts)
+      hipLaunchKernelGGL(k_conv2_fwd_mfma_bf16, dim3(B * 9), dim3(FBLK),
+                         0, s, ws.a1, params + o.w2, params + o.b2, B, ws.r2);
+    } else {
+      hipLaunchKernelGGL(k_w2_layouts, dim3((18432 + FBLK - 1) / FBLK),
+                         dim3(FBLK), 0, s, params + o.w2, ws.w2t, ws.w2rot);
+      EPOCH_CHK("k_w2_layouts");
+      hipLaunchKernelGGL(k_conv2_fwd_mfma, dim3(B * 9), dim3(FBLK),
+                         0, s, ws.a1, ws.w2t, params + o.b2, B, ws.r2);
+    }
     EPOCH_CHK("k_conv2_fwd_mfma");
     hipLaunchKernelGGL(k_pool_drop_fwd, dim3((B * 9216 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.r2, B, p1, seed, off, ws.a2,
                        ws.pidx, ws.m2);
     EPOCH_CHK("k_pool_drop_fwd");
-    hipLaunchKernelGGL(k_fc1_fwd_mfma, dim3(FC1_SPLIT), dim3(FBLK), 0, s,
-                       ws.a2, params + o.w3, B, ws.wsl);
-    EPOCH_CHK("k_fc1_fwd_mfma");
-    hipLaunchKernelGGL(k_fc1_fwd_reduce, dim3((B * 128 + FBLK - 1) / FBLK),
-                       dim3(FBLK), 0, s, ws.wsl, params + o.b3, B, p2, seed,
-                       off, ws.z3, ws.a3, ws.m3);
+    if (use_bf16) {
+      hipLaunchKernelGGL(k_fc1_fwd_mfma_bf16, dim3(FC1B_SPLIT), dim3(FBLK),
+                         0, s, ws.a2, params + o.w3, B, ws.wsl);
+      EPOCH_CHK("k_fc1_fwd_mfma_bf16");
+      hipLaunchKernelGGL(k_fc1_fwd_reduce_bf16,
+                         dim3((B * 128 + FBLK - 1) / FBLK),
+                         dim3(FBLK), 0, s, ws.wsl, params + o.b3, B, p2, seed,
+                         off, ws.z3, ws.a3, ws.m3);
+    } else {
+      hipLaunchKernelGGL(k_fc1_fwd_mfma, dim3(FC1_SPLIT), dim3(FBLK), 0, s,
+                         ws.a2, params + o.w3, B, ws.wsl);
+      EPOCH_CHK("k_fc1_fwd_mfma");
+      hipLaunchKernelGGL(k_fc1_fwd_reduce, dim3((B * 128 + FBLK - 1) / FBLK),
+                         dim3(FBLK), 0, s, ws.wsl, params + o.b3, B, p2, seed,
+                         off, ws.z3, ws.a3, ws.m3);
+    }
     EPOCH_CHK("k_fc1_fwd_reduce");
     hipLaunchKernelGGL(k_fc2_loss_fwd, dim3(B), dim3(FBLK),
                        C * (int)sizeof(float), s, ws.a3, params + o.w4,
@@ -966,11 +1283,16 @@ extern "C" void launch_cnn_epoch(
                        dim3(FBLK), 0, s, ws.da2, ws.pidx, ws.m2, ws.r2, B,
                        p1, ws.dz2);
     EPOCH_CHK("k_pool_drop_bwd");
-    hipLaunchKernelGGL(k_dz2_transpose, dim3((B * 36864 + FBLK - 1) / FBLK),
-                       dim3(FBLK), 0, s, ws.dz2, B, ws.r2);
-    EPOCH_CHK("k_dz2_transpose");
-    hipLaunchKernelGGL(k_conv2_bwd_w_mfma, dim3(6 * B), dim3(FBLK), 0, s,
-                       ws.r2, ws.a1, B, ws.wsl);
+    if (use_bf16) {
+      hipLaunchKernelGGL(k_conv2_bwd_w_mfma_bf16, dim3(6 * B), dim3(FBLK),
+                         0, s, ws.dz2, ws.a1, B, ws.wsl);
+    } else {
+      hipLaunchKernelGGL(k_dz2_transpose, dim3((B * 36864 + FBLK - 1) / FBLK),
+                         dim3(FBLK), 0, s, ws.dz2, B, ws.r2);
+      EPOCH_CHK("k_dz2_transpose");
+      hipLaunchKernelGGL(k_conv2_bwd_w_mfma, dim3(6 * B), dim3(FBLK), 0, s,
+                         ws.r2, ws.a1, B, ws.wsl);
+    }
     EPOCH_CHK("k_conv2_bwd_w_mfma");
     hipLaunchKernelGGL(k_conv2_bwd_w_fold, dim3((18432 + FBLK - 1) / FBLK),
                        dim3(FBLK), 0, s, ws.wsl, B, grads + o.w2);
@@ -978,8 +1300,12 @@ extern "C" void launch_cnn_epoch(
     hipLaunchKernelGGL(k_conv2_bwd_b, dim3(64), dim3(FBLK), 0, s,
                        ws.dz2, B, grads + o.b2);
     EPOCH_CHK("k_conv2_bwd_b");
-    hipLaunchKernelGGL(k_conv2_bwd_x_mfma, dim3(B * 11), dim3(FBLK),
-                       0, s, ws.dz2, ws.w2rot, ws.a1, B, ws.dz1);
+    if (use_bf16)
+      hipLaunchKernelGGL(k_conv2_bwd_x_mfma_bf16, dim3(B * 11), dim3(FBLK),
+                         0, s, ws.dz2, params + o.w2, ws.a1, B, ws.dz1);
+    else
+      hipLaunchKernelGGL(k_conv2_bwd_x_mfma, dim3(B * 11), dim3(FBLK),
+                         0, s, ws.dz2, ws.w2rot, ws.a1, B, ws.dz1);
     EPOCH_CHK("k_conv2_bwd_x_mfma");
     hipLaunchKernelGGL(k_conv1_bwd_w, dim3(32), dim3(1024), 0, s,
                        ws.xb, ws.dz1, B, grads + o.w1, grads + o.b1);
@@ -1015,7 +1341,7 @@ extern "C" void launch_cnn_round(
     const float* server_params, float* params, float* grads,
     float* round_accum, CnnWorkspace ws, const float* lr_t, float max_norm,
     float p1, float p2, float* stats_out, float* loss_out,
-    hipStream_t s) {
+    hipStream_t s, int use_bf16) {
   CnnOffsets o = cnn_offsets(C);
   int gp = (int)((o.total + FBLK - 1) / FBLK);
   if (gp > 2048) gp = 2048;
@@ -1025,7 +1351,7 @@ extern "C" void launch_cnn_round(
     launch_cnn_epoch(shard_x, shard_y, orders + order_offs[k], counts[k],
                      bs, C, params, grads, ws, lr_t, max_norm, p1, p2,
                      stats_out + 2 * k, loss_out + k, seeds[k], s,
-                     row_bases[k]);
+                     row_bases[k], use_bf16);
     hipLaunchKernelGGL(k_cnn_pseudo_grad, dim3(gp), dim3(FBLK), 0, s,
                        grads, server_params, params, weights[k], o.total);
     hipLaunchKernelGGL(k_cnn_axpy, dim3(gp), dim3(FBLK), 0, s,

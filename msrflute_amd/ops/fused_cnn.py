@@ -58,13 +58,17 @@ class FusedCNNEpoch:
 
     def __init__(self, arena: ParameterArena, num_classes: int, bs: int,
                  p1: float = 0.25, p2: float = 0.5,
-                 max_grad_norm: Optional[float] = None):
+                 max_grad_norm: Optional[float] = None,
+                 use_bf16: bool = False):
         assert HAS_EXT and arena.device.type == "cuda"
         self.arena = arena
         self.C = int(num_classes)
         self.bs = int(bs)
         self.p1, self.p2 = float(p1), float(p2)
         self.max_norm = float(max_grad_norm) if max_grad_norm else -1.0
+        # mixed precision: bf16 MFMA GEMMs (conv2 fwd/bwd, fc1 fwd) with
+        # fp32 accumulators/master weights; everything else stays fp32
+        self.use_bf16 = bool(use_bf16)
         dev = arena.device
         B = self.bs
         n_float = B * (784 + 21632 + 36864 + 9216 + 128 + 128 + self.C
@@ -111,5 +115,6 @@ class FusedCNNEpoch:
                      self.bs, self.C, self.arena.data, self.arena.grad,
                      self.work_f, self.work_i, self.work_b, self.work_d,
                      self.lr_t, self.max_norm, self.p1, self.p2,
-                     self.stats_acc, self.loss_acc, int(seed))
+                     self.stats_acc, self.loss_acc, int(seed),
+                     self.use_bf16)
         return n, (n + self.bs - 1) // self.bs

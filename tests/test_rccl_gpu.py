@@ -76,9 +76,11 @@ def test_full_round_through_server_with_forced_collectives(rt):
     # run in a subprocess so the module-level dataset cache and runtime
     # singleton of other tests don't interfere
     code = """
-import os, torch, torch.distributed as dist
-os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-os.environ.setdefault("MASTER_PORT", "29573")
+import os, socket, torch, torch.distributed as dist
+os.environ["MASTER_ADDR"] = "127.0.0.1"
+s = socket.socket(); s.bind(("127.0.0.1", 0))
+os.environ["MASTER_PORT"] = str(s.getsockname()[1])  # parent holds 29571
+s.close()
 dist.init_process_group("nccl", rank=0, world_size=1)
 import bench
 import sys
