@@ -425,16 +425,16 @@ void cnn_round(torch::Tensor shard_x, torch::Tensor shard_y,
 // fused LSTM sequence recurrence (lstm_seq.hip): forward over all T
 // steps in one launch; returns (h_seq, activated gates, cell states)
 std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xp,
-                                        torch::Tensor w_hh_t) {
-  check_flat(xp, "xp"); check_flat(w_hh_t, "w_hh_t");
+                                        torch::Tensor w_hh) {
+  check_flat(xp, "xp"); check_flat(w_hh, "w_hh");
   TORCH_CHECK(xp.dim() == 3 && xp.size(2) == 4 * 256 &&
-              w_hh_t.size(0) == 256 && w_hh_t.size(1) == 4 * 256,
-              "fused LSTM supports hidden size 256; pass W_hh transposed");
+              w_hh.size(0) == 256 && w_hh.size(1) == 4 * 256,
+              "fused LSTM supports hidden size 256; pass W_hh TRANSPOSED");
   long long B = xp.size(0), T = xp.size(1);
   auto h_seq = torch::empty({B, T, 256}, xp.options());
   auto gates = torch::empty({B, T, 4 * 256}, xp.options());
   auto c_seq = torch::empty({B, T, 256}, xp.options());
-  launch_lstm_seq_fwd(xp.data_ptr<float>(), w_hh_t.data_ptr<float>(),
+  launch_lstm_seq_fwd(xp.data_ptr<float>(), w_hh.data_ptr<float>(),
                       h_seq.data_ptr<float>(), gates.data_ptr<float>(),
                       c_seq.data_ptr<float>(), (int)B, (int)T, 256,
                       cur_stream());
@@ -445,6 +445,8 @@ torch::Tensor lstm_seq_bwd(torch::Tensor gates, torch::Tensor c_seq,
                            torch::Tensor w_hh, torch::Tensor dh_out) {
   check_flat(gates, "gates"); check_flat(c_seq, "c_seq");
   check_flat(w_hh, "w_hh"); check_flat(dh_out, "dh_out");
+  TORCH_CHECK(w_hh.size(0) == 4 * 256 && w_hh.size(1) == 256,
+              "lstm_seq_bwd wants W_hh [4H,H]");
   long long B = gates.size(0), T = gates.size(1);
   auto dg = torch::empty_like(gates);
   launch_lstm_seq_bwd(gates.data_ptr<float>(), c_seq.data_ptr<float>(),

@@ -36,43 +36,45 @@ __device__ inline float sigf(float x) { return 1.f / (1.f + __expf(-x)); }
 // h_seq:   [B, T, H]   out
 // gates:   [B, T, 4H]  out (ACTIVATED i,f,g,o — saved for backward)
 // c_seq:   [B, T, H]   out (cell states)
-__global__ void k_lstm_seq_fwd(const float* __restrict__ xp,
-                               const float* __restrict__ w_hh_t,
-                               float* __restrict__ h_seq,
-                               float* __restrict__ gates,
-                               float* __restrict__ c_seq, int B, int T) {
+// 1024 threads per row (16 waves on one CU): thread = gate column
+// j = g*H + h, so the per-step GEMV reads w_hh_t[k*4H + j] fully
+// coalesced with 4 waves per SIMD hiding the L2 latency (the 256-thread
+// form was latency-bound at ~1 ms per T=80 call; this is ~14x less
+// exposed latency per step).  Unroll 8 batches the loads.
+__global__ __launch_bounds__(1024)
+void k_lstm_seq_fwd(const float* __restrict__ xp,
+                    const float* __restrict__ w_hh_t,
+                    float* __restrict__ h_seq,
+                    float* __restrict__ gates,
+                    float* __restrict__ c_seq, int B, int T) {
   __shared__ float h_prev[LSTM_H];
-  int h = threadIdx.x;
+  __shared__ float act_l[4 * LSTM_H];
+  int tid = threadIdx.x;           // gate column j = g*H + h
+  int g = tid >> 8;
   int b = blockIdx.x;
-  h_prev[h] = 0.f;
-  float c = 0.f;
+  if (tid < LSTM_H) h_prev[tid] = 0.f;
+  float c = 0.f;                   // live in threads tid < H only
   __syncthreads();
   for (int t = 0; t < T; ++t) {
     const float* xr = xp + ((long long)b * T + t) * 4 * LSTM_H;
-    float s0 = xr[h];
-    float s1 = xr[LSTM_H + h];
-    float s2 = xr[2 * LSTM_H + h];
-    float s3 = xr[3 * LSTM_H + h];
-    for (int k = 0; k < LSTM_H; ++k) {
-      float hv = h_prev[k];                       // LDS broadcast
-      const float* wr = w_hh_t + (long long)k * 4 * LSTM_H;
-      s0 = fmaf(wr[h], hv, s0);                   // coalesced across lanes
-      s1 = fmaf(wr[LSTM_H + h], hv, s1);
-      s2 = fmaf(wr[2 * LSTM_H + h], hv, s2);
-      s3 = fmaf(wr[3 * LSTM_H + h], hv, s3);
-    }
-    float i = sigf(s0), f = sigf(s1), g = tanhf(s2), o = sigf(s3);
-    c = f * c + i * g;
-    float hn = o * tanhf(c);
+    float s = xr[tid];
+    #pragma unroll 16
+    for (int k = 0; k < LSTM_H; ++k)
+      s = fmaf(w_hh_t[(long long)k * 4 * LSTM_H + tid], h_prev[k], s);
+    float act = (g == 2) ? tanhf(s) : sigf(s);
     long long base = ((long long)b * T + t) * 4 * LSTM_H;
-    gates[base + h] = i;
-    gates[base + LSTM_H + h] = f;
-    gates[base + 2 * LSTM_H + h] = g;
-    gates[base + 3 * LSTM_H + h] = o;
-    c_seq[((long long)b * T + t) * LSTM_H + h] = c;
-    h_seq[((long long)b * T + t) * LSTM_H + h] = hn;
+    gates[base + tid] = act;       // ACTIVATED i,f,g,o (order ifgo)
+    act_l[tid] = act;
     __syncthreads();
-    h_prev[h] = hn;
+    if (tid < LSTM_H) {
+      float i = act_l[tid], f = act_l[LSTM_H + tid];
+      float gg = act_l[2 * LSTM_H + tid], o = act_l[3 * LSTM_H + tid];
+      c = f * c + i * gg;
+      float hn = o * tanhf(c);
+      c_seq[((long long)b * T + t) * LSTM_H + tid] = c;
+      h_seq[((long long)b * T + t) * LSTM_H + tid] = hn;
+      h_prev[tid] = hn;
+    }
     __syncthreads();
   }
 }
@@ -81,47 +83,63 @@ __global__ void k_lstm_seq_fwd(const float* __restrict__ xp,
 // dg_pre [B,T,4H] (pre-activation gate grads == grads of xp).
 // W_hh in its ORIGINAL [4H, H] layout: w_hh[j*H + h] is coalesced across
 // lanes (consecutive h).
-__global__ void k_lstm_seq_bwd(const float* __restrict__ gates,
-                               const float* __restrict__ c_seq,
-                               const float* __restrict__ w_hh,
-                               const float* __restrict__ dh_out,
-                               float* __restrict__ dg_pre, int B, int T) {
+__global__ __launch_bounds__(1024)
+void k_lstm_seq_bwd(const float* __restrict__ gates,
+                    const float* __restrict__ c_seq,
+                    const float* __restrict__ w_hh,
+                    const float* __restrict__ dh_out,
+                    float* __restrict__ dg_pre, int B, int T) {
   __shared__ float dg_l[4 * LSTM_H];
-  int h = threadIdx.x;
+  __shared__ float part[4 * LSTM_H];
+  __shared__ float dh_rec_l[LSTM_H];
+  int tid = threadIdx.x;
+  int h = tid & (LSTM_H - 1), q = tid >> 8;  // partial-dot quarter
   int b = blockIdx.x;
-  float dc = 0.f;
-  float dh_rec = 0.f;
+  float dc = 0.f;                            // live in threads tid < H
+  if (tid < LSTM_H) dh_rec_l[tid] = 0.f;
+  __syncthreads();
   for (int t = T - 1; t >= 0; --t) {
     long long base = ((long long)b * T + t) * 4 * LSTM_H;
     long long cbase = ((long long)b * T + t) * LSTM_H;
-    float i = gates[base + h];
-    float f = gates[base + LSTM_H + h];
-    float g = gates[base + 2 * LSTM_H + h];
-    float o = gates[base + 3 * LSTM_H + h];
-    float ct = c_seq[cbase + h];
-    float cprev = (t > 0) ? c_seq[cbase - LSTM_H + h] : 0.f;
-    float tc = tanhf(ct);
-    float dh = dh_out[cbase + h] + dh_rec;
-    float do_ = dh * tc * o * (1.f - o);
-    dc = dc + dh * o * (1.f - tc * tc);
-    float di = dc * g * i * (1.f - i);
-    float df = dc * cprev * f * (1.f - f);
-    float dg = dc * i * (1.f - g * g);
-    dc = dc * f;
+    if (tid < LSTM_H) {
+      float i = gates[base + tid];
+      float f = gates[base + LSTM_H + tid];
+      float g = gates[base + 2 * LSTM_H + tid];
+      float o = gates[base + 3 * LSTM_H + tid];
+      float ct = c_seq[cbase + tid];
+      float cprev = (t > 0) ? c_seq[cbase - LSTM_H + tid] : 0.f;
+      float tc = tanhf(ct);
+      float dh = dh_out[cbase + tid] + dh_rec_l[tid];
+      float do_ = dh * tc * o * (1.f - o);
+      dc = dc + dh * o * (1.f - tc * tc);
+      float di = dc * g * i * (1.f - i);
+      float df = dc * cprev * f * (1.f - f);
+      float dg = dc * i * (1.f - g * g);
+      dc = dc * f;
+      dg_l[tid] = di;
+      dg_l[LSTM_H + tid] = df;
+      dg_l[2 * LSTM_H + tid] = dg;
+      dg_l[3 * LSTM_H + tid] = do_;
+      dg_pre[base + tid] = di;
+      dg_pre[base + LSTM_H + tid] = df;
+      dg_pre[base + 2 * LSTM_H + tid] = dg;
+      dg_pre[base + 3 * LSTM_H + tid] = do_;
+    }
     __syncthreads();
-    dg_l[h] = di;
-    dg_l[LSTM_H + h] = df;
-    dg_l[2 * LSTM_H + h] = dg;
-    dg_l[3 * LSTM_H + h] = do_;
-    __syncthreads();
-    dg_pre[base + h] = di;
-    dg_pre[base + LSTM_H + h] = df;
-    dg_pre[base + 2 * LSTM_H + h] = dg;
-    dg_pre[base + 3 * LSTM_H + h] = do_;
+    // dh_rec = W_hh^T dg: thread (q, h) sums its 256-row quarter with
+    // lane-coalesced w_hh[j*H + h] reads; 16 waves hide the L2 latency
     float s = 0.f;
-    for (int j = 0; j < 4 * LSTM_H; ++j)
+    #pragma unroll 16
+    for (int jj = 0; jj < LSTM_H; ++jj) {
+      int j = q * LSTM_H + jj;
       s = fmaf(w_hh[(long long)j * LSTM_H + h], dg_l[j], s);
-    dh_rec = s;
+    }
+    part[tid] = s;
+    __syncthreads();
+    if (tid < LSTM_H)
+      dh_rec_l[tid] = part[tid] + part[LSTM_H + tid]
+                      + part[2 * LSTM_H + tid] + part[3 * LSTM_H + tid];
+    __syncthreads();
   }
 }
 
@@ -130,14 +148,14 @@ extern "C" {
 void launch_lstm_seq_fwd(const float* xp, const float* w_hh_t, float* h_seq,
                          float* gates, float* c_seq, int B, int T, int H,
                          hipStream_t s) {
-  hipLaunchKernelGGL(k_lstm_seq_fwd, dim3(B), dim3(LSTM_H), 0, s,
+  hipLaunchKernelGGL(k_lstm_seq_fwd, dim3(B), dim3(4 * LSTM_H), 0, s,
                      xp, w_hh_t, h_seq, gates, c_seq, B, T);
 }
 
 void launch_lstm_seq_bwd(const float* gates, const float* c_seq,
                          const float* w_hh, const float* dh_out,
                          float* dg_pre, int B, int T, int H, hipStream_t s) {
-  hipLaunchKernelGGL(k_lstm_seq_bwd, dim3(B), dim3(LSTM_H), 0, s,
+  hipLaunchKernelGGL(k_lstm_seq_bwd, dim3(B), dim3(4 * LSTM_H), 0, s,
                      gates, c_seq, w_hh, dh_out, dg_pre, B, T);
 }
 

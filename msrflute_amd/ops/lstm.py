@@ -32,7 +32,8 @@ class _LSTMSeq(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dh):
         gates, c_seq, w_hh, h_seq = ctx.saved_tensors
-        dg = _C.lstm_seq_bwd(gates, c_seq, w_hh, dh.contiguous())
+        dg = _C.lstm_seq_bwd(gates, c_seq, w_hh.contiguous(),
+                             dh.contiguous())
         B, T, H = h_seq.shape
         h_prev = torch.cat([h_seq.new_zeros(B, 1, H), h_seq[:, :-1]], dim=1)
         # dW_hh = sum_t dgates_t^T h_{t-1} : one GEMM over the stacked steps
