@@ -64,7 +64,9 @@ def build_config(args):
             "seed": 1234,
         },
         "client_config": {
-            "mixed_precision": args.dtype if args.dtype != "fp32" else "",
+            "mixed_precision": (getattr(args, "dtype", "fp32")
+                                if getattr(args, "dtype", "fp32") != "fp32"
+                                else ""),
             "parallel_clients": int(os.environ.get("BENCH_PAR", "8")),
             "use_fused_cnn": os.environ.get("BENCH_FUSED", "1") == "1",
             "do_profiling": False,
