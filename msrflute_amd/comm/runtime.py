@@ -134,12 +134,68 @@ class FedRuntime:
         return ("gloo", (h1, h2, host if grad.is_cuda else None, grad),
                 wsum, None)
 
+    # -- quantized wire -------------------------------
+    # The reference keeps quantize-dequantize IN PLACE client-side and
+    # ships fp32 (extensions/quantization/quant.py:42-50) — that contract
+    # is preserved.  This OPT-IN transport additionally compresses the
+    # round's all-reduce itself: each rank all_gathers int8 codes + one
+    # fp32 scale per arena segment (per-layer absmax/127) instead of fp32
+    # gradients — ~4x fewer bytes per direction on xGMI — and every rank
+    # dequant-accumulates the N contributions in fixed rank order, so all
+    # replicas compute the identical sum.  config:
+    # client_config.quant_wire: true (server side reads it).
+    def begin_grad_reduce_quant(self, grad: torch.Tensor,
+                                local_weight_sum: float,
+                                seg_expand: torch.Tensor):
+        """Quantized round reduce: all_gather(int8 codes + segment scales).
+        ``seg_expand[i]`` = segment id of element i (precomputed from the
+        arena's layout)."""
+        if not self._active:
+            return ("local", None, None, float(local_weight_sum))
+        n_seg = int(seg_expand.max().item()) + 1 if seg_expand.numel() else 1
+        absmax = torch.zeros(n_seg, dtype=torch.float32, device=grad.device)
+        absmax.scatter_reduce_(0, seg_expand, grad.abs(), reduce="amax")
+        scales = absmax / 127.0
+        safe = torch.where(scales > 0, scales, torch.ones_like(scales))
+        codes = torch.clamp(torch.round(grad / safe[seg_expand]),
+                            -127, 127).to(torch.int8)
+        wsum = torch.tensor([local_weight_sum], dtype=torch.float64,
+                            device=grad.device if self.backend == "nccl"
+                            else "cpu")
+        if self.backend == "gloo":
+            codes_h = codes.cpu() if codes.is_cuda else codes
+            scales_h = scales.cpu() if scales.is_cuda else scales
+            all_codes = [torch.empty_like(codes_h) for _ in range(self.size)]
+            all_scales = [torch.empty_like(scales_h) for _ in range(self.size)]
+            dist.all_gather(all_codes, codes_h)
+            dist.all_gather(all_scales, scales_h)
+            dist.all_reduce(wsum, op=dist.ReduceOp.SUM)
+            return ("quant", (all_codes, all_scales, seg_expand, grad),
+                    wsum, None)
+        all_codes = [torch.empty_like(codes) for _ in range(self.size)]
+        all_scales = [torch.empty_like(scales) for _ in range(self.size)]
+        dist.all_gather(all_codes, codes)
+        dist.all_gather(all_scales, scales)
+        dist.all_reduce(wsum, op=dist.ReduceOp.SUM)
+        return ("quant", (all_codes, all_scales, seg_expand, grad),
+                wsum, None)
+
     def finish_grad_reduce(self, handle) -> float:
         """Join the round reduce; returns the global weight sum.  The
         reduced gradient is in place in the arena afterwards."""
         kind, a, wsum, local = handle
         if kind == "local":
             return local
+        if kind == "quant":
+            all_codes, all_scales, seg_expand, grad = a
+            dev = grad.device
+            acc = torch.zeros_like(grad)
+            for codes_r, scales_r in zip(all_codes, all_scales):
+                c = codes_r.to(dev, torch.float32)
+                s = scales_r.to(dev, torch.float32)
+                acc += c * s[seg_expand]
+            grad.copy_(acc)
+            return float(wsum.item())
         if kind == "nccl":
             torch.cuda.current_stream().wait_event(a)
             return float(wsum.item())  # syncs the scalar only

@@ -110,6 +110,18 @@ class FedAvg(BaseStrategy):
                 or self.runtime is None or self._pending_reduce is not None):
             return
         local_weight_sum = float(sum(self.client_weights))
+        if self.client_config.get("quant_wire", False):
+            # 8-bit codes + per-segment scales on the wire (opt-in;
+            # runtime.begin_grad_reduce_quant)
+            arena = worker_trainer.arena
+            if getattr(self, "_seg_expand", None) is None:
+                lengths = torch.tensor(arena.numels, dtype=torch.int64)
+                self._seg_expand = torch.repeat_interleave(
+                    torch.arange(len(arena.numels)), lengths
+                ).to(arena.device)
+            self._pending_reduce = self.runtime.begin_grad_reduce_quant(
+                arena.grad, local_weight_sum, self._seg_expand)
+            return
         self._pending_reduce = self.runtime.begin_grad_reduce(
             worker_trainer.arena.grad, local_weight_sum)
 

@@ -100,3 +100,24 @@ bench.main()
                        cwd=os.path.dirname(os.path.dirname(__file__)))
     assert r.returncode == 0, r.stderr[-3000:]
     assert '"metric"' in r.stdout
+
+
+def test_quant_wire_device_path(rt):
+    """Device-side quantized transport (int8 codes + segment scales) on a
+    1-rank RCCL group: dequant(quant(g)) within per-segment bound."""
+    torch.manual_seed(3)
+    seg_expand = torch.cat([torch.zeros(100_000, dtype=torch.int64),
+                            torch.ones(50_000, dtype=torch.int64)]).cuda()
+    g = torch.cat([torch.randn(100_000, device="cuda") * 2.0,
+                   torch.randn(50_000, device="cuda") * 0.05])
+    ref = g.clone()
+    h = rt.begin_grad_reduce_quant(g, 4.0, seg_expand)
+    wsum = rt.finish_grad_reduce(h)
+    torch.cuda.synchronize()
+    assert wsum == 4.0
+    s0 = ref[:100_000].abs().max() / 127.0
+    s1 = ref[100_000:].abs().max() / 127.0
+    err0 = (g[:100_000] - ref[:100_000]).abs().max()
+    err1 = (g[100_000:] - ref[100_000:]).abs().max()
+    assert float(err0) <= float(s0) / 2 + 1e-7
+    assert float(err1) <= float(s1) / 2 + 1e-7
