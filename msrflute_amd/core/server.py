@@ -135,8 +135,20 @@ class OptimizationServer:
         # order-free strategy, a pool of stream-parallel executors trains
         # several clients concurrently (ClientPool)
         n_par = int(config["client_config"].get("parallel_clients", 8))
+        # DGA in fast-aggregation mode is order-free like FedAvg (weighted
+        # sum; softmax weights are per-client local), so it runs on the
+        # stream pool too — the BASELINE config-5 shape (DGA + local DP +
+        # 8-bit quant) trains with the fused epoch + pooled streams.
+        # Stacked modes (staleness simulation, RL reweighting) need the
+        # per-client stack and keep the single executor.
+        dga_poolable = (config["strategy"] == "DGA"
+                        and server_config.get("fast_aggregation", True)
+                        and not server_config.get("wantRL", False)
+                        and float(server_config.get("stale_prob", 0.0) or 0.0)
+                        == 0.0)
         if (torch.cuda.is_available() and n_par > 1
-                and config["strategy"] in ("FedAvg", "FedProx")
+                and (config["strategy"] in ("FedAvg", "FedProx")
+                     or dga_poolable)
                 and server_config.get("type") != "personalization"
                 and server_config.get("fast_aggregation", True)
                 and not config.get("dump_norm_stats", False)

@@ -123,6 +123,10 @@ def update_privacy_accountant(config, num_clients, curr_iter,
     if dp_config is None or not (dp_config.get("enable_global_dp", False)
                                  or dp_config.get("enable_local_dp", False)):
         return None
+    if dp_config.get("eps", 0) < 0 and dp_config.get("global_sigma") is None:
+        # eps < 0 = clip-only local DP (reference privacy/__init__.py:
+        # 167-171): no noise is added, so there is nothing to account
+        return None
     from . import analysis as privacy_analysis
 
     K = 1
