@@ -21,21 +21,83 @@ from ..core.dataloader import BaseDataLoader
 from ..core.dataset import BaseDataset
 
 
+def _normalize_blob(blob):
+    """Normalize LEAF/FedML variants of the blob convention to FLUTE's
+    ``{users, num_samples, user_data, user_data_label}``:
+
+    * LEAF/FedML JSON stores ``user_data[user] = {"x": ..., "y": ...}``
+      (reference experiments/cv_lr_mnist/dataloaders/preprocessing.py:42-68
+      splits x/y the same way);
+    * ``num_samples`` may be missing (derived from the x lists).
+    """
+    ud = blob.get("user_data", {})
+    first = next(iter(ud.values()), None)
+    if isinstance(first, dict) and "y" in first and "x" in first \
+            and "user_data_label" not in blob:
+        blob = dict(blob)
+        blob["user_data"] = {u: v["x"] for u, v in ud.items()}
+        blob["user_data_label"] = {u: v["y"] for u, v in ud.items()}
+        ud = blob["user_data"]
+    if "num_samples" not in blob and "users" in blob:
+        blob = dict(blob)
+        blob["num_samples"] = [len(ud[u]) for u in blob["users"]]
+    return blob
+
+
+def load_hdf5_blob(path):
+    """Read the reference's HDF5 federated layout (groups ``users``,
+    ``num_samples``, ``user_data/<user>``[, ``user_data_label/<user>``] —
+    reference testing/create_data.py:95-140, experiments/*/preprocess.py).
+    Requires h5py at runtime; this offline image ships without it, so the
+    call fails with an actionable message instead of an ImportError."""
+    try:
+        import h5py
+    except ImportError as e:
+        raise RuntimeError(
+            "reading .hdf5 federated datasets requires h5py (pip install "
+            "h5py on a networked machine); offline runs can convert with "
+            "tools/create_data.py to .pt/.npz/.json instead") from e
+    with h5py.File(path, "r") as f:
+        users = [u.decode() if isinstance(u, bytes) else str(u)
+                 for u in f["users"][()]]
+        blob = {"users": users,
+                "num_samples": list(f["num_samples"][()]),
+                "user_data": {}}
+        for u in users:
+            g = f["user_data"][u]
+            # either a dataset per user or an {x, y} group
+            if hasattr(g, "keys") and "x" in g:
+                blob["user_data"][u] = np.asarray(g["x"])
+                if "y" in g:
+                    blob.setdefault("user_data_label", {})[u] = \
+                        np.asarray(g["y"])
+            else:
+                blob["user_data"][u] = np.asarray(g)
+        if "user_data_label" in f:
+            blob["user_data_label"] = {
+                u: np.asarray(f["user_data_label"][u]) for u in users}
+    return _normalize_blob(blob)
+
+
 def load_blob(data):
-    """Load a data blob from dict / .json / .npz / torch .pt file."""
+    """Load a data blob from dict / .json / .npz / torch .pt / .hdf5 file
+    (LEAF/FedML per-user {x, y} layouts are normalized — _normalize_blob)."""
     if isinstance(data, dict):
-        return data
+        return _normalize_blob(data)
     if data is None:
         raise ValueError("no data provided")
     path = str(data)
     if path.endswith(".npz"):
         z = np.load(path, allow_pickle=True)
-        return {k: z[k].item() if z[k].dtype == object and z[k].shape == () else z[k]
-                for k in z.files}
+        return _normalize_blob(
+            {k: z[k].item() if z[k].dtype == object and z[k].shape == ()
+             else z[k] for k in z.files})
     if path.endswith(".pt"):
-        return torch.load(path, weights_only=False)
+        return _normalize_blob(torch.load(path, weights_only=False))
+    if path.endswith(".hdf5") or path.endswith(".h5"):
+        return load_hdf5_blob(path)
     with open(path, "r") as f:
-        return json.load(f)
+        return _normalize_blob(json.load(f))
 
 
 class ArrayDataset(BaseDataset):
