@@ -84,6 +84,9 @@ class OptimizationServer:
             model_type=self.model_type, decoder_config=decoder_config,
             arena=arena, val_fn=self.evaluation.make_val_fn())
         self.metrics["worker_trainer"] = self.worker_trainer
+        # the RL path may force a val pass (run_lr_scheduler -> val_fn)
+        # before any Evaluation.run has populated it
+        self.evaluation.worker_trainer = self.worker_trainer
 
         # server-side replay trainer (reference: server.py:130-151)
         self.server_replay_iterations = None

@@ -66,6 +66,7 @@ class RL:
         self.gamma = rl_config.get("gamma", 0.99)
         self.epsilon = rl_config.get("epsilon", 0.5)
         self.epsilon_decay = rl_config.get("epsilon_decay", 0.99)
+        self.final_epsilon = rl_config.get("final_epsilon", 0.01)
         self.hidden_dim = rl_config.get("hidden_dim", 512)
         self.lr = rl_config.get("lr", 0.001)
         self.batch_size = rl_config.get("batch_size", 16)
@@ -99,11 +100,13 @@ class RL:
         out_dim = self.out_dim or (len(state) // 4)
         self._ensure_model(len(state), out_dim)
         if self._rng.random() < self.epsilon:
-            self.epsilon *= self.epsilon_decay
+            self.epsilon = max(self.epsilon * self.epsilon_decay,
+                               self.final_epsilon)
             action = np.array([self._rng.uniform(-1, 1) for _ in range(out_dim)],
                               dtype=np.float32)
             return action
-        self.epsilon *= self.epsilon_decay
+        self.epsilon = max(self.epsilon * self.epsilon_decay,
+                           self.final_epsilon)
         with torch.no_grad():
             t = to_device(torch.from_numpy(state)).unsqueeze(0)
             q = self.model(t).squeeze(0).detach().cpu().numpy()
@@ -116,32 +119,64 @@ class RL:
         self.rl_losses = losses
 
     def train(self, batch):
-        """One replay-memory training step (reference: RL.py:206-262):
-        push (state, action, reward), then fit Q(state) toward
-        action*reward on a sampled minibatch."""
+        """One replay step (reference: RL.py:206-262): push the round's
+        (state, action, reward) transition, then fit the DQN on a sampled
+        minibatch with ``q = Q(s)·a`` toward the γ-discounted target
+        ``y = r + γ·max Q(s')``.
+
+        The reference computes exactly ``q = Σ Q(s)·a`` and ``y = r`` —
+        its own comment ("set y_j to r_j for terminal state, otherwise to
+        r_j + gamma*max(Q)") documents the discounted form but every FL
+        round was treated as terminal.  Rounds are NOT terminal (the next
+        round's state is the successor), so this implements the
+        documented intent: each transition's next state is backfilled
+        when the following round arrives, and transitions still awaiting
+        a successor use y = r.  γ=0 reproduces the reference's update
+        exactly."""
         state, action, reward = batch
         state = np.asarray(state, dtype=np.float32)
         action = np.asarray(action, dtype=np.float32)
-        self.memory.append((state, action, float(reward[0])))
+        r = float(np.asarray(reward).reshape(-1)[0])
+        # backfill the previous transition's successor state
+        if self.memory:
+            s_prev, a_prev, r_prev, _ = self.memory[-1]
+            self.memory[-1] = (s_prev, a_prev, r_prev, state)
+        self.memory.append((state, action, r, None))
         self._ensure_model(len(state), len(action))
 
         n = min(self.batch_size, len(self.memory))
         sample = self._rng.sample(list(self.memory), n)
-        # pad/truncate stored episodes to the current dims (cohort size may vary)
+        # pad/truncate stored episodes to the current dims (cohort size
+        # varies round to round)
         S = np.zeros((n, self._state_dim), dtype=np.float32)
-        T_ = np.zeros((n, self._out_dim), dtype=np.float32)
-        for i, (s, a, r) in enumerate(sample):
+        A = np.zeros((n, self._out_dim), dtype=np.float32)
+        R = np.zeros(n, dtype=np.float32)
+        SN = np.zeros((n, self._state_dim), dtype=np.float32)
+        has_next = np.zeros(n, dtype=np.float32)
+        for i, (s, a, rr, sn) in enumerate(sample):
             S[i, :min(len(s), self._state_dim)] = s[:self._state_dim]
-            t = a * (1.0 if r >= 0 else -1.0) * max(abs(r), 0.1)
-            T_[i, :min(len(t), self._out_dim)] = t[:self._out_dim]
+            A[i, :min(len(a), self._out_dim)] = a[:self._out_dim]
+            R[i] = rr
+            if sn is not None:
+                SN[i, :min(len(sn), self._state_dim)] = sn[:self._state_dim]
+                has_next[i] = 1.0
         S_t = to_device(torch.from_numpy(S))
-        T_t = to_device(torch.from_numpy(T_))
+        A_t = to_device(torch.from_numpy(A))
+        R_t = to_device(torch.from_numpy(R))
         self.model.train()
+        with torch.no_grad():
+            next_max = self.model(to_device(torch.from_numpy(SN))).max(dim=1).values
+        y = (R_t + self.gamma * to_device(torch.from_numpy(has_next))
+             * next_max).detach()
+        q = (self.model(S_t) * A_t).sum(dim=1)
         self.optimizer.zero_grad()
-        loss = torch.nn.functional.mse_loss(self.model(S_t), T_t)
+        loss = torch.nn.functional.mse_loss(q, y)
         loss.backward()
         self.optimizer.step()
-        self.running_loss = float(loss.item())
+        # running EMA like the reference (RL.py:258-262)
+        cur = float(loss.item())
+        self.running_loss = cur if self.running_loss == 0.0 \
+            else 0.95 * self.running_loss + 0.05 * cur
         self.runningLoss = self.running_loss
         return self.running_loss
 
