@@ -118,6 +118,48 @@ class BERT(BaseModel):
                 "perplexity": {"value": float(torch.exp(loss).item()),
                                "higher_is_better": False}}
 
+    def prediction_loop(self, dataloader, runtime=None, max_tokens=2 ** 22):
+        """Distributed prediction loop (reference: model.py:300-380, which
+        vendors HF's DistributedTensorGatherer).  Each rank runs its
+        dataloader, collects masked-position predictions and label ids,
+        and the per-rank arrays are gathered across ranks through the
+        runtime's padded row-gather (comm/runtime.all_gather_rows — the
+        engine's replacement for the vendored gatherer).  Returns
+        {"predictions", "label_ids", "metrics"} like the reference's
+        PredictionOutput; ``max_tokens`` bounds host memory.
+        """
+        import torch as _t
+        self.set_eval()
+        preds, labels, loss_sum, n_batches = [], [], 0.0, 0
+        with _t.no_grad():
+            for batch in dataloader.create_loader() \
+                    if hasattr(dataloader, "create_loader") else dataloader:
+                inputs = self._prepare(batch)
+                lab = inputs.pop("labels")
+                out = self.model(**inputs)
+                loss_sum += float(self._loss_from_logits(out.logits, lab))
+                n_batches += 1
+                mask = lab != -100
+                preds.append(out.logits.argmax(dim=-1)[mask].cpu())
+                labels.append(lab[mask].cpu())
+                if sum(p.numel() for p in preds) > max_tokens:
+                    break
+        p = (_t.cat(preds) if preds else _t.zeros(0, dtype=_t.long))
+        l = (_t.cat(labels) if labels else _t.zeros(0, dtype=_t.long))
+        rows = _t.stack([p.double(), l.double()], dim=1)  # [n, 2]
+        if runtime is not None and runtime.size > 1:
+            counts = runtime.all_gather_object(int(rows.shape[0]))
+            gathered = runtime.all_gather_rows(rows, counts)
+            rows = _t.cat(gathered, dim=0)
+        p_all, l_all = rows[:, 0].long(), rows[:, 1].long()
+        n = max(int(p_all.numel()), 1)
+        acc = float((p_all == l_all).sum()) / n
+        mean_loss = loss_sum / max(n_batches, 1)
+        metrics = {"eval_loss": mean_loss,
+                   "perplexity": float(torch.exp(torch.tensor(mean_loss))),
+                   "acc": acc}
+        return {"predictions": p_all, "label_ids": l_all, "metrics": metrics}
+
     def set_eval(self):
         self.eval()
         self.model.eval()

@@ -48,10 +48,17 @@ server.run_stats = {k: [] for k in [
 for i in range(3):
     server.run_one_round(i, housekeeping=False)
 torch.cuda.synchronize()
+import cProfile, pstats, io
+pr = cProfile.Profile()
 t0 = time.time(); N = 10
+pr.enable()
 for i in range(3, 3 + N):
     server.run_one_round(i, housekeeping=False)
+pr.disable()
 torch.cuda.synchronize()
+buf = io.StringIO()
+pstats.Stats(pr, stream=buf).sort_stats("cumulative").print_stats(22)
+print(buf.getvalue()[:3200])
 ms = (time.time() - t0) / N * 1000
 print(f"PAR={par} ms/round={ms:.1f}")
 acc = dict(server.executor.perf_acc)
