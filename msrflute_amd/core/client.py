@@ -706,6 +706,25 @@ def _pool_run_fused_round_batch(self, client_ids, initial_lr, iteration,
         n = store.offsets[i + 1] - store.offsets[i]
         if n == 0 or (dms is not None and n > dms):
             return None
+
+    # cross-client MEGA round: one launch set per batch-step covering all
+    # K clients (fp32 path; bf16 keeps the per-executor fused rounds)
+    fc = prim.fused_cnn
+    if (not fc.use_bf16
+            and prim.client_config.get("use_mega_round", True)):
+        if getattr(self, "_mega", None) is None:
+            from ..ops.fused_cnn import MegaRound
+            self._mega = MegaRound(prim.arena, fc.C, fc.bs, fc.p1, fc.p2,
+                                   prim.client_config["data_config"]["train"]
+                                   .get("max_grad_norm"))
+        if self._mega.supports(len(client_ids)):
+            out = self._mega.run(store, ds, client_ids, seeds, initial_lr,
+                                 self.server_arena, self.round_accums[0])
+            if out is not None:
+                self._streams_dirty = True
+                prim.perf_acc["clients"] = (prim.perf_acc.get("clients", 0)
+                                            + len(client_ids))
+                return out
     for ex in self.executors[1:]:
         if not ex._shard_store_tried:
             ex._shard_store = store

@@ -99,6 +99,23 @@ extern "C" void launch_conv2_bwd_w_mfma_bf16(const float*, const float*, int,
                                              float*, float*, float*,
                                              hipStream_t);
 
+extern "C" void launch_cnn_round_mega(
+    const float* shard_x, const long long* shard_y,
+    const long long* orders_dev,
+    const long long* row_bases_dev, const long long* order_offs_dev,
+    const long long* counts_dev, const long long* counts_host,
+    const float* weights_dev, const long long* seeds_dev,
+    int K, int bs, int C,
+    const float* server_params, float* params_stack, float* grads_stack,
+    float* round_accum,
+    float* xb, float* a1, float* r2, float* a2, float* z3, float* a3,
+    float* dlogits, float* dz3, float* da2, float* dz2, float* dz1,
+    float* w2t_stack, float* w2rot_stack, float* slab,
+    int* yb, unsigned char* pidx, unsigned char* m2, unsigned char* m3,
+    double* acc2k,
+    const float* lr_t, float max_norm, float p1, float p2,
+    float* stats_out, float* loss_out, hipStream_t s);
+
 extern "C" void launch_cnn_round(
     const float* shard_x, const long long* shard_y, const long long* orders,
     const long long* row_bases, const long long* order_offs,
@@ -663,7 +680,107 @@ torch::Tensor dbg_mfma_bf16_probe(torch::Tensor A, torch::Tensor B) {
   return D;
 }
 
+// MEGA round: every sampled client of the round in ONE launch set per
+// batch-step (fused_cnn_mega.hip).  Caller provides the per-client
+// metadata both on device (kernel use) and host (max_batches), the
+// K*P parameter/gradient stacks and a float workspace sliced here.
+void cnn_round_mega(torch::Tensor shard_x, torch::Tensor shard_y,
+                    torch::Tensor orders_dev, torch::Tensor row_bases_dev,
+                    torch::Tensor order_offs_dev, torch::Tensor counts_dev,
+                    torch::Tensor counts_host, torch::Tensor weights_dev,
+                    torch::Tensor seeds_dev, int64_t bs, int64_t C,
+                    torch::Tensor server_params, torch::Tensor params_stack,
+                    torch::Tensor grads_stack, torch::Tensor round_accum,
+                    torch::Tensor work_f, torch::Tensor work_i,
+                    torch::Tensor work_b, torch::Tensor work_d,
+                    torch::Tensor lr_t, double max_norm, double p1,
+                    double p2, torch::Tensor stats_out,
+                    torch::Tensor loss_out) {
+  check_flat(server_params, "server_params");
+  check_flat(params_stack, "params_stack");
+  check_flat(grads_stack, "grads_stack");
+  check_flat(round_accum, "round_accum");
+  check_flat(work_f, "work_f");
+  TORCH_CHECK(orders_dev.is_cuda() && row_bases_dev.is_cuda() &&
+              order_offs_dev.is_cuda() && counts_dev.is_cuda() &&
+              weights_dev.is_cuda() && seeds_dev.is_cuda(),
+              "mega metadata must be device tensors");
+  TORCH_CHECK(!counts_host.is_cuda() &&
+              counts_host.scalar_type() == torch::kInt64);
+  int K = (int)counts_host.numel();
+  TORCH_CHECK(row_bases_dev.numel() == K && order_offs_dev.numel() == K &&
+              counts_dev.numel() == K && weights_dev.numel() == K &&
+              seeds_dev.numel() == K, "metadata lengths must equal K");
+  TORCH_CHECK(bs >= 1 && bs <= 32);
+  long long P = server_params.numel();
+  TORCH_CHECK(params_stack.numel() >= (long long)K * P &&
+              grads_stack.numel() >= (long long)K * P,
+              "parameter stacks too small");
+  TORCH_CHECK(stats_out.numel() >= 2 * K && loss_out.numel() >= K);
+  {
+    long long n_rows = shard_y.numel();
+    long long n_orders = orders_dev.numel();
+    auto rb = row_bases_dev.cpu();
+    auto oo = order_offs_dev.cpu();
+    auto rba = rb.accessor<int64_t, 1>();
+    auto ooa = oo.accessor<int64_t, 1>();
+    auto ct = counts_host.accessor<int64_t, 1>();
+    for (int k = 0; k < K; ++k) {
+      TORCH_CHECK(ct[k] > 0 && rba[k] >= 0 && ooa[k] >= 0 &&
+                  rba[k] + ct[k] <= n_rows &&
+                  ooa[k] + ct[k] <= n_orders,
+                  "client ", k, ": metadata out of bounds");
+    }
+  }
+  int G = K * (int)bs;
+  float* f = work_f.data_ptr<float>();
+  auto take = [&](long long n) { float* p = f; f += n; return p; };
+  float* xb = take((long long)G * 784);
+  float* a1 = take((long long)G * 21632);
+  float* r2 = take((long long)G * 36864);
+  float* a2 = take((long long)G * 9216);
+  float* z3 = take((long long)G * 128);
+  float* a3 = take((long long)G * 128);
+  float* dlg = take((long long)G * C);
+  float* dz3 = take((long long)G * 128);
+  float* da2 = take((long long)G * 9216);
+  float* dz2 = take((long long)G * 36864);
+  float* dz1 = take((long long)G * 21632);
+  float* w2t = take((long long)K * 18432);
+  float* w2rot = take((long long)K * 18432);
+  float* slab = take((long long)G * 18432);
+  TORCH_CHECK(f - work_f.data_ptr<float>() <= work_f.numel(),
+              "mega float workspace too small");
+  TORCH_CHECK(work_i.numel() >= G, "mega int workspace too small");
+  unsigned char* u = work_b.data_ptr<unsigned char>();
+  unsigned char* pidx = u;
+  unsigned char* m2 = u + (long long)G * 9216;
+  unsigned char* m3 = u + (long long)G * 9216 * 2;
+  TORCH_CHECK(work_b.numel() >= (long long)G * (9216 * 2 + 128),
+              "mega byte workspace too small");
+  TORCH_CHECK(work_d.numel() >= 2 * K, "mega double workspace too small");
+  launch_cnn_round_mega(
+      shard_x.data_ptr<float>(),
+      reinterpret_cast<const long long*>(shard_y.data_ptr<int64_t>()),
+      reinterpret_cast<const long long*>(orders_dev.data_ptr<int64_t>()),
+      reinterpret_cast<const long long*>(row_bases_dev.data_ptr<int64_t>()),
+      reinterpret_cast<const long long*>(order_offs_dev.data_ptr<int64_t>()),
+      reinterpret_cast<const long long*>(counts_dev.data_ptr<int64_t>()),
+      reinterpret_cast<const long long*>(counts_host.data_ptr<int64_t>()),
+      weights_dev.data_ptr<float>(),
+      reinterpret_cast<const long long*>(seeds_dev.data_ptr<int64_t>()),
+      K, (int)bs, (int)C, server_params.data_ptr<float>(),
+      params_stack.data_ptr<float>(), grads_stack.data_ptr<float>(),
+      round_accum.data_ptr<float>(),
+      xb, a1, r2, a2, z3, a3, dlg, dz3, da2, dz2, dz1, w2t, w2rot, slab,
+      work_i.data_ptr<int>(), pidx, m2, m3, work_d.data_ptr<double>(),
+      lr_t.data_ptr<float>(), (float)max_norm, (float)p1, (float)p2,
+      stats_out.data_ptr<float>(), loss_out.data_ptr<float>(),
+      cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("cnn_round_mega", &cnn_round_mega);
   m.def("dbg_mfma_bf16_probe", &dbg_mfma_bf16_probe);
   m.def("dbg_conv2_fwd_mfma_bf16", &dbg_conv2_fwd_mfma_bf16);
   m.def("dbg_conv2_bwd_x_mfma_bf16", &dbg_conv2_bwd_x_mfma_bf16);

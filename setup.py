@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "msrflute_amd/csrc/bindings.cpp",
         "msrflute_amd/csrc/flat_ops.hip",
         "msrflute_amd/csrc/fused_cnn.hip",
+        "msrflute_amd/csrc/fused_cnn_mega.hip",
         "msrflute_amd/csrc/lstm_seq.hip",
         "msrflute_amd/csrc/gru_seq.hip",
     ],
