@@ -710,13 +710,13 @@ def _pool_run_fused_round_batch(self, client_ids, initial_lr, iteration,
     # cross-client MEGA round: one launch set per batch-step covering all
     # K clients (fp32 path; bf16 keeps the per-executor fused rounds)
     fc = prim.fused_cnn
-    if (not fc.use_bf16
-            and prim.client_config.get("use_mega_round", True)):
+    if prim.client_config.get("use_mega_round", True):
         if getattr(self, "_mega", None) is None:
             from ..ops.fused_cnn import MegaRound
             self._mega = MegaRound(prim.arena, fc.C, fc.bs, fc.p1, fc.p2,
                                    prim.client_config["data_config"]["train"]
-                                   .get("max_grad_norm"))
+                                   .get("max_grad_norm"),
+                                   use_bf16=fc.use_bf16)
         if self._mega.supports(len(client_ids)):
             out = self._mega.run(store, ds, client_ids, seeds, initial_lr,
                                  self.server_arena, self.round_accums[0])

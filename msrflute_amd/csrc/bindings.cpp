@@ -114,7 +114,7 @@ extern "C" void launch_cnn_round_mega(
     int* yb, unsigned char* pidx, unsigned char* m2, unsigned char* m3,
     double* acc2k,
     const float* lr_t, float max_norm, float p1, float p2,
-    float* stats_out, float* loss_out, hipStream_t s);
+    float* stats_out, float* loss_out, hipStream_t s, int use_bf16);
 
 extern "C" void launch_cnn_round(
     const float* shard_x, const long long* shard_y, const long long* orders,
@@ -695,7 +695,7 @@ void cnn_round_mega(torch::Tensor shard_x, torch::Tensor shard_y,
                     torch::Tensor work_b, torch::Tensor work_d,
                     torch::Tensor lr_t, double max_norm, double p1,
                     double p2, torch::Tensor stats_out,
-                    torch::Tensor loss_out) {
+                    torch::Tensor loss_out, bool use_bf16) {
   check_flat(server_params, "server_params");
   check_flat(params_stack, "params_stack");
   check_flat(grads_stack, "grads_stack");
@@ -776,7 +776,7 @@ void cnn_round_mega(torch::Tensor shard_x, torch::Tensor shard_y,
       work_i.data_ptr<int>(), pidx, m2, m3, work_d.data_ptr<double>(),
       lr_t.data_ptr<float>(), (float)max_norm, (float)p1, (float)p2,
       stats_out.data_ptr<float>(), loss_out.data_ptr<float>(),
-      cur_stream());
+      cur_stream(), use_bf16 ? 1 : 0);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

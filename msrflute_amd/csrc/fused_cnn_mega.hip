@@ -144,6 +144,7 @@ void k_conv2_fwd_mfma_mb(const float* __restrict__ a1,
   int kc = lane >> 4;
   f32x4 acc[4] = {f32x4{0,0,0,0}, f32x4{0,0,0,0},
                   f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  #pragma unroll 4
   for (int k0 = 0; k0 < 288; k0 += 4) {
     int kk = k0 + kc;
     int ci = kk / 9, rem = kk % 9, kh = rem / 3, kw = rem % 3;
@@ -527,6 +528,7 @@ void k_conv2_bwd_w_mfma_mb(const float* __restrict__ dz2t,
   int kc = lane >> 4;
   const float* dzb = dz2t + g * 36864;
   f32x4 acc[3] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  #pragma unroll 4
   for (int k0 = 0; k0 < 576; k0 += 4) {
     int o = k0 + kc;
     int yy = o / 24, xx = o % 24;
@@ -603,6 +605,7 @@ void k_conv2_bwd_x_mfma_mb(const float* __restrict__ dz2,
   bool mrow = m < 676;
   int kc = lane >> 4;
   f32x4 acc[2] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  #pragma unroll 4
   for (int k0 = 0; k0 < 576; k0 += 4) {
     int kk = k0 + kc;
     int co = kk / 9, rem = kk % 9, kh = rem / 3, kw = rem % 3;
@@ -775,6 +778,247 @@ __global__ void k_accum_mb(float* __restrict__ round_accum,
 }
 
 // ---------------------------------------------------------------------------
+// bf16 mega variants (mixed precision, fp32 master — same recipe as the
+// per-client bf16 kernels in fused_cnn.hip: operands materialized as
+// k-contiguous bf16 LDS tiles with padded strides, f32 accumulators,
+// MFMA v_mfma_f32_16x16x32_bf16; the inner loops read ONLY LDS, so the
+// L2-latency bound of the f32 variants does not apply).
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8m;
+
+__device__ __forceinline__ bf16x8m ld_bf16x8m(const __bf16* p) {
+  return *reinterpret_cast<const bf16x8m*>(p);
+}
+
+#define MC2_LD 296
+__global__ __launch_bounds__(256)
+void k_conv2_fwd_mfma_mb_bf16(const float* __restrict__ a1,
+                              const float* __restrict__ params, long long P,
+                              long long ow2, long long ob2, int bs, int K,
+                              float* __restrict__ r2) {
+  __shared__ __bf16 imc[64 * MC2_LD];
+  __shared__ __bf16 wb[64 * MC2_LD];
+  long long g = blockIdx.x / 9;
+  int mt = blockIdx.x % 9;
+  int k = (int)(g / bs);
+  const float* a1b = a1 + g * 21632;
+  const float* w2 = params + (long long)k * P + ow2;
+  const float* b2 = params + (long long)k * P + ob2;
+  for (int i = threadIdx.x; i < 64 * 288; i += 256) {
+    int mr = i / 288, kk = i % 288;
+    int m = mt * 64 + mr, yy = m / 24, xx = m % 24;
+    int ci = kk / 9, rem = kk % 9, kh = rem / 3, kw = rem % 3;
+    imc[mr * MC2_LD + kk] = (__bf16)a1b[ci * 676 + (yy + kh) * 26 + xx + kw];
+    wb[mr * MC2_LD + kk] = (__bf16)w2[i];  // mr doubles as co
+  }
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int il = lane & 15, kc8 = (lane >> 4) * 8;
+  f32x4 acc[4] = {f32x4{0,0,0,0}, f32x4{0,0,0,0},
+                  f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  for (int k0 = 0; k0 < 288; k0 += 32) {
+    bf16x8m a = ld_bf16x8m(&imc[(w * 16 + il) * MC2_LD + k0 + kc8]);
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      bf16x8m bv = ld_bf16x8m(&wb[(nt * 16 + il) * MC2_LD + k0 + kc8]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc[nt], 0, 0, 0);
+    }
+  }
+  int om = mt * 64 + w * 16 + (lane >> 4) * 4;
+  #pragma unroll
+  for (int r = 0; r < 4; ++r)
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      int co = nt * 16 + il;
+      float v = acc[nt][r] + b2[co];
+      r2[(g * 64 + co) * 576 + om + r] = v > 0.f ? v : 0.f;
+    }
+}
+
+#define MC2B_LD 584
+__global__ __launch_bounds__(256)
+void k_conv2_bwd_x_mfma_mb_bf16(const float* __restrict__ dz2,
+                                const float* __restrict__ params,
+                                long long P, long long ow2, int bs, int K,
+                                const float* __restrict__ a1,
+                                float* __restrict__ dz1) {
+  __shared__ __bf16 imc[64 * MC2B_LD];
+  __shared__ __bf16 wt[32 * MC2B_LD];
+  long long g = blockIdx.x / 11;
+  int mt = blockIdx.x % 11;
+  int k = (int)(g / bs);
+  const float* dzb = dz2 + g * 36864;
+  const float* w2 = params + (long long)k * P + ow2;
+  for (int i = threadIdx.x; i < 64 * 576; i += 256) {
+    int mr = i / 576, kk = i % 576;
+    int m = mt * 64 + mr, p = m / 26, q = m % 26;
+    int co = kk / 9, rem = kk % 9, kh = rem / 3, kw = rem % 3;
+    int y = p - kh, x = q - kw;
+    imc[mr * MC2B_LD + kk] =
+        (__bf16)((m < 676 && y >= 0 && y < 24 && x >= 0 && x < 24)
+                     ? dzb[co * 576 + y * 24 + x] : 0.f);
+  }
+  for (int i = threadIdx.x; i < 32 * 576; i += 256) {
+    int ci = i / 576, kk = i % 576;
+    int co = kk / 9, rem = kk % 9;
+    wt[ci * MC2B_LD + kk] = (__bf16)w2[(long long)co * 288 + ci * 9 + rem];
+  }
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int il = lane & 15, kc8 = (lane >> 4) * 8;
+  f32x4 acc[2] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  for (int k0 = 0; k0 < 576; k0 += 32) {
+    bf16x8m a = ld_bf16x8m(&imc[(w * 16 + il) * MC2B_LD + k0 + kc8]);
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      bf16x8m bv = ld_bf16x8m(&wt[(nt * 16 + il) * MC2B_LD + k0 + kc8]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc[nt], 0, 0, 0);
+    }
+  }
+  int om = mt * 64 + w * 16 + (lane >> 4) * 4;
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    if (om + r >= 676) continue;
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      int ci = nt * 16 + il;
+      long long o = (g * 32 + ci) * 676 + om + r;
+      dz1[o] = a1[o] > 0.f ? acc[nt][r] : 0.f;
+    }
+  }
+}
+
+__global__ __launch_bounds__(256)
+void k_conv2_bwd_w_mfma_mb_bf16(const float* __restrict__ dz2,
+                                const float* __restrict__ a1, int bs, int K,
+                                float* __restrict__ slab) {
+  __shared__ __bf16 dzb16[64 * MC2B_LD];
+  __shared__ __bf16 imt[48 * MC2B_LD];
+  int nb = blockIdx.x % 6;
+  long long g = blockIdx.x / 6;
+  const float* dzb = dz2 + g * 36864;
+  const float* a1b = a1 + g * 21632;
+  for (int i = threadIdx.x; i < 64 * 576; i += 256) {
+    int co = i / 576, o = i % 576;
+    dzb16[co * MC2B_LD + o] = (__bf16)dzb[i];
+  }
+  for (int i = threadIdx.x; i < 48 * 576; i += 256) {
+    int nr = i / 576, o = i % 576;
+    int n = nb * 48 + nr;
+    int ci = n / 9, rem = n % 9, kh = rem / 3, kw = rem % 3;
+    int yy = o / 24, xx = o % 24;
+    imt[nr * MC2B_LD + o] = (__bf16)a1b[ci * 676 + (yy + kh) * 26 + xx + kw];
+  }
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int il = lane & 15, kc8 = (lane >> 4) * 8;
+  f32x4 acc[3] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  for (int k0 = 0; k0 < 576; k0 += 32) {
+    bf16x8m a = ld_bf16x8m(&dzb16[(w * 16 + il) * MC2B_LD + k0 + kc8]);
+    #pragma unroll
+    for (int nt = 0; nt < 3; ++nt) {
+      bf16x8m bv = ld_bf16x8m(&imt[(nt * 16 + il) * MC2B_LD + k0 + kc8]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc[nt], 0, 0, 0);
+    }
+  }
+  float* out = slab + g * 18432;
+  int orow = w * 16 + (lane >> 4) * 4;
+  #pragma unroll
+  for (int r = 0; r < 4; ++r)
+    #pragma unroll
+    for (int nt = 0; nt < 3; ++nt)
+      out[(orow + r) * 288 + nb * 48 + nt * 16 + il] = acc[nt][r];
+}
+
+#define MFC1B_SPLIT 36
+#define MFC1B_CH 256
+#define MFC1B_LD 264
+__global__ __launch_bounds__(256)
+void k_fc1_fwd_mfma_mb_bf16(const float* __restrict__ a2,
+                            const float* __restrict__ params, long long P,
+                            long long ow3, int bs, int K,
+                            float* __restrict__ slab) {
+  __shared__ __bf16 lw[128 * MFC1B_LD];
+  __shared__ __bf16 la[32 * MFC1B_LD];
+  int k = blockIdx.x / MFC1B_SPLIT;
+  int s = blockIdx.x % MFC1B_SPLIT;
+  int k_base = s * MFC1B_CH;
+  const float* w3 = params + (long long)k * P + ow3;
+  const float* a2k = a2 + (long long)k * bs * 9216;
+  for (int i = threadIdx.x; i < 128 * MFC1B_CH; i += 256) {
+    int row = i / MFC1B_CH, kk = i % MFC1B_CH;
+    lw[row * MFC1B_LD + kk] = (__bf16)w3[(long long)row * 9216 + k_base + kk];
+  }
+  for (int i = threadIdx.x; i < 32 * MFC1B_CH; i += 256) {
+    int bu = i / MFC1B_CH, kk = i % MFC1B_CH;
+    la[bu * MFC1B_LD + kk] =
+        (__bf16)(bu < bs ? a2k[(long long)bu * 9216 + k_base + kk] : 0.f);
+  }
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int il = lane & 15, kc8 = (lane >> 4) * 8;
+  f32x4 acc[2][2] = {{f32x4{0,0,0,0}, f32x4{0,0,0,0}},
+                     {f32x4{0,0,0,0}, f32x4{0,0,0,0}}};
+  for (int k0 = 0; k0 < MFC1B_CH; k0 += 32) {
+    bf16x8m a0 = ld_bf16x8m(&lw[(w * 32 + il) * MFC1B_LD + k0 + kc8]);
+    bf16x8m a1v = ld_bf16x8m(&lw[(w * 32 + 16 + il) * MFC1B_LD + k0 + kc8]);
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      bf16x8m bv = ld_bf16x8m(&la[(u * 16 + il) * MFC1B_LD + k0 + kc8]);
+      acc[0][u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bv, acc[0][u], 0, 0, 0);
+      acc[1][u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1v, bv, acc[1][u], 0, 0, 0);
+    }
+  }
+  float* slk = slab + (long long)k * MFC1B_SPLIT * bs * 128;
+  #pragma unroll
+  for (int tt = 0; tt < 2; ++tt)
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int bu = u * 16 + il;
+      if (bu >= bs) continue;
+      int j = w * 32 + tt * 16 + (lane >> 4) * 4;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        slk[((long long)s * bs + bu) * 128 + j + r] = acc[tt][u][r];
+    }
+}
+
+__global__ void k_fc1_fwd_reduce_mb_bf16(const float* __restrict__ slab,
+                                         const float* __restrict__ params,
+                                         long long P, long long ob3, int bs,
+                                         int K, float p2,
+                                         const long long* __restrict__ seeds,
+                                         unsigned long long offset,
+                                         float* __restrict__ z3,
+                                         float* __restrict__ a3,
+                                         unsigned char* __restrict__ m3) {
+  long long total = (long long)K * bs * 128;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long g = i / 128;
+    int j = (int)(i % 128);
+    int k = (int)(g / bs), b = (int)(g % bs);
+    const float* slk = slab + (long long)k * MFC1B_SPLIT * bs * 128;
+    float t = params[(long long)k * P + ob3 + j];
+    for (int s = 0; s < MFC1B_SPLIT; ++s)
+      t += slk[((long long)s * bs + b) * 128 + j];
+    z3[i] = t;
+    float r = t > 0.f ? t : 0.f;
+    unsigned char keep = 1;
+    if (p2 > 0.f) {
+      long long li = (long long)b * 128 + j;
+      hiprandStatePhilox4_32_10_t st;
+      hiprand_init((unsigned long long)seeds[k] ^ 0x9e3779b97f4a7c15ull,
+                   (unsigned long long)li, offset, &st);
+      keep = hiprand_uniform(&st) >= p2;
+    }
+    m3[i] = keep;
+    a3[i] = keep ? r / (1.f - p2) : 0.f;
+  }
+}
+
+
+// ---------------------------------------------------------------------------
 // host driver: one launch set per batch-step for ALL K clients
 // ---------------------------------------------------------------------------
 extern "C" void launch_cnn_round_mega(
@@ -792,7 +1036,7 @@ extern "C" void launch_cnn_round_mega(
     int* yb, unsigned char* pidx, unsigned char* m2, unsigned char* m3,
     double* acc2k,
     const float* lr_t, float max_norm, float p1, float p2,
-    float* stats_out, float* loss_out, hipStream_t s) {
+    float* stats_out, float* loss_out, hipStream_t s, int use_bf16) {
   MegaOffsets o = mega_offsets(C);
   long long P = o.total;
   int G = K * bs;
@@ -816,22 +1060,36 @@ extern "C" void launch_cnn_round_mega(
                        dim3((int)(((long long)G * 21632 + FBLK - 1) / FBLK)),
                        dim3(FBLK), 0, s, xb, params_stack, P, o.w1, o.b1,
                        bs, K, a1);
-    hipLaunchKernelGGL(k_w2_layouts_mb,
-                       dim3((int)(((long long)K * 18432 + FBLK - 1) / FBLK)),
-                       dim3(FBLK), 0, s, params_stack, P, o.w2, K,
-                       w2t_stack, w2rot_stack);
-    hipLaunchKernelGGL(k_conv2_fwd_mfma_mb, dim3(G * 9), dim3(FBLK), 0, s,
-                       a1, w2t_stack, params_stack, P, o.b2, bs, K, r2);
+    if (use_bf16) {
+      hipLaunchKernelGGL(k_conv2_fwd_mfma_mb_bf16, dim3(G * 9), dim3(FBLK),
+                         0, s, a1, params_stack, P, o.w2, o.b2, bs, K, r2);
+    } else {
+      hipLaunchKernelGGL(k_w2_layouts_mb,
+                         dim3((int)(((long long)K * 18432 + FBLK - 1) / FBLK)),
+                         dim3(FBLK), 0, s, params_stack, P, o.w2, K,
+                         w2t_stack, w2rot_stack);
+      hipLaunchKernelGGL(k_conv2_fwd_mfma_mb, dim3(G * 9), dim3(FBLK), 0, s,
+                         a1, w2t_stack, params_stack, P, o.b2, bs, K, r2);
+    }
     hipLaunchKernelGGL(k_pool_drop_fwd_mb,
                        dim3((int)(((long long)G * 9216 + FBLK - 1) / FBLK)),
                        dim3(FBLK), 0, s, r2, bs, K, p1, seeds_dev, off,
                        a2, pidx, m2);
-    hipLaunchKernelGGL(k_fc1_fwd_mfma_mb, dim3(K * FC1M_SPLIT), dim3(FBLK),
-                       0, s, a2, params_stack, P, o.w3, bs, K, slab);
-    hipLaunchKernelGGL(k_fc1_fwd_reduce_mb,
-                       dim3((int)(((long long)G * 128 + FBLK - 1) / FBLK)),
-                       dim3(FBLK), 0, s, slab, params_stack, P, o.b3, bs, K,
-                       p2, seeds_dev, off, z3, a3, m3);
+    if (use_bf16) {
+      hipLaunchKernelGGL(k_fc1_fwd_mfma_mb_bf16, dim3(K * 36), dim3(FBLK),
+                         0, s, a2, params_stack, P, o.w3, bs, K, slab);
+      hipLaunchKernelGGL(k_fc1_fwd_reduce_mb_bf16,
+                         dim3((int)(((long long)G * 128 + FBLK - 1) / FBLK)),
+                         dim3(FBLK), 0, s, slab, params_stack, P, o.b3, bs,
+                         K, p2, seeds_dev, off, z3, a3, m3);
+    } else {
+      hipLaunchKernelGGL(k_fc1_fwd_mfma_mb, dim3(K * FC1M_SPLIT), dim3(FBLK),
+                         0, s, a2, params_stack, P, o.w3, bs, K, slab);
+      hipLaunchKernelGGL(k_fc1_fwd_reduce_mb,
+                         dim3((int)(((long long)G * 128 + FBLK - 1) / FBLK)),
+                         dim3(FBLK), 0, s, slab, params_stack, P, o.b3, bs,
+                         K, p2, seeds_dev, off, z3, a3, m3);
+    }
     hipLaunchKernelGGL(k_fc2_loss_fwd_mb, dim3(G), dim3(FBLK),
                        C * (int)sizeof(float), s, a3, params_stack, P, o.w4,
                        o.b4, counts_dev, t, bs, K, C, yb, dlogits, loss_out);
@@ -851,18 +1109,27 @@ extern "C" void launch_cnn_round_mega(
     hipLaunchKernelGGL(k_pool_drop_bwd_mb,
                        dim3((int)(((long long)G * 9216 + FBLK - 1) / FBLK)),
                        dim3(FBLK), 0, s, da2, pidx, m2, r2, bs, K, p1, dz2);
-    hipLaunchKernelGGL(k_dz2_transpose_mb,
-                       dim3((int)(((long long)G * 36864 + FBLK - 1) / FBLK)),
-                       dim3(FBLK), 0, s, dz2, bs, K, r2);  // r2 free now
-    hipLaunchKernelGGL(k_conv2_bwd_w_mfma_mb, dim3(6 * G), dim3(FBLK), 0, s,
-                       r2, a1, bs, K, slab);
+    if (use_bf16) {
+      hipLaunchKernelGGL(k_conv2_bwd_w_mfma_mb_bf16, dim3(6 * G), dim3(FBLK),
+                         0, s, dz2, a1, bs, K, slab);
+    } else {
+      hipLaunchKernelGGL(k_dz2_transpose_mb,
+                         dim3((int)(((long long)G * 36864 + FBLK - 1) / FBLK)),
+                         dim3(FBLK), 0, s, dz2, bs, K, r2);  // r2 free now
+      hipLaunchKernelGGL(k_conv2_bwd_w_mfma_mb, dim3(6 * G), dim3(FBLK), 0, s,
+                         r2, a1, bs, K, slab);
+    }
     int perk_fold = (18432 + FBLK - 1) / FBLK;
     hipLaunchKernelGGL(k_conv2_bwd_w_fold_mb, dim3(K * perk_fold),
                        dim3(FBLK), 0, s, slab, grads_stack, P, o.w2, bs, K);
     hipLaunchKernelGGL(k_conv2_bwd_b_mb, dim3(K * 64), dim3(FBLK), 0, s,
                        dz2, grads_stack, P, o.b2, bs, K);
-    hipLaunchKernelGGL(k_conv2_bwd_x_mfma_mb, dim3(G * 11), dim3(FBLK),
-                       0, s, dz2, w2rot_stack, a1, bs, K, dz1);
+    if (use_bf16)
+      hipLaunchKernelGGL(k_conv2_bwd_x_mfma_mb_bf16, dim3(G * 11), dim3(FBLK),
+                         0, s, dz2, params_stack, P, o.w2, bs, K, a1, dz1);
+    else
+      hipLaunchKernelGGL(k_conv2_bwd_x_mfma_mb, dim3(G * 11), dim3(FBLK),
+                         0, s, dz2, w2rot_stack, a1, bs, K, dz1);
     hipLaunchKernelGGL(k_conv1_bwd_w_mb, dim3(K * 32), dim3(1024), 0, s,
                        xb, dz1, grads_stack, P, o.w1, o.b1, bs, K);
     hipMemsetAsync(acc2k, 0, 2 * K * sizeof(double), s);

@@ -130,7 +130,8 @@ class MegaRound:
     epoch (client-local indices)."""
 
     def __init__(self, arena: ParameterArena, num_classes: int, bs: int,
-                 p1: float, p2: float, max_grad_norm, k_cap: int = 32):
+                 p1: float, p2: float, max_grad_norm, k_cap: int = 32,
+                 use_bf16: bool = False):
         assert HAS_EXT and arena.device.type == "cuda"
         self.arena = arena
         self.C = int(num_classes)
@@ -138,6 +139,7 @@ class MegaRound:
         self.p1, self.p2 = float(p1), float(p2)
         self.max_norm = float(max_grad_norm) if max_grad_norm else -1.0
         self.k_cap = int(k_cap)
+        self.use_bf16 = bool(use_bf16)
         self._alloc_k = 0
         dev = arena.device
         self.lr_t = torch.zeros(1, dtype=torch.float32, device=dev)
@@ -232,7 +234,7 @@ class MegaRound:
             server_arena.data, self.params_stack, self.grads_stack,
             round_accum, self.work_f, self.work_i, self.work_b, self.work_d,
             self.lr_t, self.max_norm, self.p1, self.p2,
-            self.stats_out, self.loss_out)
+            self.stats_out, self.loss_out, self.use_bf16)
         now = _time.time()
         outputs = []
         for k, cid in enumerate(client_ids):
