@@ -687,7 +687,7 @@ __global__ void k_conv1_bwd_w_mb(const float* __restrict__ x,
 // + stats accumulate + SGD in one elementwise pass.
 // one block per (client, contiguous chunk): block-local tree reduce,
 // ONE f64 atomicAdd pair per block (~K * ceil(P/65536) atomics per step)
-#define SUMSQ_CHUNK 65536
+#define SUMSQ_CHUNK 32768
 __global__ void k_sumsq_mb(const float* __restrict__ grads, long long P,
                            int blocks_per_k, double* __restrict__ acc) {
   int k = blockIdx.x / blocks_per_k;
@@ -696,10 +696,27 @@ __global__ void k_sumsq_mb(const float* __restrict__ grads, long long P,
   long long hi = lo + SUMSQ_CHUNK;
   if (hi > P) hi = P;
   const float* g = grads + (long long)k * P;
-  double fs = 0.0, fq = 0.0;
-  for (long long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-    double v = (double)g[i];
-    fs += v; fq += v * v;
+  // float4 loads + independent accumulator pairs break the dependent
+  // f64-add chain (was ~74 us per step at 190 underfilled blocks)
+  double s0 = 0.0, s1 = 0.0, s2 = 0.0, s3 = 0.0;
+  double q0 = 0.0, q1 = 0.0, q2 = 0.0, q3 = 0.0;
+  // scalar loads (the k*P segment bases are not 16 B aligned); the win
+  // is the 4 independent f64 accumulator chains
+  long long i = lo + (long long)threadIdx.x * 4;
+  for (; i + 3 < hi; i += (long long)blockDim.x * 4) {
+    double a = g[i], b = g[i + 1], cc = g[i + 2], d = g[i + 3];
+    s0 += a; q0 += a * a;
+    s1 += b; q1 += b * b;
+    s2 += cc; q2 += cc * cc;
+    s3 += d; q3 += d * d;
+  }
+  for (long long j = i; j < hi; ++j) {  // ragged tail (P % 4)
+    double v = (double)g[j];
+    s0 += v; q0 += v * v;
+  }
+  double fs = (s0 + s1) + (s2 + s3);
+  double fq = (q0 + q1) + (q2 + q3);
+  {
   }
   for (int d = 32; d > 0; d >>= 1) {
     fs += __shfl_down(fs, d, 64);
@@ -790,6 +807,22 @@ __device__ __forceinline__ bf16x8m ld_bf16x8m(const __bf16* p) {
   return *reinterpret_cast<const bf16x8m*>(p);
 }
 
+// [k][ci][k'=(co,kh,kw)] bf16 layout for the bwd-data B operand (built
+// once per batch into the w2rot_stack buffer reinterpreted as bf16)
+__global__ void k_w2rotbf_mb(const float* __restrict__ params, long long P,
+                             long long ow2, int K,
+                             __bf16* __restrict__ w2rotbf) {
+  long long total = (long long)K * 18432;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    int k = (int)(i / 18432), r = (int)(i % 18432);
+    int co = r / 288, rem9 = r % 288;
+    int ci = rem9 / 9, rem = rem9 % 9;
+    w2rotbf[(long long)k * 18432 + (long long)ci * 576 + co * 9 + rem] =
+        (__bf16)params[(long long)k * P + ow2 + r];
+  }
+}
+
 #define MC2_LD 296
 __global__ __launch_bounds__(256)
 void k_conv2_fwd_mfma_mb_bf16(const float* __restrict__ a1,
@@ -836,75 +869,75 @@ void k_conv2_fwd_mfma_mb_bf16(const float* __restrict__ a1,
 }
 
 #define MC2B_LD 584
+// 32-row m-tiles (imc 37.4 KB -> 4 blocks/CU of 4 waves hide the staging
+// latency that bound the 64-row form to ~340 us); the B operand streams
+// from the pre-built global w2rotbf (18 16-B reads per lane, unrolled).
+// Wave w: m-subtile (w & 1), ci-tile (w >> 1).
 __global__ __launch_bounds__(256)
 void k_conv2_bwd_x_mfma_mb_bf16(const float* __restrict__ dz2,
-                                const float* __restrict__ params,
-                                long long P, long long ow2, int bs, int K,
+                                const __bf16* __restrict__ w2rotbf,
+                                int bs, int K,
                                 const float* __restrict__ a1,
                                 float* __restrict__ dz1) {
-  __shared__ __bf16 imc[64 * MC2B_LD];
-  __shared__ __bf16 wt[32 * MC2B_LD];
-  long long g = blockIdx.x / 11;
-  int mt = blockIdx.x % 11;
+  __shared__ __bf16 imc[32 * MC2B_LD];
+  long long g = blockIdx.x / 22;
+  int mt = blockIdx.x % 22;
   int k = (int)(g / bs);
   const float* dzb = dz2 + g * 36864;
-  const float* w2 = params + (long long)k * P + ow2;
-  for (int i = threadIdx.x; i < 64 * 576; i += 256) {
+  const __bf16* wtk = w2rotbf + (long long)k * 18432;
+  for (int i = threadIdx.x; i < 32 * 576; i += 256) {
     int mr = i / 576, kk = i % 576;
-    int m = mt * 64 + mr, p = m / 26, q = m % 26;
+    int m = mt * 32 + mr, p = m / 26, q = m % 26;
     int co = kk / 9, rem = kk % 9, kh = rem / 3, kw = rem % 3;
     int y = p - kh, x = q - kw;
     imc[mr * MC2B_LD + kk] =
         (__bf16)((m < 676 && y >= 0 && y < 24 && x >= 0 && x < 24)
                      ? dzb[co * 576 + y * 24 + x] : 0.f);
   }
-  for (int i = threadIdx.x; i < 32 * 576; i += 256) {
-    int ci = i / 576, kk = i % 576;
-    int co = kk / 9, rem = kk % 9;
-    wt[ci * MC2B_LD + kk] = (__bf16)w2[(long long)co * 288 + ci * 9 + rem];
-  }
   __syncthreads();
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
   int il = lane & 15, kc8 = (lane >> 4) * 8;
-  f32x4 acc[2] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  int msub = w & 1, nt = w >> 1;
+  f32x4 acc = {0, 0, 0, 0};
+  #pragma unroll 3
   for (int k0 = 0; k0 < 576; k0 += 32) {
-    bf16x8m a = ld_bf16x8m(&imc[(w * 16 + il) * MC2B_LD + k0 + kc8]);
-    #pragma unroll
-    for (int nt = 0; nt < 2; ++nt) {
-      bf16x8m bv = ld_bf16x8m(&wt[(nt * 16 + il) * MC2B_LD + k0 + kc8]);
-      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc[nt], 0, 0, 0);
-    }
+    bf16x8m a = ld_bf16x8m(&imc[(msub * 16 + il) * MC2B_LD + k0 + kc8]);
+    bf16x8m bv = ld_bf16x8m(&wtk[(long long)(nt * 16 + il) * 576
+                                 + k0 + kc8]);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc, 0, 0, 0);
   }
-  int om = mt * 64 + w * 16 + (lane >> 4) * 4;
+  int om = mt * 32 + msub * 16 + (lane >> 4) * 4;
+  int ci = nt * 16 + il;
   #pragma unroll
   for (int r = 0; r < 4; ++r) {
     if (om + r >= 676) continue;
-    #pragma unroll
-    for (int nt = 0; nt < 2; ++nt) {
-      int ci = nt * 16 + il;
-      long long o = (g * 32 + ci) * 676 + om + r;
-      dz1[o] = a1[o] > 0.f ? acc[nt][r] : 0.f;
-    }
+    long long o = (g * 32 + ci) * 676 + om + r;
+    dz1[o] = a1[o] > 0.f ? acc[r] : 0.f;
   }
 }
 
+// block = 32 co-rows x 32 n-cols over K=576 (dzb16 37.4 KB + imt
+// 37.4 KB -> 2 blocks/CU); grid = G * 2(co) * 9(n).  Wave w: co-subtile
+// (w & 1), n-subtile (w >> 1).
 __global__ __launch_bounds__(256)
 void k_conv2_bwd_w_mfma_mb_bf16(const float* __restrict__ dz2,
                                 const float* __restrict__ a1, int bs, int K,
                                 float* __restrict__ slab) {
-  __shared__ __bf16 dzb16[64 * MC2B_LD];
-  __shared__ __bf16 imt[48 * MC2B_LD];
-  int nb = blockIdx.x % 6;
-  long long g = blockIdx.x / 6;
-  const float* dzb = dz2 + g * 36864;
+  __shared__ __bf16 dzb16[32 * MC2B_LD];
+  __shared__ __bf16 imt[32 * MC2B_LD];
+  int blk = blockIdx.x % 18;
+  int ch = blk / 9;         // co half (0: rows 0-31, 1: rows 32-63)
+  int nb = blk % 9;         // 32-col n-block
+  long long g = blockIdx.x / 18;
+  const float* dzb = dz2 + g * 36864 + (long long)ch * 32 * 576;
   const float* a1b = a1 + g * 21632;
-  for (int i = threadIdx.x; i < 64 * 576; i += 256) {
-    int co = i / 576, o = i % 576;
-    dzb16[co * MC2B_LD + o] = (__bf16)dzb[i];
+  for (int i = threadIdx.x; i < 32 * 576; i += 256) {
+    int co = i / 576;
+    dzb16[co * MC2B_LD + (i % 576)] = (__bf16)dzb[i];
   }
-  for (int i = threadIdx.x; i < 48 * 576; i += 256) {
+  for (int i = threadIdx.x; i < 32 * 576; i += 256) {
     int nr = i / 576, o = i % 576;
-    int n = nb * 48 + nr;
+    int n = nb * 32 + nr;
     int ci = n / 9, rem = n % 9, kh = rem / 3, kw = rem % 3;
     int yy = o / 24, xx = o % 24;
     imt[nr * MC2B_LD + o] = (__bf16)a1b[ci * 676 + (yy + kh) * 26 + xx + kw];
@@ -912,22 +945,19 @@ void k_conv2_bwd_w_mfma_mb_bf16(const float* __restrict__ dz2,
   __syncthreads();
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
   int il = lane & 15, kc8 = (lane >> 4) * 8;
-  f32x4 acc[3] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}};
+  int csub = w & 1, nt = w >> 1;
+  f32x4 acc = {0, 0, 0, 0};
+  #pragma unroll 3
   for (int k0 = 0; k0 < 576; k0 += 32) {
-    bf16x8m a = ld_bf16x8m(&dzb16[(w * 16 + il) * MC2B_LD + k0 + kc8]);
-    #pragma unroll
-    for (int nt = 0; nt < 3; ++nt) {
-      bf16x8m bv = ld_bf16x8m(&imt[(nt * 16 + il) * MC2B_LD + k0 + kc8]);
-      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc[nt], 0, 0, 0);
-    }
+    bf16x8m a = ld_bf16x8m(&dzb16[(csub * 16 + il) * MC2B_LD + k0 + kc8]);
+    bf16x8m bv = ld_bf16x8m(&imt[(nt * 16 + il) * MC2B_LD + k0 + kc8]);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc, 0, 0, 0);
   }
   float* out = slab + g * 18432;
-  int orow = w * 16 + (lane >> 4) * 4;
+  int orow = ch * 32 + csub * 16 + (lane >> 4) * 4;
   #pragma unroll
   for (int r = 0; r < 4; ++r)
-    #pragma unroll
-    for (int nt = 0; nt < 3; ++nt)
-      out[(orow + r) * 288 + nb * 48 + nt * 16 + il] = acc[nt][r];
+    out[(orow + r) * 288 + nb * 32 + nt * 16 + il] = acc[r];
 }
 
 #define MFC1B_SPLIT 36
@@ -1061,6 +1091,10 @@ extern "C" void launch_cnn_round_mega(
                        dim3(FBLK), 0, s, xb, params_stack, P, o.w1, o.b1,
                        bs, K, a1);
     if (use_bf16) {
+      hipLaunchKernelGGL(k_w2rotbf_mb,
+                         dim3((int)(((long long)K * 18432 + FBLK - 1) / FBLK)),
+                         dim3(FBLK), 0, s, params_stack, P, o.w2, K,
+                         reinterpret_cast<__bf16*>(w2rot_stack));
       hipLaunchKernelGGL(k_conv2_fwd_mfma_mb_bf16, dim3(G * 9), dim3(FBLK),
                          0, s, a1, params_stack, P, o.w2, o.b2, bs, K, r2);
     } else {
@@ -1110,7 +1144,7 @@ extern "C" void launch_cnn_round_mega(
                        dim3((int)(((long long)G * 9216 + FBLK - 1) / FBLK)),
                        dim3(FBLK), 0, s, da2, pidx, m2, r2, bs, K, p1, dz2);
     if (use_bf16) {
-      hipLaunchKernelGGL(k_conv2_bwd_w_mfma_mb_bf16, dim3(6 * G), dim3(FBLK),
+      hipLaunchKernelGGL(k_conv2_bwd_w_mfma_mb_bf16, dim3(18 * G), dim3(FBLK),
                          0, s, dz2, a1, bs, K, slab);
     } else {
       hipLaunchKernelGGL(k_dz2_transpose_mb,
@@ -1125,8 +1159,10 @@ extern "C" void launch_cnn_round_mega(
     hipLaunchKernelGGL(k_conv2_bwd_b_mb, dim3(K * 64), dim3(FBLK), 0, s,
                        dz2, grads_stack, P, o.b2, bs, K);
     if (use_bf16)
-      hipLaunchKernelGGL(k_conv2_bwd_x_mfma_mb_bf16, dim3(G * 11), dim3(FBLK),
-                         0, s, dz2, params_stack, P, o.w2, bs, K, a1, dz1);
+      hipLaunchKernelGGL(k_conv2_bwd_x_mfma_mb_bf16, dim3(G * 22), dim3(FBLK),
+                         0, s, dz2,
+                         reinterpret_cast<const __bf16*>(w2rot_stack),
+                         bs, K, a1, dz1);
     else
       hipLaunchKernelGGL(k_conv2_bwd_x_mfma_mb, dim3(G * 11), dim3(FBLK),
                          0, s, dz2, w2rot_stack, a1, bs, K, dz1);
