@@ -26,6 +26,9 @@
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
+#define MC2F_LD 585  /* padded f32 LDS row stride (585 %% 32 = 9, odd ->
+                        conflict-free 16-lane fragment groups) */
+
 struct MegaOffsets {
   long long w1, b1, w2, b2, w3, b3, w4, b4, total;
 };
@@ -512,42 +515,57 @@ __global__ void k_dz2_transpose_mb(const float* __restrict__ dz2, int bs,
   }
 }
 
-// conv2 backward weights (f32 MFMA): grid = 6 * G; per-row slabs then fold
+// conv2 backward weights (f32 MFMA): 32x32 (co x n) tiles, K=576
+// staged in TWO 288-deep chunks so the live LDS is 74 KB -> 2
+// blocks/CU (a one-shot 576-deep stage was 149 KB -> 1 block/CU and
+// measured WORSE than the 64-row form).  A = dz2t rows; B = im2col^T.
+// grid = G * 18 (2 co-halves x 9 n-blocks); one accumulator per wave.
+#define MC2FW_LD 289  /* 289 %% 32 = 1 -> conflict-free fragment groups */
 __global__ __launch_bounds__(256)
 void k_conv2_bwd_w_mfma_mb(const float* __restrict__ dz2t,
                            const float* __restrict__ a1, int bs, int K,
                            float* __restrict__ slab) {
-  __shared__ float lds[32 * 676];
-  int nb = blockIdx.x % 6;
-  long long g = blockIdx.x / 6;
-  const float* src = a1 + g * 21632;
-  for (int i = threadIdx.x; i < 21632; i += 256) lds[i] = src[i];
-  __syncthreads();
+  __shared__ float dzl[32 * MC2FW_LD];
+  __shared__ float imt[32 * MC2FW_LD];
+  int blk = blockIdx.x % 18;
+  int ch = blk / 9, nb = blk % 9;
+  long long g = blockIdx.x / 18;
+  const float* dzb = dz2t + g * 36864;   // [o][co] layout
+  const float* a1b = a1 + g * 21632;
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
-  int co = w * 16 + (lane & 15);
-  int kc = lane >> 4;
-  const float* dzb = dz2t + g * 36864;
-  f32x4 acc[3] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}, f32x4{0,0,0,0}};
-  #pragma unroll 4
-  for (int k0 = 0; k0 < 576; k0 += 4) {
-    int o = k0 + kc;
-    int yy = o / 24, xx = o % 24;
-    float a = dzb[o * 64 + co];
-    #pragma unroll
-    for (int nt = 0; nt < 3; ++nt) {
-      int n = nb * 48 + nt * 16 + (lane & 15);
+  int il = lane & 15, kc = lane >> 4;
+  int csub = w & 1, nt = w >> 1;
+  f32x4 acc = {0, 0, 0, 0};
+  for (int c = 0; c < 2; ++c) {
+    int obase = c * 288;
+    __syncthreads();
+    for (int i = threadIdx.x; i < 32 * 288; i += 256) {
+      // thread-consecutive co -> 128 B contiguous dz2t reads per o
+      int o = obase + i / 32, co = i % 32;
+      dzl[co * MC2FW_LD + (o - obase)] = dzb[o * 64 + ch * 32 + co];
+    }
+    for (int i = threadIdx.x; i < 32 * 288; i += 256) {
+      int nr = i / 288, o = obase + i % 288;
+      int n = nb * 32 + nr;
       int ci = n / 9, rem = n % 9, kh = rem / 3, kw = rem % 3;
-      float bv = lds[ci * 676 + (yy + kh) * 26 + xx + kw];
-      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[nt], 0, 0, 0);
+      int yy = o / 24, xx = o % 24;
+      imt[nr * MC2FW_LD + (o - obase)] =
+          a1b[ci * 676 + (yy + kh) * 26 + xx + kw];
+    }
+    __syncthreads();
+    #pragma unroll 8
+    for (int k0 = 0; k0 < 288; k0 += 4) {
+      int o = k0 + kc;
+      float a = dzl[(csub * 16 + il) * MC2FW_LD + o];
+      float bv = imt[(nt * 16 + il) * MC2FW_LD + o];
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
     }
   }
   float* out = slab + g * 18432;
-  int orow = w * 16 + (lane >> 4) * 4;
+  int orow = ch * 32 + csub * 16 + (lane >> 4) * 4;
   #pragma unroll
   for (int r = 0; r < 4; ++r)
-    #pragma unroll
-    for (int nt = 0; nt < 3; ++nt)
-      out[(orow + r) * 288 + nb * 48 + nt * 16 + (lane & 15)] = acc[nt][r];
+    out[(orow + r) * 288 + nb * 32 + nt * 16 + il] = acc[r];
 }
 
 __global__ void k_conv2_bwd_w_fold_mb(const float* __restrict__ slab,
@@ -585,51 +603,49 @@ __global__ void k_conv2_bwd_b_mb(const float* __restrict__ dz2,
   }
 }
 
-// conv2 backward data (f32 MFMA): grid = G * 11
+// conv2 backward data (f32 MFMA): 32-row materialized-im2col tiles
+// (74.9 KB LDS -> 2 blocks/CU), one accumulator per wave (wave w:
+// m-subtile w&1, ci-tile w>>1), B streamed from the per-batch global
+// w2rot layout with lane-contiguous 64 B groups.  grid = G * 22.
 __global__ __launch_bounds__(256)
 void k_conv2_bwd_x_mfma_mb(const float* __restrict__ dz2,
                            const float* __restrict__ w2rot_stack,
                            const float* __restrict__ a1, int bs, int K,
                            float* __restrict__ dz1) {
-  __shared__ float lds[64 * 576];
-  long long g = blockIdx.x / 11;
-  int mt = blockIdx.x % 11;
+  __shared__ float imc[32 * MC2F_LD];
+  long long g = blockIdx.x / 22;
+  int mt = blockIdx.x % 22;
   int k = (int)(g / bs);
-  const float* src = dz2 + g * 36864;
-  for (int i = threadIdx.x; i < 36864; i += 256) lds[i] = src[i];
-  __syncthreads();
+  const float* dzb = dz2 + g * 36864;
   const float* w2rot = w2rot_stack + (long long)k * 18432;
-  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
-  int m = mt * 64 + w * 16 + (lane & 15);
-  int p = m / 26, q = m % 26;
-  bool mrow = m < 676;
-  int kc = lane >> 4;
-  f32x4 acc[2] = {f32x4{0,0,0,0}, f32x4{0,0,0,0}};
-  #pragma unroll 4
-  for (int k0 = 0; k0 < 576; k0 += 4) {
-    int kk = k0 + kc;
+  for (int i = threadIdx.x; i < 32 * 576; i += 256) {
+    int mr = i / 576, kk = i % 576;
+    int m = mt * 32 + mr, p = m / 26, q = m % 26;
     int co = kk / 9, rem = kk % 9, kh = rem / 3, kw = rem % 3;
     int y = p - kh, x = q - kw;
-    float a = (mrow && y >= 0 && y < 24 && x >= 0 && x < 24)
-                  ? lds[co * 576 + y * 24 + x] : 0.f;
-    const float* wrow = w2rot + (long long)kk * 32 + (lane & 15);
-    #pragma unroll
-    for (int nt = 0; nt < 2; ++nt) {
-      float bv = wrow[nt * 16];
-      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc[nt], 0, 0, 0);
-    }
+    imc[mr * MC2F_LD + kk] =
+        (m < 676 && y >= 0 && y < 24 && x >= 0 && x < 24)
+            ? dzb[co * 576 + y * 24 + x] : 0.f;
   }
-  int om = mt * 64 + w * 16 + (lane >> 4) * 4;
-  int cl = lane & 15;
+  __syncthreads();
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int il = lane & 15, kc = lane >> 4;
+  int msub = w & 1, nt = w >> 1;
+  f32x4 acc = {0, 0, 0, 0};
+  #pragma unroll 8
+  for (int k0 = 0; k0 < 576; k0 += 4) {
+    int kk = k0 + kc;
+    float a = imc[(msub * 16 + il) * MC2F_LD + kk];
+    float bv = w2rot[(long long)kk * 32 + nt * 16 + il];
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+  }
+  int om = mt * 32 + msub * 16 + (lane >> 4) * 4;
+  int ci = nt * 16 + il;
   #pragma unroll
   for (int r = 0; r < 4; ++r) {
     if (om + r >= 676) continue;
-    #pragma unroll
-    for (int nt = 0; nt < 2; ++nt) {
-      int ci = nt * 16 + cl;
-      long long o = (g * 32 + ci) * 676 + om + r;
-      dz1[o] = a1[o] > 0.f ? acc[nt][r] : 0.f;
-    }
+    long long o = (g * 32 + ci) * 676 + om + r;
+    dz1[o] = a1[o] > 0.f ? acc[r] : 0.f;
   }
 }
 
@@ -1150,8 +1166,8 @@ extern "C" void launch_cnn_round_mega(
       hipLaunchKernelGGL(k_dz2_transpose_mb,
                          dim3((int)(((long long)G * 36864 + FBLK - 1) / FBLK)),
                          dim3(FBLK), 0, s, dz2, bs, K, r2);  // r2 free now
-      hipLaunchKernelGGL(k_conv2_bwd_w_mfma_mb, dim3(6 * G), dim3(FBLK), 0, s,
-                         r2, a1, bs, K, slab);
+      hipLaunchKernelGGL(k_conv2_bwd_w_mfma_mb, dim3(18 * G), dim3(FBLK),
+                         0, s, r2, a1, bs, K, slab);
     }
     int perk_fold = (18432 + FBLK - 1) / FBLK;
     hipLaunchKernelGGL(k_conv2_bwd_w_fold_mb, dim3(K * perk_fold),
@@ -1164,7 +1180,7 @@ extern "C" void launch_cnn_round_mega(
                          reinterpret_cast<const __bf16*>(w2rot_stack),
                          bs, K, a1, dz1);
     else
-      hipLaunchKernelGGL(k_conv2_bwd_x_mfma_mb, dim3(G * 11), dim3(FBLK),
+      hipLaunchKernelGGL(k_conv2_bwd_x_mfma_mb, dim3(G * 22), dim3(FBLK),
                          0, s, dz2, w2rot_stack, a1, bs, K, dz1);
     hipLaunchKernelGGL(k_conv1_bwd_w_mb, dim3(K * 32), dim3(1024), 0, s,
                        xb, dz1, grads_stack, P, o.w1, o.b1, bs, K);
