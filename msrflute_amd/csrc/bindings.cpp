@@ -53,6 +53,12 @@ extern "C" void launch_lstm_seq_fwd(const float*, const float*, float*,
 extern "C" void launch_lstm_seq_bwd(const float*, const float*, const float*,
                                     const float*, float*, int, int, int,
                                     hipStream_t);
+extern "C" void launch_lstm_seq_fwd_b(const float*, const float*, float*,
+                                      float*, float*, int, int, int,
+                                      hipStream_t);
+extern "C" void launch_lstm_seq_bwd_b(const float*, const float*,
+                                      const float*, const float*, float*,
+                                      int, int, int, hipStream_t);
 extern "C" void launch_gru_seq_fwd(const float*, const float*, const float*,
                                    float*, float*, float*, int, int,
                                    hipStream_t);
@@ -115,6 +121,13 @@ extern "C" void launch_cnn_round_mega(
     double* acc2k,
     const float* lr_t, float max_norm, float p1, float p2,
     float* stats_out, float* loss_out, hipStream_t s, int use_bf16);
+
+extern "C" void launch_mega_clip_sgd(float*, float*, long long, int,
+                                     double*, float, const float*, float*,
+                                     hipStream_t);
+extern "C" void launch_mega_pseudo_accum(float*, const float*, const float*,
+                                         const float*, float*, long long,
+                                         int, hipStream_t);
 
 extern "C" void launch_cnn_round(
     const float* shard_x, const long long* shard_y, const long long* orders,
@@ -473,6 +486,51 @@ torch::Tensor lstm_seq_bwd(torch::Tensor gates, torch::Tensor c_seq,
   return dg;
 }
 
+// batched (cross-client) LSTM recurrence: xp [R, T, 4H], K-stacked
+// transposed hidden weights [K, H, 4H]; row r belongs to client
+// r / rows_per_client (Shakespeare mega round)
+std::vector<torch::Tensor> lstm_seq_fwd_b(torch::Tensor xp,
+                                          torch::Tensor w_hh_t_stack,
+                                          int64_t rows_per_client) {
+  check_flat(xp, "xp"); check_flat(w_hh_t_stack, "w_hh_t_stack");
+  TORCH_CHECK(xp.dim() == 3 && xp.size(2) == 4 * 256,
+              "xp must be [R, T, 4*256]");
+  long long R = xp.size(0), T = xp.size(1);
+  TORCH_CHECK(rows_per_client > 0 && R % rows_per_client == 0);
+  long long K = R / rows_per_client;
+  TORCH_CHECK(w_hh_t_stack.numel() == K * 256 * 4 * 256,
+              "w_hh_t_stack must be [K, 256, 4*256]");
+  auto h_seq = torch::empty({R, T, 256}, xp.options());
+  auto gates = torch::empty({R, T, 4 * 256}, xp.options());
+  auto c_seq = torch::empty({R, T, 256}, xp.options());
+  launch_lstm_seq_fwd_b(xp.data_ptr<float>(),
+                        w_hh_t_stack.data_ptr<float>(),
+                        h_seq.data_ptr<float>(), gates.data_ptr<float>(),
+                        c_seq.data_ptr<float>(), (int)R,
+                        (int)rows_per_client, (int)T, cur_stream());
+  return {h_seq, gates, c_seq};
+}
+
+torch::Tensor lstm_seq_bwd_b(torch::Tensor gates, torch::Tensor c_seq,
+                             torch::Tensor w_hh_stack,
+                             torch::Tensor dh_out,
+                             int64_t rows_per_client) {
+  check_flat(gates, "gates"); check_flat(c_seq, "c_seq");
+  check_flat(w_hh_stack, "w_hh_stack"); check_flat(dh_out, "dh_out");
+  long long R = gates.size(0), T = gates.size(1);
+  TORCH_CHECK(rows_per_client > 0 && R % rows_per_client == 0);
+  long long K = R / rows_per_client;
+  TORCH_CHECK(w_hh_stack.numel() == K * 4 * 256 * 256,
+              "w_hh_stack must be [K, 4*256, 256]");
+  auto dg = torch::empty_like(gates);
+  launch_lstm_seq_bwd_b(gates.data_ptr<float>(), c_seq.data_ptr<float>(),
+                        w_hh_stack.data_ptr<float>(),
+                        dh_out.data_ptr<float>(), dg.data_ptr<float>(),
+                        (int)R, (int)rows_per_client, (int)T,
+                        cur_stream());
+  return dg;
+}
+
 // fused GRU sequence recurrence (gru_seq.hip)
 std::vector<torch::Tensor> gru_seq_fwd(torch::Tensor gi, torch::Tensor whh_t,
                                        torch::Tensor b_hh) {
@@ -779,8 +837,51 @@ void cnn_round_mega(torch::Tensor shard_x, torch::Tensor shard_y,
       cur_stream(), use_bf16 ? 1 : 0);
 }
 
+// per-client clip + sufficient stats + SGD over a K-stacked flat arena
+// (graph-capture safe; stats_out[2k] accumulates clipped Σg / Σg²)
+void mega_clip_sgd(torch::Tensor params_stack, torch::Tensor grads_stack,
+                   int64_t K, torch::Tensor acc2k, double max_norm,
+                   torch::Tensor lr_t, torch::Tensor stats_out) {
+  check_flat(params_stack, "params_stack");
+  check_flat(grads_stack, "grads_stack");
+  check_flat(lr_t, "lr_t"); check_flat(stats_out, "stats_out");
+  TORCH_CHECK(K > 0 && params_stack.numel() % K == 0 &&
+              params_stack.numel() == grads_stack.numel());
+  TORCH_CHECK(acc2k.is_cuda() && acc2k.scalar_type() == torch::kFloat64 &&
+              acc2k.numel() >= 2 * K);
+  TORCH_CHECK(stats_out.numel() >= 2 * K);
+  long long P = params_stack.numel() / K;
+  launch_mega_clip_sgd(params_stack.data_ptr<float>(),
+                       grads_stack.data_ptr<float>(), P, (int)K,
+                       acc2k.data_ptr<double>(), (float)max_norm,
+                       lr_t.data_ptr<float>(), stats_out.data_ptr<float>(),
+                       cur_stream());
+}
+
+void mega_pseudo_accum(torch::Tensor grads_stack, torch::Tensor server,
+                       torch::Tensor params_stack, torch::Tensor weights_dev,
+                       torch::Tensor round_accum) {
+  check_flat(grads_stack, "grads_stack"); check_flat(server, "server");
+  check_flat(params_stack, "params_stack");
+  check_flat(weights_dev, "weights_dev");
+  check_flat(round_accum, "round_accum");
+  long long P = server.numel();
+  TORCH_CHECK(P > 0 && params_stack.numel() % P == 0);
+  int K = (int)(params_stack.numel() / P);
+  TORCH_CHECK(weights_dev.numel() >= K && round_accum.numel() == P &&
+              grads_stack.numel() == params_stack.numel());
+  launch_mega_pseudo_accum(grads_stack.data_ptr<float>(),
+                           server.data_ptr<float>(),
+                           params_stack.data_ptr<float>(),
+                           weights_dev.data_ptr<float>(),
+                           round_accum.data_ptr<float>(), P, K,
+                           cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cnn_round_mega", &cnn_round_mega);
+  m.def("mega_clip_sgd", &mega_clip_sgd);
+  m.def("mega_pseudo_accum", &mega_pseudo_accum);
   m.def("dbg_mfma_bf16_probe", &dbg_mfma_bf16_probe);
   m.def("dbg_conv2_fwd_mfma_bf16", &dbg_conv2_fwd_mfma_bf16);
   m.def("dbg_conv2_bwd_x_mfma_bf16", &dbg_conv2_bwd_x_mfma_bf16);
@@ -810,6 +911,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cnn_epoch", &cnn_epoch);
   m.def("cnn_round", &cnn_round);
   m.def("lstm_seq_fwd", &lstm_seq_fwd);
+  m.def("lstm_seq_fwd_b", &lstm_seq_fwd_b);
+  m.def("lstm_seq_bwd_b", &lstm_seq_bwd_b);
   m.def("gru_seq_fwd", &gru_seq_fwd);
   m.def("gru_seq_bwd", &gru_seq_bwd);
   m.def("lstm_seq_bwd", &lstm_seq_bwd);

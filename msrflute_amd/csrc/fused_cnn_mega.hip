@@ -1200,3 +1200,41 @@ extern "C" void launch_cnn_round_mega(
   hipLaunchKernelGGL(k_accum_mb, dim3(gp), dim3(FBLK), 0, s,
                      round_accum, grads_stack, P, K);
 }
+
+
+// ---------------------------------------------------------------------------
+// standalone entries for K-stacked flat arenas (reused by the
+// Shakespeare mega round's graph-captured epoch): per-client clip +
+// stats + SGD, and weighted pseudo-grad + deterministic accumulate.
+// ---------------------------------------------------------------------------
+extern "C" void launch_mega_clip_sgd(float* params_stack, float* grads_stack,
+                                     long long P, int K, double* acc2k,
+                                     float max_norm, const float* lr_t,
+                                     float* stats_out, hipStream_t s) {
+  hipMemsetAsync(acc2k, 0, 2 * K * sizeof(double), s);
+  int blocks_per_k = (int)((P + SUMSQ_CHUNK - 1) / SUMSQ_CHUNK);
+  hipLaunchKernelGGL(k_sumsq_mb, dim3(K * blocks_per_k), dim3(FBLK), 0, s,
+                     grads_stack, P, blocks_per_k, acc2k);
+  hipLaunchKernelGGL(k_stats_mb, dim3((K + 63) / 64), dim3(64), 0, s,
+                     acc2k, K, max_norm, 1e-6f, stats_out);
+  long long kp = (long long)K * P;
+  int gkp = (int)((kp + FBLK - 1) / FBLK); if (gkp > 4096) gkp = 4096;
+  hipLaunchKernelGGL(k_clip_sgd_mb, dim3(gkp), dim3(FBLK), 0, s,
+                     params_stack, grads_stack, P, K, acc2k, max_norm,
+                     1e-6f, lr_t, stats_out);
+}
+
+extern "C" void launch_mega_pseudo_accum(float* grads_stack,
+                                         const float* server,
+                                         const float* params_stack,
+                                         const float* weights_dev,
+                                         float* round_accum, long long P,
+                                         int K, hipStream_t s) {
+  long long kp = (long long)K * P;
+  int gkp = (int)((kp + FBLK - 1) / FBLK); if (gkp > 4096) gkp = 4096;
+  int gp = (int)((P + FBLK - 1) / FBLK); if (gp > 2048) gp = 2048;
+  hipLaunchKernelGGL(k_pseudo_grad_mb, dim3(gkp), dim3(FBLK), 0, s,
+                     grads_stack, server, params_stack, weights_dev, P, K);
+  hipLaunchKernelGGL(k_accum_mb, dim3(gp), dim3(FBLK), 0, s,
+                     round_accum, grads_stack, P, K);
+}

@@ -160,3 +160,133 @@ void launch_lstm_seq_bwd(const float* gates, const float* c_seq,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// cross-client batched variants (Shakespeare mega round): the grid is
+// ALL clients' rows (row r belongs to client r / rows_per_client) and
+// each block indexes its client's weights in the K-stacked buffers —
+// one launch covers the whole cohort's recurrence instead of one
+// kernel per client on co-scheduling-limited streams.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024)
+void k_lstm_seq_fwd_b(const float* __restrict__ xp,
+                      const float* __restrict__ w_hh_t_stack,
+                      float* __restrict__ h_seq,
+                      float* __restrict__ gates,
+                      float* __restrict__ c_seq, int rows_per_client,
+                      int T) {
+  __shared__ float h_prev[LSTM_H];
+  __shared__ float act_l[4 * LSTM_H];
+  int tid = threadIdx.x;
+  int g = tid >> 8;
+  int b = blockIdx.x;
+  const float* w_hh_t = w_hh_t_stack
+      + (long long)(b / rows_per_client) * LSTM_H * 4 * LSTM_H;
+  if (tid < LSTM_H) h_prev[tid] = 0.f;
+  float c = 0.f;
+  __syncthreads();
+  for (int t = 0; t < T; ++t) {
+    const float* xr = xp + ((long long)b * T + t) * 4 * LSTM_H;
+    float s = xr[tid];
+    #pragma unroll 16
+    for (int k = 0; k < LSTM_H; ++k)
+      s = fmaf(w_hh_t[(long long)k * 4 * LSTM_H + tid], h_prev[k], s);
+    float act = (g == 2) ? tanhf(s) : sigf(s);
+    long long base = ((long long)b * T + t) * 4 * LSTM_H;
+    gates[base + tid] = act;
+    act_l[tid] = act;
+    __syncthreads();
+    if (tid < LSTM_H) {
+      float i = act_l[tid], f = act_l[LSTM_H + tid];
+      float gg = act_l[2 * LSTM_H + tid], o = act_l[3 * LSTM_H + tid];
+      c = f * c + i * gg;
+      float hn = o * tanhf(c);
+      c_seq[((long long)b * T + t) * LSTM_H + tid] = c;
+      h_seq[((long long)b * T + t) * LSTM_H + tid] = hn;
+      h_prev[tid] = hn;
+    }
+    __syncthreads();
+  }
+}
+
+__global__ __launch_bounds__(1024)
+void k_lstm_seq_bwd_b(const float* __restrict__ gates,
+                      const float* __restrict__ c_seq,
+                      const float* __restrict__ w_hh_stack,
+                      const float* __restrict__ dh_out,
+                      float* __restrict__ dg_pre, int rows_per_client,
+                      int T) {
+  __shared__ float dg_l[4 * LSTM_H];
+  __shared__ float part[4 * LSTM_H];
+  __shared__ float dh_rec_l[LSTM_H];
+  int tid = threadIdx.x;
+  int h = tid & (LSTM_H - 1), q = tid >> 8;
+  int b = blockIdx.x;
+  const float* w_hh = w_hh_stack
+      + (long long)(b / rows_per_client) * 4 * LSTM_H * LSTM_H;
+  float dc = 0.f;
+  if (tid < LSTM_H) dh_rec_l[tid] = 0.f;
+  __syncthreads();
+  for (int t = T - 1; t >= 0; --t) {
+    long long base = ((long long)b * T + t) * 4 * LSTM_H;
+    long long cbase = ((long long)b * T + t) * LSTM_H;
+    if (tid < LSTM_H) {
+      float i = gates[base + tid];
+      float f = gates[base + LSTM_H + tid];
+      float g = gates[base + 2 * LSTM_H + tid];
+      float o = gates[base + 3 * LSTM_H + tid];
+      float ct = c_seq[cbase + tid];
+      float cprev = (t > 0) ? c_seq[cbase - LSTM_H + tid] : 0.f;
+      float tc = tanhf(ct);
+      float dh = dh_out[cbase + tid] + dh_rec_l[tid];
+      float do_ = dh * tc * o * (1.f - o);
+      dc = dc + dh * o * (1.f - tc * tc);
+      float di = dc * g * i * (1.f - i);
+      float df = dc * cprev * f * (1.f - f);
+      float dg = dc * i * (1.f - g * g);
+      dc = dc * f;
+      dg_l[tid] = di;
+      dg_l[LSTM_H + tid] = df;
+      dg_l[2 * LSTM_H + tid] = dg;
+      dg_l[3 * LSTM_H + tid] = do_;
+      dg_pre[base + tid] = di;
+      dg_pre[base + LSTM_H + tid] = df;
+      dg_pre[base + 2 * LSTM_H + tid] = dg;
+      dg_pre[base + 3 * LSTM_H + tid] = do_;
+    }
+    __syncthreads();
+    float s = 0.f;
+    #pragma unroll 16
+    for (int jj = 0; jj < LSTM_H; ++jj) {
+      int j = q * LSTM_H + jj;
+      s = fmaf(w_hh[(long long)j * LSTM_H + h], dg_l[j], s);
+    }
+    part[tid] = s;
+    __syncthreads();
+    if (tid < LSTM_H)
+      dh_rec_l[tid] = part[tid] + part[LSTM_H + tid]
+                      + part[2 * LSTM_H + tid] + part[3 * LSTM_H + tid];
+    __syncthreads();
+  }
+}
+
+extern "C" void launch_lstm_seq_fwd_b(const float* xp,
+                                      const float* w_hh_t_stack,
+                                      float* h_seq, float* gates,
+                                      float* c_seq, int R,
+                                      int rows_per_client, int T,
+                                      hipStream_t s) {
+  hipLaunchKernelGGL(k_lstm_seq_fwd_b, dim3(R), dim3(4 * LSTM_H), 0, s,
+                     xp, w_hh_t_stack, h_seq, gates, c_seq,
+                     rows_per_client, T);
+}
+
+extern "C" void launch_lstm_seq_bwd_b(const float* gates, const float* c_seq,
+                                      const float* w_hh_stack,
+                                      const float* dh_out, float* dg_pre,
+                                      int R, int rows_per_client, int T,
+                                      hipStream_t s) {
+  hipLaunchKernelGGL(k_lstm_seq_bwd_b, dim3(R), dim3(4 * LSTM_H), 0, s,
+                     gates, c_seq, w_hh_stack, dh_out, dg_pre,
+                     rows_per_client, T);
+}
