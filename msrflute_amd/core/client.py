@@ -678,6 +678,62 @@ def _fused_round_impl(ex, client_ids, initial_lr, seeds):
     return outputs
 
 
+def _try_mega_lstm_round(self, prim, data_cfg, client_ids, initial_lr,
+                         seeds):
+    """Cross-client MEGA round for the Shakespeare char-LSTM
+    (ops/mega_shakespeare.py): all K clients' epochs in one graph-captured
+    launch set (batched recurrence kernels + bmm projections over
+    K-stacked weights).  Returns outputs or None if not this task /
+    ineligible."""
+    if self._mega_lstm is False:
+        return None
+    if not prim.client_config.get("use_mega_round", True):
+        return None
+    if prim.client_config.get("mixed_precision"):
+        return None  # the LSTM path is fp32
+    if prim.arena is None or prim.arena.device.type != "cuda":
+        return None
+    opt_cfg = prim.client_config.get("optimizer_config", {})
+    if (opt_cfg.get("type", "sgd") != "sgd"
+            or opt_cfg.get("momentum", 0) or opt_cfg.get("weight_decay", 0)
+            or opt_cfg.get("nesterov", False)):
+        return None
+    if self._mega_lstm is None:
+        from ..ops import HAS_EXT
+        from ..ops.mega_shakespeare import (ShakespeareMegaRound,
+                                            matches_char_lstm)
+        if not HAS_EXT or matches_char_lstm(prim.arena) is None:
+            self._mega_lstm = False  # not this model: stop probing
+            return None
+        self._mega_lstm = ShakespeareMegaRound(
+            prim.arena, data_cfg.get("batch_size", 4),
+            data_cfg.get("max_grad_norm"))
+    if not self._mega_lstm.supports(len(client_ids)):
+        return None
+    store = prim._get_shard_store(data_cfg)
+    if store is None or store.x.dim() != 2:
+        return None
+    import msrflute_amd.core.client as cm
+    ds = cm.train_dataset
+    if ds is None:
+        return None
+    dms = data_cfg.get("desired_max_samples", None)
+    for cid in client_ids:
+        i = store.user_pos.get(ds.user_list[cid])
+        if i is None:
+            return None
+        n = store.offsets[i + 1] - store.offsets[i]
+        if n == 0 or (dms is not None and n > dms):
+            return None
+    out = self._mega_lstm.run(store, ds, client_ids, seeds, initial_lr,
+                              self.server_arena, self.round_accums[0])
+    if out is not None:
+        self._streams_dirty = True
+        prim.perf_acc["clients"] = (prim.perf_acc.get("clients", 0)
+                                    + len(client_ids))
+    return out
+
+
 def _pool_run_fused_round_batch(self, client_ids, initial_lr, iteration,
                                 seeds):
     """ClientPool: one _C.cnn_round per executor (on its stream) covering
@@ -686,10 +742,16 @@ def _pool_run_fused_round_batch(self, client_ids, initial_lr, iteration,
     Eligibility is validated UP FRONT so no partial state is mutated on
     the fallback path."""
     prim = self.executors[0]
-    if prim.fused_cnn is None:
-        return None
     data_cfg = prim.client_config["data_config"]["train"]
     if not _fused_round_eligible(prim, data_cfg):
+        return None
+    if getattr(self, "_mega_lstm", "missing") == "missing":
+        self._mega_lstm = None
+    out = _try_mega_lstm_round(self, prim, data_cfg, client_ids,
+                               initial_lr, seeds)
+    if out is not None:
+        return out
+    if prim.fused_cnn is None:
         return None
     store = prim._get_shard_store(data_cfg)
     if store is None or store.x[0].numel() != 784:
