@@ -373,8 +373,14 @@ class Trainer(TrainerBase):
         n_graph_batches = 0
 
         # probe one batch: graphs need dict batches of plain (x, y) tensors
-        # (nlg/mlm/newsrec batch shapes run the generic eager path)
+        # (nlg/mlm/newsrec batch shapes run the generic eager path).
+        # The probe draws through the loader's shuffle, which consumes a
+        # randperm from the per-client-seeded host RNG — restore the state
+        # so every path (eager, graphed, fused round, mega round) trains
+        # on the SAME first-draw shuffle order for a given (round, client)
+        rng_state = torch.get_rng_state()
         probe = next(iter(self.train_dataloader.create_loader()), None)
+        torch.set_rng_state(rng_state)
         if not (isinstance(probe, dict) and torch.is_tensor(probe.get("x"))
                 and torch.is_tensor(probe.get("y"))):
             self.graph_cache = None

@@ -213,6 +213,8 @@ class ShakespeareMegaRound:
         with torch.cuda.graph(graph):
             epoch_body()
         g["graph"] = graph
+        # the capture baked these data pointers in: pin their lifetime
+        g["_pins"] = (x_all, y_all, server)
         torch.cuda.synchronize()
         return g
 
