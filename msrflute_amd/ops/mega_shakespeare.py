@@ -30,6 +30,7 @@ only GEMM scheduling (bmm vs mm) and float accumulation orders differ.
 
 from __future__ import annotations
 
+import os
 import time
 from typing import Dict, Optional, Tuple
 
@@ -128,10 +129,15 @@ class ShakespeareMegaRound:
         bs, V, E = self.bs, self.V, self.E
         R, T = x.shape
         v = self._views(flat)
-        # per-client embedding: offset rows into the stacked table
+        # per-client embedding: offset rows into the stacked table.
+        # index_select, NOT F.embedding: embedding_dense_backward's
+        # implementation is index-value-dependent, which is unsafe under
+        # hipGraph capture (captured on round-1 data, replayed on other
+        # rounds'); index_select's backward is a plain index_add with
+        # value-independent kernels
         offs = (torch.arange(K, device=x.device) * V).repeat_interleave(bs)
-        e = F.embedding((x + offs[:, None]).view(-1),
-                        v["net.embeddings.weight"].reshape(K * V, E))
+        e = v["net.embeddings.weight"].reshape(K * V, E).index_select(
+            0, (x + offs[:, None]).view(-1))
         e = e.view(K, bs * T, E)
         b0 = (v["net.lstm.bias_ih_l0"] + v["net.lstm.bias_hh_l0"])
         xp0 = torch.baddbmm(b0.unsqueeze(1), e,
@@ -201,22 +207,31 @@ class ShakespeareMegaRound:
                                  flat.data.reshape(-1), g["weights"],
                                  g["accum"])
 
+        g["body"] = epoch_body
+        g["graph"] = None
+        # the compute reads these tensors by baked pointer: pin lifetime
+        g["_pins"] = (x_all, y_all, server)
+        return g
+
+    def _capture(self, g):
+        """Warm up + capture the epoch.  MUST run after the round's real
+        idx/ymask/lr are staged: capturing on degenerate data risks
+        value-dependent kernel selection that faults on replay."""
+        if os.environ.get("MEGA_SHK_EAGER") == "1":
+            return
         # warmup (establishes flat.grad + autograd buffers), then capture;
         # nothing here writes the server arena (flat is the working copy)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(2):
-                epoch_body()
+                g["body"]()
         torch.cuda.current_stream().wait_stream(s)
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):
-            epoch_body()
+            g["body"]()
         g["graph"] = graph
-        # the capture baked these data pointers in: pin their lifetime
-        g["_pins"] = (x_all, y_all, server)
         torch.cuda.synchronize()
-        return g
 
     # ------------------------------------------------------------------
     def run(self, store, ds, client_ids, seeds, initial_lr: float,
@@ -265,10 +280,19 @@ class ShakespeareMegaRound:
         g["idx"].copy_(idx)
         g["ymask"].copy_(mask)
         g["weights"].copy_(torch.tensor([float(c) for c in counts]))
+        self.lr_t.fill_(float(initial_lr))
+        if g["graph"] is None and not g.get("_captured"):
+            # first use of this shape: capture on THIS round's real data
+            self._capture(g)
+            g["_captured"] = True
+        # warmup/capture side effects land in these accumulators: zero
+        # them after capture, before the replay that counts
         g["stats_out"].zero_()
         g["accum"].zero_()
-        self.lr_t.fill_(float(initial_lr))
-        g["graph"].replay()
+        if g["graph"] is None:
+            g["body"]()
+        else:
+            g["graph"].replay()
         # fold the epoch's accumulated weighted pseudo-gradients
         round_accum += g["accum"]
 
