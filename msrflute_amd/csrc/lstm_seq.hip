@@ -57,10 +57,21 @@ void k_lstm_seq_fwd(const float* __restrict__ xp,
   __syncthreads();
   for (int t = 0; t < T; ++t) {
     const float* xr = xp + ((long long)b * T + t) * 4 * LSTM_H;
-    float s = xr[tid];
-    #pragma unroll 16
-    for (int k = 0; k < LSTM_H; ++k)
-      s = fmaf(w_hh_t[(long long)k * 4 * LSTM_H + tid], h_prev[k], s);
+    // 4 independent accumulator chains: the single-chain form was
+    // bound by the fmaf dependency latency (256 serial FMAs per step)
+    float s0 = xr[tid], s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    #pragma unroll 8
+    for (int k = 0; k < LSTM_H / 4; ++k) {
+      s0 = fmaf(w_hh_t[(long long)k * 4 * LSTM_H + tid],
+                h_prev[k], s0);
+      s1 = fmaf(w_hh_t[(long long)(64 + k) * 4 * LSTM_H + tid],
+                h_prev[64 + k], s1);
+      s2 = fmaf(w_hh_t[(long long)(128 + k) * 4 * LSTM_H + tid],
+                h_prev[128 + k], s2);
+      s3 = fmaf(w_hh_t[(long long)(192 + k) * 4 * LSTM_H + tid],
+                h_prev[192 + k], s3);
+    }
+    float s = (s0 + s1) + (s2 + s3);
     float act = (g == 2) ? tanhf(s) : sigf(s);
     long long base = ((long long)b * T + t) * 4 * LSTM_H;
     gates[base + tid] = act;       // ACTIVATED i,f,g,o (order ifgo)
@@ -128,12 +139,18 @@ void k_lstm_seq_bwd(const float* __restrict__ gates,
     __syncthreads();
     // dh_rec = W_hh^T dg: thread (q, h) sums its 256-row quarter with
     // lane-coalesced w_hh[j*H + h] reads; 16 waves hide the L2 latency
-    float s = 0.f;
-    #pragma unroll 16
-    for (int jj = 0; jj < LSTM_H; ++jj) {
+    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    #pragma unroll 8
+    for (int jj = 0; jj < LSTM_H / 4; ++jj) {
       int j = q * LSTM_H + jj;
-      s = fmaf(w_hh[(long long)j * LSTM_H + h], dg_l[j], s);
+      s0 = fmaf(w_hh[(long long)j * LSTM_H + h], dg_l[j], s0);
+      s1 = fmaf(w_hh[(long long)(j + 64) * LSTM_H + h], dg_l[j + 64], s1);
+      s2 = fmaf(w_hh[(long long)(j + 128) * LSTM_H + h],
+                dg_l[j + 128], s2);
+      s3 = fmaf(w_hh[(long long)(j + 192) * LSTM_H + h],
+                dg_l[j + 192], s3);
     }
+    float s = (s0 + s1) + (s2 + s3);
     part[tid] = s;
     __syncthreads();
     if (tid < LSTM_H)
@@ -187,10 +204,21 @@ void k_lstm_seq_fwd_b(const float* __restrict__ xp,
   __syncthreads();
   for (int t = 0; t < T; ++t) {
     const float* xr = xp + ((long long)b * T + t) * 4 * LSTM_H;
-    float s = xr[tid];
-    #pragma unroll 16
-    for (int k = 0; k < LSTM_H; ++k)
-      s = fmaf(w_hh_t[(long long)k * 4 * LSTM_H + tid], h_prev[k], s);
+    // 4 independent accumulator chains: the single-chain form was
+    // bound by the fmaf dependency latency (256 serial FMAs per step)
+    float s0 = xr[tid], s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    #pragma unroll 8
+    for (int k = 0; k < LSTM_H / 4; ++k) {
+      s0 = fmaf(w_hh_t[(long long)k * 4 * LSTM_H + tid],
+                h_prev[k], s0);
+      s1 = fmaf(w_hh_t[(long long)(64 + k) * 4 * LSTM_H + tid],
+                h_prev[64 + k], s1);
+      s2 = fmaf(w_hh_t[(long long)(128 + k) * 4 * LSTM_H + tid],
+                h_prev[128 + k], s2);
+      s3 = fmaf(w_hh_t[(long long)(192 + k) * 4 * LSTM_H + tid],
+                h_prev[192 + k], s3);
+    }
+    float s = (s0 + s1) + (s2 + s3);
     float act = (g == 2) ? tanhf(s) : sigf(s);
     long long base = ((long long)b * T + t) * 4 * LSTM_H;
     gates[base + tid] = act;
@@ -255,12 +283,18 @@ void k_lstm_seq_bwd_b(const float* __restrict__ gates,
       dg_pre[base + 3 * LSTM_H + tid] = do_;
     }
     __syncthreads();
-    float s = 0.f;
-    #pragma unroll 16
-    for (int jj = 0; jj < LSTM_H; ++jj) {
+    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    #pragma unroll 8
+    for (int jj = 0; jj < LSTM_H / 4; ++jj) {
       int j = q * LSTM_H + jj;
-      s = fmaf(w_hh[(long long)j * LSTM_H + h], dg_l[j], s);
+      s0 = fmaf(w_hh[(long long)j * LSTM_H + h], dg_l[j], s0);
+      s1 = fmaf(w_hh[(long long)(j + 64) * LSTM_H + h], dg_l[j + 64], s1);
+      s2 = fmaf(w_hh[(long long)(j + 128) * LSTM_H + h],
+                dg_l[j + 128], s2);
+      s3 = fmaf(w_hh[(long long)(j + 192) * LSTM_H + h],
+                dg_l[j + 192], s3);
     }
+    float s = (s0 + s1) + (s2 + s3);
     part[tid] = s;
     __syncthreads();
     if (tid < LSTM_H)
@@ -270,170 +304,12 @@ void k_lstm_seq_bwd_b(const float* __restrict__ gates,
   }
 }
 
-// ---------------------------------------------------------------------------
-// client-major variants (rows_per_client == 4, the benchmark batch):
-// ONE block per CLIENT processes its 4 rows together, so every W_hh
-// element loaded from L2/HBM is reused 4x in registers.  The row-major
-// _b kernels above re-read the client's 1 MB weight matrix per row per
-// timestep (R blocks x T x 1 MB), which rocprof showed as the bench's
-// bandwidth bound (~0.9 ms per call); here traffic drops 4x and each
-// client's weights stay hot in its XCD's L2 across timesteps.
-// ---------------------------------------------------------------------------
-__global__ __launch_bounds__(1024)
-void k_lstm_seq_fwd_c(const float* __restrict__ xp,
-                      const float* __restrict__ w_hh_t_stack,
-                      float* __restrict__ h_seq,
-                      float* __restrict__ gates,
-                      float* __restrict__ c_seq, int T) {
-  __shared__ float h_prev[4][LSTM_H];
-  __shared__ float act_l[4][4 * LSTM_H];
-  int tid = threadIdx.x;              // gate column j = g*H + h
-  int g = tid >> 8;
-  long long r0 = (long long)blockIdx.x * 4;  // first row of this client
-  const float* w_hh_t = w_hh_t_stack
-      + (long long)blockIdx.x * LSTM_H * 4 * LSTM_H;
-  if (tid < LSTM_H)
-    for (int r = 0; r < 4; ++r) h_prev[r][tid] = 0.f;
-  float c0 = 0.f, c1 = 0.f, c2 = 0.f, c3 = 0.f;
-  __syncthreads();
-  for (int t = 0; t < T; ++t) {
-    long long xb = (r0 * T + t) * 4 * LSTM_H;
-    long long rstride = (long long)T * 4 * LSTM_H;
-    float s0 = xp[xb + tid];
-    float s1 = xp[xb + rstride + tid];
-    float s2 = xp[xb + 2 * rstride + tid];
-    float s3 = xp[xb + 3 * rstride + tid];
-    #pragma unroll 8
-    for (int k = 0; k < LSTM_H; ++k) {
-      float wv = w_hh_t[(long long)k * 4 * LSTM_H + tid];
-      s0 = fmaf(wv, h_prev[0][k], s0);
-      s1 = fmaf(wv, h_prev[1][k], s1);
-      s2 = fmaf(wv, h_prev[2][k], s2);
-      s3 = fmaf(wv, h_prev[3][k], s3);
-    }
-    float a0 = (g == 2) ? tanhf(s0) : sigf(s0);
-    float a1 = (g == 2) ? tanhf(s1) : sigf(s1);
-    float a2 = (g == 2) ? tanhf(s2) : sigf(s2);
-    float a3 = (g == 2) ? tanhf(s3) : sigf(s3);
-    gates[xb + tid] = a0;
-    gates[xb + rstride + tid] = a1;
-    gates[xb + 2 * rstride + tid] = a2;
-    gates[xb + 3 * rstride + tid] = a3;
-    act_l[0][tid] = a0; act_l[1][tid] = a1;
-    act_l[2][tid] = a2; act_l[3][tid] = a3;
-    __syncthreads();
-    if (tid < LSTM_H) {
-      long long cb = (r0 * T + t) * LSTM_H;
-      long long crs = (long long)T * LSTM_H;
-      #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float i = act_l[r][tid], f = act_l[r][LSTM_H + tid];
-        float gg = act_l[r][2 * LSTM_H + tid];
-        float o = act_l[r][3 * LSTM_H + tid];
-        float* cr = (r == 0) ? &c0 : (r == 1) ? &c1 : (r == 2) ? &c2 : &c3;
-        *cr = f * *cr + i * gg;
-        float hn = o * tanhf(*cr);
-        c_seq[cb + r * crs + tid] = *cr;
-        h_seq[cb + r * crs + tid] = hn;
-        h_prev[r][tid] = hn;
-      }
-    }
-    __syncthreads();
-  }
-}
-
-__global__ __launch_bounds__(1024)
-void k_lstm_seq_bwd_c(const float* __restrict__ gates,
-                      const float* __restrict__ c_seq,
-                      const float* __restrict__ w_hh_stack,
-                      const float* __restrict__ dh_out,
-                      float* __restrict__ dg_pre, int T) {
-  __shared__ float dg_l[4][4 * LSTM_H];
-  __shared__ float part[4 * LSTM_H];
-  __shared__ float dh_rec_l[4][LSTM_H];
-  int tid = threadIdx.x;
-  int h = tid & (LSTM_H - 1), q = tid >> 8;
-  long long r0 = (long long)blockIdx.x * 4;
-  const float* w_hh = w_hh_stack
-      + (long long)blockIdx.x * 4 * LSTM_H * LSTM_H;
-  float dc0 = 0.f, dc1 = 0.f, dc2 = 0.f, dc3 = 0.f;
-  if (tid < LSTM_H)
-    for (int r = 0; r < 4; ++r) dh_rec_l[r][tid] = 0.f;
-  __syncthreads();
-  for (int t = T - 1; t >= 0; --t) {
-    long long rstride4 = (long long)T * 4 * LSTM_H;
-    long long rstride1 = (long long)T * LSTM_H;
-    long long base = (r0 * T + t) * 4 * LSTM_H;
-    long long cbase = (r0 * T + t) * LSTM_H;
-    if (tid < LSTM_H) {
-      #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        long long b4 = base + r * rstride4;
-        long long c1b = cbase + r * rstride1;
-        float i = gates[b4 + tid];
-        float f = gates[b4 + LSTM_H + tid];
-        float g = gates[b4 + 2 * LSTM_H + tid];
-        float o = gates[b4 + 3 * LSTM_H + tid];
-        float ct = c_seq[c1b + tid];
-        float cprev = (t > 0) ? c_seq[c1b - LSTM_H + tid] : 0.f;
-        float tc = tanhf(ct);
-        float dh = dh_out[c1b + tid] + dh_rec_l[r][tid];
-        float* dcr = (r == 0) ? &dc0 : (r == 1) ? &dc1
-                     : (r == 2) ? &dc2 : &dc3;
-        float do_ = dh * tc * o * (1.f - o);
-        float dc = *dcr + dh * o * (1.f - tc * tc);
-        float di = dc * g * i * (1.f - i);
-        float df = dc * cprev * f * (1.f - f);
-        float dg = dc * i * (1.f - g * g);
-        *dcr = dc * f;
-        dg_l[r][tid] = di;
-        dg_l[r][LSTM_H + tid] = df;
-        dg_l[r][2 * LSTM_H + tid] = dg;
-        dg_l[r][3 * LSTM_H + tid] = do_;
-        dg_pre[b4 + tid] = di;
-        dg_pre[b4 + LSTM_H + tid] = df;
-        dg_pre[b4 + 2 * LSTM_H + tid] = dg;
-        dg_pre[b4 + 3 * LSTM_H + tid] = do_;
-      }
-    }
-    __syncthreads();
-    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    #pragma unroll 8
-    for (int jj = 0; jj < LSTM_H; ++jj) {
-      int j = q * LSTM_H + jj;
-      float wv = w_hh[(long long)j * LSTM_H + h];
-      s0 = fmaf(wv, dg_l[0][j], s0);
-      s1 = fmaf(wv, dg_l[1][j], s1);
-      s2 = fmaf(wv, dg_l[2][j], s2);
-      s3 = fmaf(wv, dg_l[3][j], s3);
-    }
-    // reduce the 4 gate-quarter partials per row sequentially through
-    // one LDS plane (saves 12 KB over a [4][4H] partial buffer)
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float s = (r == 0) ? s0 : (r == 1) ? s1 : (r == 2) ? s2 : s3;
-      part[tid] = s;
-      __syncthreads();
-      if (tid < LSTM_H)
-        dh_rec_l[r][tid] = part[tid] + part[LSTM_H + tid]
-                           + part[2 * LSTM_H + tid]
-                           + part[3 * LSTM_H + tid];
-      __syncthreads();
-    }
-  }
-}
-
 extern "C" void launch_lstm_seq_fwd_b(const float* xp,
                                       const float* w_hh_t_stack,
                                       float* h_seq, float* gates,
                                       float* c_seq, int R,
                                       int rows_per_client, int T,
                                       hipStream_t s) {
-  if (rows_per_client == 4) {
-    hipLaunchKernelGGL(k_lstm_seq_fwd_c, dim3(R / 4), dim3(4 * LSTM_H), 0,
-                       s, xp, w_hh_t_stack, h_seq, gates, c_seq, T);
-    return;
-  }
   hipLaunchKernelGGL(k_lstm_seq_fwd_b, dim3(R), dim3(4 * LSTM_H), 0, s,
                      xp, w_hh_t_stack, h_seq, gates, c_seq,
                      rows_per_client, T);
@@ -444,11 +320,6 @@ extern "C" void launch_lstm_seq_bwd_b(const float* gates, const float* c_seq,
                                       const float* dh_out, float* dg_pre,
                                       int R, int rows_per_client, int T,
                                       hipStream_t s) {
-  if (rows_per_client == 4) {
-    hipLaunchKernelGGL(k_lstm_seq_bwd_c, dim3(R / 4), dim3(4 * LSTM_H), 0,
-                       s, gates, c_seq, w_hh_stack, dh_out, dg_pre, T);
-    return;
-  }
   hipLaunchKernelGGL(k_lstm_seq_bwd_b, dim3(R), dim3(4 * LSTM_H), 0, s,
                      gates, c_seq, w_hh_stack, dh_out, dg_pre,
                      rows_per_client, T);
