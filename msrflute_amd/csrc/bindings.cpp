@@ -458,8 +458,9 @@ std::vector<torch::Tensor> lstm_seq_fwd(torch::Tensor xp,
                                         torch::Tensor w_hh) {
   check_flat(xp, "xp"); check_flat(w_hh, "w_hh");
   TORCH_CHECK(xp.dim() == 3 && xp.size(2) == 4 * 256 &&
-              w_hh.size(0) == 256 && w_hh.size(1) == 4 * 256,
-              "fused LSTM supports hidden size 256; pass W_hh TRANSPOSED");
+              w_hh.numel() == 256 * 4 * 256,
+              "fused LSTM supports hidden size 256; pass W_hh "
+              "float4-packed [H/4, 4H, 4] (ops/lstm._pack_fwd)");
   long long B = xp.size(0), T = xp.size(1);
   auto h_seq = torch::empty({B, T, 256}, xp.options());
   auto gates = torch::empty({B, T, 4 * 256}, xp.options());
@@ -475,8 +476,9 @@ torch::Tensor lstm_seq_bwd(torch::Tensor gates, torch::Tensor c_seq,
                            torch::Tensor w_hh, torch::Tensor dh_out) {
   check_flat(gates, "gates"); check_flat(c_seq, "c_seq");
   check_flat(w_hh, "w_hh"); check_flat(dh_out, "dh_out");
-  TORCH_CHECK(w_hh.size(0) == 4 * 256 && w_hh.size(1) == 256,
-              "lstm_seq_bwd wants W_hh [4H,H]");
+  TORCH_CHECK(w_hh.numel() == 4 * 256 * 256,
+              "lstm_seq_bwd wants W_hh float4-packed [4H/4, H, 4] "
+              "(ops/lstm._pack_bwd)");
   long long B = gates.size(0), T = gates.size(1);
   auto dg = torch::empty_like(gates);
   launch_lstm_seq_bwd(gates.data_ptr<float>(), c_seq.data_ptr<float>(),

@@ -52,24 +52,25 @@ void k_lstm_seq_fwd(const float* __restrict__ xp,
   int tid = threadIdx.x;           // gate column j = g*H + h
   int g = tid >> 8;
   int b = blockIdx.x;
+  const float4* wp = reinterpret_cast<const float4*>(w_hh_t);
+  const float4* hp4 = reinterpret_cast<const float4*>(h_prev);
   if (tid < LSTM_H) h_prev[tid] = 0.f;
   float c = 0.f;                   // live in threads tid < H only
   __syncthreads();
   for (int t = 0; t < T; ++t) {
     const float* xr = xp + ((long long)b * T + t) * 4 * LSTM_H;
-    // 4 independent accumulator chains: the single-chain form was
-    // bound by the fmaf dependency latency (256 serial FMAs per step)
+    // float4-packed weights ([H/4, 4H, 4]: one b128 load covers 4 k's)
+    // + 4 independent accumulator chains: the b32 form was bound by
+    // exposed L2 latency x load count (256 loads/step/thread -> 64)
     float s0 = xr[tid], s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    #pragma unroll 8
-    for (int k = 0; k < LSTM_H / 4; ++k) {
-      s0 = fmaf(w_hh_t[(long long)k * 4 * LSTM_H + tid],
-                h_prev[k], s0);
-      s1 = fmaf(w_hh_t[(long long)(64 + k) * 4 * LSTM_H + tid],
-                h_prev[64 + k], s1);
-      s2 = fmaf(w_hh_t[(long long)(128 + k) * 4 * LSTM_H + tid],
-                h_prev[128 + k], s2);
-      s3 = fmaf(w_hh_t[(long long)(192 + k) * 4 * LSTM_H + tid],
-                h_prev[192 + k], s3);
+    #pragma unroll 16
+    for (int kk = 0; kk < LSTM_H / 4; ++kk) {
+      float4 wv = wp[(long long)kk * 4 * LSTM_H + tid];
+      float4 hv = hp4[kk];
+      s0 = fmaf(wv.x, hv.x, s0);
+      s1 = fmaf(wv.y, hv.y, s1);
+      s2 = fmaf(wv.z, hv.z, s2);
+      s3 = fmaf(wv.w, hv.w, s3);
     }
     float s = (s0 + s1) + (s2 + s3);
     float act = (g == 2) ? tanhf(s) : sigf(s);
@@ -106,6 +107,8 @@ void k_lstm_seq_bwd(const float* __restrict__ gates,
   int tid = threadIdx.x;
   int h = tid & (LSTM_H - 1), q = tid >> 8;  // partial-dot quarter
   int b = blockIdx.x;
+  const float4* wpB = reinterpret_cast<const float4*>(w_hh);
+  const float4* dgp4 = reinterpret_cast<const float4*>(dg_l);
   float dc = 0.f;                            // live in threads tid < H
   if (tid < LSTM_H) dh_rec_l[tid] = 0.f;
   __syncthreads();
@@ -140,15 +143,15 @@ void k_lstm_seq_bwd(const float* __restrict__ gates,
     // dh_rec = W_hh^T dg: thread (q, h) sums its 256-row quarter with
     // lane-coalesced w_hh[j*H + h] reads; 16 waves hide the L2 latency
     float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    #pragma unroll 8
+    #pragma unroll 16
     for (int jj = 0; jj < LSTM_H / 4; ++jj) {
-      int j = q * LSTM_H + jj;
-      s0 = fmaf(w_hh[(long long)j * LSTM_H + h], dg_l[j], s0);
-      s1 = fmaf(w_hh[(long long)(j + 64) * LSTM_H + h], dg_l[j + 64], s1);
-      s2 = fmaf(w_hh[(long long)(j + 128) * LSTM_H + h],
-                dg_l[j + 128], s2);
-      s3 = fmaf(w_hh[(long long)(j + 192) * LSTM_H + h],
-                dg_l[j + 192], s3);
+      int jg = q * (LSTM_H / 4) + jj;    // float4 group of 4 j's
+      float4 wv = wpB[(long long)jg * LSTM_H + h];
+      float4 dv = dgp4[jg];
+      s0 = fmaf(wv.x, dv.x, s0);
+      s1 = fmaf(wv.y, dv.y, s1);
+      s2 = fmaf(wv.z, dv.z, s2);
+      s3 = fmaf(wv.w, dv.w, s3);
     }
     float s = (s0 + s1) + (s2 + s3);
     part[tid] = s;
@@ -199,24 +202,25 @@ void k_lstm_seq_fwd_b(const float* __restrict__ xp,
   int b = blockIdx.x;
   const float* w_hh_t = w_hh_t_stack
       + (long long)(b / rows_per_client) * LSTM_H * 4 * LSTM_H;
+  const float4* wp = reinterpret_cast<const float4*>(w_hh_t);
+  const float4* hp4 = reinterpret_cast<const float4*>(h_prev);
   if (tid < LSTM_H) h_prev[tid] = 0.f;
   float c = 0.f;
   __syncthreads();
   for (int t = 0; t < T; ++t) {
     const float* xr = xp + ((long long)b * T + t) * 4 * LSTM_H;
-    // 4 independent accumulator chains: the single-chain form was
-    // bound by the fmaf dependency latency (256 serial FMAs per step)
+    // float4-packed weights ([H/4, 4H, 4]: one b128 load covers 4 k's)
+    // + 4 independent accumulator chains: the b32 form was bound by
+    // exposed L2 latency x load count (256 loads/step/thread -> 64)
     float s0 = xr[tid], s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    #pragma unroll 8
-    for (int k = 0; k < LSTM_H / 4; ++k) {
-      s0 = fmaf(w_hh_t[(long long)k * 4 * LSTM_H + tid],
-                h_prev[k], s0);
-      s1 = fmaf(w_hh_t[(long long)(64 + k) * 4 * LSTM_H + tid],
-                h_prev[64 + k], s1);
-      s2 = fmaf(w_hh_t[(long long)(128 + k) * 4 * LSTM_H + tid],
-                h_prev[128 + k], s2);
-      s3 = fmaf(w_hh_t[(long long)(192 + k) * 4 * LSTM_H + tid],
-                h_prev[192 + k], s3);
+    #pragma unroll 16
+    for (int kk = 0; kk < LSTM_H / 4; ++kk) {
+      float4 wv = wp[(long long)kk * 4 * LSTM_H + tid];
+      float4 hv = hp4[kk];
+      s0 = fmaf(wv.x, hv.x, s0);
+      s1 = fmaf(wv.y, hv.y, s1);
+      s2 = fmaf(wv.z, hv.z, s2);
+      s3 = fmaf(wv.w, hv.w, s3);
     }
     float s = (s0 + s1) + (s2 + s3);
     float act = (g == 2) ? tanhf(s) : sigf(s);
@@ -252,6 +256,8 @@ void k_lstm_seq_bwd_b(const float* __restrict__ gates,
   int b = blockIdx.x;
   const float* w_hh = w_hh_stack
       + (long long)(b / rows_per_client) * 4 * LSTM_H * LSTM_H;
+  const float4* wpB = reinterpret_cast<const float4*>(w_hh);
+  const float4* dgp4 = reinterpret_cast<const float4*>(dg_l);
   float dc = 0.f;
   if (tid < LSTM_H) dh_rec_l[tid] = 0.f;
   __syncthreads();
@@ -284,15 +290,15 @@ void k_lstm_seq_bwd_b(const float* __restrict__ gates,
     }
     __syncthreads();
     float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    #pragma unroll 8
+    #pragma unroll 16
     for (int jj = 0; jj < LSTM_H / 4; ++jj) {
-      int j = q * LSTM_H + jj;
-      s0 = fmaf(w_hh[(long long)j * LSTM_H + h], dg_l[j], s0);
-      s1 = fmaf(w_hh[(long long)(j + 64) * LSTM_H + h], dg_l[j + 64], s1);
-      s2 = fmaf(w_hh[(long long)(j + 128) * LSTM_H + h],
-                dg_l[j + 128], s2);
-      s3 = fmaf(w_hh[(long long)(j + 192) * LSTM_H + h],
-                dg_l[j + 192], s3);
+      int jg = q * (LSTM_H / 4) + jj;    // float4 group of 4 j's
+      float4 wv = wpB[(long long)jg * LSTM_H + h];
+      float4 dv = dgp4[jg];
+      s0 = fmaf(wv.x, dv.x, s0);
+      s1 = fmaf(wv.y, dv.y, s1);
+      s2 = fmaf(wv.z, dv.z, s2);
+      s3 = fmaf(wv.w, dv.w, s3);
     }
     float s = (s0 + s1) + (s2 + s3);
     part[tid] = s;

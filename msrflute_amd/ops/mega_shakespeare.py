@@ -71,9 +71,11 @@ class _LSTMSeqB(torch.autograd.Function):
     @staticmethod
     def forward(ctx, xp, w_hh_s, rows_per_client):
         K = w_hh_s.shape[0]
-        w_t = w_hh_s.transpose(1, 2).contiguous()
+        # per-client float4 packing (ops/lstm._pack_fwd, K-stacked)
+        w_p = w_hh_s.transpose(1, 2).reshape(K, H // 4, 4, 4 * H) \
+            .permute(0, 1, 3, 2).contiguous()
         h_seq, gates, c_seq = _C.lstm_seq_fwd_b(
-            xp.contiguous(), w_t.reshape(-1), rows_per_client)
+            xp.contiguous(), w_p.reshape(-1), rows_per_client)
         ctx.save_for_backward(gates, c_seq, w_hh_s, h_seq)
         ctx.rows_per_client = rows_per_client
         return h_seq
@@ -83,8 +85,8 @@ class _LSTMSeqB(torch.autograd.Function):
         gates, c_seq, w_hh_s, h_seq = ctx.saved_tensors
         rpc = ctx.rows_per_client
         K = w_hh_s.shape[0]
-        dg = _C.lstm_seq_bwd_b(gates, c_seq,
-                               w_hh_s.contiguous().reshape(-1),
+        w_pb = w_hh_s.reshape(K, H, 4, H).permute(0, 1, 3, 2).contiguous()
+        dg = _C.lstm_seq_bwd_b(gates, c_seq, w_pb.reshape(-1),
                                dh.contiguous(), rpc)
         R, T, _ = h_seq.shape
         h_prev = torch.cat([h_seq.new_zeros(R, 1, H), h_seq[:, :-1]], dim=1)

@@ -42,13 +42,14 @@ xp = torch.randn(B, T, 4*H, device="cuda")
 whh = torch.randn(4*H, H, device="cuda")
 torch.cuda.synchronize(); t0=time.time()
 for _ in range(N):
-    h_seq, gates, c_seq = _C.lstm_seq_fwd(xp, whh.t().contiguous())
+    from msrflute_amd.ops.lstm import _pack_fwd, _pack_bwd
+    h_seq, gates, c_seq = _C.lstm_seq_fwd(xp, _pack_fwd(whh))
 torch.cuda.synchronize()
 print(f"lstm_seq_fwd per call: {(time.time()-t0)/N*1000:.3f} ms (T={T})")
 dh = torch.randn_like(h_seq)
 torch.cuda.synchronize(); t0=time.time()
 for _ in range(N):
-    dg = _C.lstm_seq_bwd(gates, c_seq, whh, dh)
+    dg = _C.lstm_seq_bwd(gates, c_seq, _pack_bwd(whh), dh)
 torch.cuda.synchronize()
 print(f"lstm_seq_bwd per call: {(time.time()-t0)/N*1000:.3f} ms")
 
