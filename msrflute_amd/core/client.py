@@ -734,6 +734,61 @@ def _try_mega_lstm_round(self, prim, data_cfg, client_ids, initial_lr,
     return out
 
 
+def _try_mega_resnet_round(self, prim, data_cfg, client_ids, initial_lr,
+                           seeds):
+    """Cross-client MEGA round for the fed-CIFAR100 ResNet-18
+    (ops/mega_resnet.py): all K clients' epochs as one graph-captured
+    launch set of grouped convs + GroupNorm over K-stacked weights."""
+    if self._mega_resnet is False:
+        return None
+    if not prim.client_config.get("use_mega_round", True):
+        return None
+    if prim.client_config.get("mixed_precision"):
+        return None
+    if prim.arena is None or prim.arena.device.type != "cuda":
+        return None
+    opt_cfg = prim.client_config.get("optimizer_config", {})
+    if (opt_cfg.get("type", "sgd") != "sgd"
+            or opt_cfg.get("momentum", 0) or opt_cfg.get("weight_decay", 0)
+            or opt_cfg.get("nesterov", False)):
+        return None
+    if self._mega_resnet is None:
+        from ..ops import HAS_EXT
+        from ..ops.mega_resnet import ResNetMegaRound, matches_resnet18
+        cpg = int(prim.model_config.get("group_norm", 0) or 0)
+        if (not HAS_EXT or cpg <= 0
+                or matches_resnet18(prim.arena) is None):
+            self._mega_resnet = False  # not this model: stop probing
+            return None
+        self._mega_resnet = ResNetMegaRound(
+            prim.arena, data_cfg.get("batch_size", 20),
+            data_cfg.get("max_grad_norm"), cpg)
+    if not self._mega_resnet.supports(len(client_ids)):
+        return None
+    store = prim._get_shard_store(data_cfg)
+    if store is None or store.x.dim() != 4:
+        return None
+    import msrflute_amd.core.client as cm
+    ds = cm.train_dataset
+    if ds is None:
+        return None
+    dms = data_cfg.get("desired_max_samples", None)
+    for cid in client_ids:
+        i = store.user_pos.get(ds.user_list[cid])
+        if i is None:
+            return None
+        n = store.offsets[i + 1] - store.offsets[i]
+        if n == 0 or (dms is not None and n > dms):
+            return None
+    out = self._mega_resnet.run(store, ds, client_ids, seeds, initial_lr,
+                                self.server_arena, self.round_accums[0])
+    if out is not None:
+        self._streams_dirty = True
+        prim.perf_acc["clients"] = (prim.perf_acc.get("clients", 0)
+                                    + len(client_ids))
+    return out
+
+
 def _pool_run_fused_round_batch(self, client_ids, initial_lr, iteration,
                                 seeds):
     """ClientPool: one _C.cnn_round per executor (on its stream) covering
@@ -747,8 +802,13 @@ def _pool_run_fused_round_batch(self, client_ids, initial_lr, iteration,
         return None
     if getattr(self, "_mega_lstm", "missing") == "missing":
         self._mega_lstm = None
+    if getattr(self, "_mega_resnet", "missing") == "missing":
+        self._mega_resnet = None
     out = _try_mega_lstm_round(self, prim, data_cfg, client_ids,
                                initial_lr, seeds)
+    if out is None:
+        out = _try_mega_resnet_round(self, prim, data_cfg, client_ids,
+                                     initial_lr, seeds)
     if out is not None:
         return out
     if prim.fused_cnn is None:
