@@ -198,6 +198,10 @@ class ResNetMegaRound:
         return g
 
     def _capture(self, g):
+        # exhaustive MIOpen find during warmup: grouped-conv algorithms
+        # from immediate mode are fallbacks; the capture then bakes the
+        # found kernels (find cost is paid once, outside the hot loop)
+        torch.backends.cudnn.benchmark = True
         if os.environ.get("MEGA_RESNET_EAGER") == "1":
             return
         s = torch.cuda.Stream()
