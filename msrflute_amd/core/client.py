@@ -738,8 +738,17 @@ def _try_mega_resnet_round(self, prim, data_cfg, client_ids, initial_lr,
                            seeds):
     """Cross-client MEGA round for the fed-CIFAR100 ResNet-18
     (ops/mega_resnet.py): all K clients' epochs as one graph-captured
-    launch set of grouped convs + GroupNorm over K-stacked weights."""
+    launch set of grouped convs + GroupNorm over K-stacked weights.
+
+    OPT-IN (use_mega_round_resnet): measured SLOWER than the per-client
+    epoch-graph path on ROCm 7.2 — MIOpen decomposes groups=K convs into
+    K per-group kernels (no batching win) and grouped bwd-weight falls
+    to multi-ms CK batched-GEMM fallbacks (PERF.md, profiles/).  The
+    formulation itself is exact (tests/test_mega_cpu.py, f64) and flips
+    on wholesale when a ROCm release makes grouped conv competitive."""
     if self._mega_resnet is False:
+        return None
+    if not prim.client_config.get("use_mega_round_resnet", False):
         return None
     if not prim.client_config.get("use_mega_round", True):
         return None
