@@ -374,13 +374,22 @@ class Trainer(TrainerBase):
 
         # probe one batch: graphs need dict batches of plain (x, y) tensors
         # (nlg/mlm/newsrec batch shapes run the generic eager path).
-        # The probe draws through the loader's shuffle, which consumes a
-        # randperm from the per-client-seeded host RNG — restore the state
-        # so every path (eager, graphed, fused round, mega round) trains
-        # on the SAME first-draw shuffle order for a given (round, client)
-        rng_state = torch.get_rng_state()
-        probe = next(iter(self.train_dataloader.create_loader()), None)
-        torch.set_rng_state(rng_state)
+        # The probe must NOT consume the per-client-seeded host RNG (a
+        # shuffled draw would shift the epoch's randperm, diverging this
+        # path from the eager/fused/mega paths' first-draw order): peek
+        # with shuffle off when the loader supports it, else save/restore
+        # the RNG state (state clone costs ~0.5 ms — only for the generic
+        # loaders, whose tasks run eager anyway)
+        dl = self.train_dataloader
+        if hasattr(dl, "shuffle"):
+            sh = dl.shuffle
+            dl.shuffle = False
+            probe = next(iter(dl.create_loader()), None)
+            dl.shuffle = sh
+        else:
+            rng_state = torch.get_rng_state()
+            probe = next(iter(dl.create_loader()), None)
+            torch.set_rng_state(rng_state)
         if not (isinstance(probe, dict) and torch.is_tensor(probe.get("x"))
                 and torch.is_tensor(probe.get("y"))):
             self.graph_cache = None
