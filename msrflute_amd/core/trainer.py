@@ -372,14 +372,34 @@ class Trainer(TrainerBase):
         g = None
         n_graph_batches = 0
 
-        # probe one batch: graphs need dict batches of plain (x, y) tensors
-        # (nlg/mlm/newsrec batch shapes run the generic eager path).
-        # The probe must NOT consume the per-client-seeded host RNG (a
-        # shuffled draw would shift the epoch's randperm, diverging this
-        # path from the eager/fused/mega paths' first-draw order): peek
-        # with shuffle off when the loader supports it, else save/restore
-        # the RNG state (state clone costs ~0.5 ms — only for the generic
-        # loaders, whose tasks run eager anyway)
+        cache.set_lr(get_lr(self.optimizer) if self.optimizer is not None
+                     else float(cache.lr_t[0]))
+
+        # whole-epoch fast path: device-resident uniform shard ⇒ the entire
+        # local epoch is ONE graph replay (ops/graphs.py GraphedClientEpoch).
+        # Attempted FIRST — it needs no probe batch, and the probe costs
+        # real host time on fast tasks (an extra loader iteration per
+        # client on a ~5 ms round)
+        ds = getattr(self.train_dataloader, "dataset", None)
+        bs = getattr(self.train_dataloader, "batch_size", 0)
+        if (ds is not None and bs
+                and torch.is_tensor(getattr(ds, "x", None)) and ds.x.is_cuda
+                and torch.is_tensor(getattr(ds, "y", None))
+                and getattr(self.train_dataloader, "shuffle", False)
+                and (desired_max_samples is None
+                     or desired_max_samples >= len(ds.x))):
+            from ..ops.graphs import epoch_graph_for
+            eg = epoch_graph_for(cache, ds.x, ds.y, bs)
+            if eg is not None:
+                return self._run_epoch_one_graph(eg, ds, bs, cache)
+
+        # probe one batch: per-batch graphs need dict batches of plain
+        # (x, y) tensors (nlg/mlm/newsrec batch shapes run the generic
+        # eager path).  The probe must NOT consume the per-client-seeded
+        # host RNG (a shuffled draw would shift the epoch's randperm,
+        # diverging this path from the eager/fused/mega paths' first-draw
+        # order): peek with shuffle off when the loader supports it, else
+        # save/restore the RNG state
         dl = self.train_dataloader
         if hasattr(dl, "shuffle"):
             sh = dl.shuffle
@@ -394,24 +414,6 @@ class Trainer(TrainerBase):
                 and torch.is_tensor(probe.get("y"))):
             self.graph_cache = None
             return self.run_train_epoch(desired_max_samples)
-
-        cache.set_lr(get_lr(self.optimizer) if self.optimizer is not None
-                     else float(cache.lr_t[0]))
-
-        # whole-epoch fast path: device-resident uniform shard ⇒ the entire
-        # local epoch is ONE graph replay (ops/graphs.py GraphedClientEpoch)
-        ds = getattr(self.train_dataloader, "dataset", None)
-        bs = getattr(self.train_dataloader, "batch_size", 0)
-        if (ds is not None and bs
-                and torch.is_tensor(getattr(ds, "x", None)) and ds.x.is_cuda
-                and torch.is_tensor(getattr(ds, "y", None))
-                and getattr(self.train_dataloader, "shuffle", False)
-                and (desired_max_samples is None
-                     or desired_max_samples >= len(ds.x))):
-            from ..ops.graphs import epoch_graph_for
-            eg = epoch_graph_for(cache, ds.x, ds.y, bs)
-            if eg is not None:
-                return self._run_epoch_one_graph(eg, ds, bs, cache)
 
         train_loader = self.train_dataloader.create_loader()
         for batch in train_loader:
