@@ -110,7 +110,6 @@ class ShakespeareMegaRound:
         self.V, self.E = vc
         self.lr_t = torch.zeros(1, dtype=torch.float32, device=arena.device)
         self._graphs: Dict[Tuple, dict] = {}
-        self._pins = None
 
     def supports(self, K: int) -> bool:
         return 0 < K <= self.k_cap
