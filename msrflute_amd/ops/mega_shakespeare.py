@@ -10,11 +10,12 @@ ALL K sampled clients train together with K-stacked per-client weights:
   is a strided view into it, so autograd accumulates a [K, P] gradient
   stack and the per-client clip + sufficient-stats + SGD is the SAME
   gfx950 kernel suite as the CNN mega round (_C.mega_clip_sgd);
-* batched compute: per-client embedding via one offset F.embedding,
-  input/output projections via torch.bmm over the K-stacked weights
-  (hipBLASLt batched GEMMs), and the T-step recurrences via the
-  cross-client kernels (_C.lstm_seq_fwd_b/_bwd_b — grid = all K*bs rows,
-  each block indexing its client's W_hh);
+* batched compute: per-client embedding via one offset index_select
+  (value-independent backward — capture-safe), input/output projections
+  via torch.bmm over the K-stacked weights (hipBLASLt batched GEMMs),
+  and the T-step recurrences via the cross-client kernels
+  (_C.lstm_seq_fwd_b/_bwd_b — grid = all K*bs rows, each block indexing
+  its client's float4-packed W_hh);
 * the WHOLE local epoch (all batch-steps, forward + autograd backward +
   clip/SGD) is captured as ONE hipGraph keyed by (K, steps, bs, T):
   per round the host refreshes the static gather-index/label buffers
