@@ -206,6 +206,14 @@ void k_lstm_seq_fwd_b(const float* __restrict__ xp,
   const float4* hp4 = reinterpret_cast<const float4*>(h_prev);
   if (tid < LSTM_H) h_prev[tid] = 0.f;
   float c = 0.f;
+  // weights are time-invariant: keep the first WREG float4 groups of
+  // this thread's column resident in registers across ALL T steps
+  // (removes 1/4 of the latency-bound per-step loads)
+  #define WREG_F 16
+  float4 wreg[WREG_F];
+  #pragma unroll
+  for (int kk = 0; kk < WREG_F; ++kk)
+    wreg[kk] = wp[(long long)kk * 4 * LSTM_H + tid];
   __syncthreads();
   for (int t = 0; t < T; ++t) {
     const float* xr = xp + ((long long)b * T + t) * 4 * LSTM_H;
@@ -213,8 +221,16 @@ void k_lstm_seq_fwd_b(const float* __restrict__ xp,
     // + 4 independent accumulator chains: the b32 form was bound by
     // exposed L2 latency x load count (256 loads/step/thread -> 64)
     float s0 = xr[tid], s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    #pragma unroll
+    for (int kk = 0; kk < WREG_F; ++kk) {
+      float4 hv = hp4[kk];
+      s0 = fmaf(wreg[kk].x, hv.x, s0);
+      s1 = fmaf(wreg[kk].y, hv.y, s1);
+      s2 = fmaf(wreg[kk].z, hv.z, s2);
+      s3 = fmaf(wreg[kk].w, hv.w, s3);
+    }
     #pragma unroll 16
-    for (int kk = 0; kk < LSTM_H / 4; ++kk) {
+    for (int kk = WREG_F; kk < LSTM_H / 4; ++kk) {
       float4 wv = wp[(long long)kk * 4 * LSTM_H + tid];
       float4 hv = hp4[kk];
       s0 = fmaf(wv.x, hv.x, s0);
@@ -260,6 +276,13 @@ void k_lstm_seq_bwd_b(const float* __restrict__ gates,
   const float4* dgp4 = reinterpret_cast<const float4*>(dg_l);
   float dc = 0.f;
   if (tid < LSTM_H) dh_rec_l[tid] = 0.f;
+  // register-resident first WREG_B weight groups of this thread's
+  // quarter (time-invariant across the BPTT loop)
+  #define WREG_B 16
+  float4 wregB[WREG_B];
+  #pragma unroll
+  for (int jj = 0; jj < WREG_B; ++jj)
+    wregB[jj] = wpB[(long long)(q * (LSTM_H / 4) + jj) * LSTM_H + h];
   __syncthreads();
   for (int t = T - 1; t >= 0; --t) {
     long long base = ((long long)b * T + t) * 4 * LSTM_H;
@@ -290,8 +313,16 @@ void k_lstm_seq_bwd_b(const float* __restrict__ gates,
     }
     __syncthreads();
     float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    #pragma unroll
+    for (int jj = 0; jj < WREG_B; ++jj) {
+      float4 dv = dgp4[q * (LSTM_H / 4) + jj];
+      s0 = fmaf(wregB[jj].x, dv.x, s0);
+      s1 = fmaf(wregB[jj].y, dv.y, s1);
+      s2 = fmaf(wregB[jj].z, dv.z, s2);
+      s3 = fmaf(wregB[jj].w, dv.w, s3);
+    }
     #pragma unroll 16
-    for (int jj = 0; jj < LSTM_H / 4; ++jj) {
+    for (int jj = WREG_B; jj < LSTM_H / 4; ++jj) {
       int jg = q * (LSTM_H / 4) + jj;    // float4 group of 4 j's
       float4 wv = wpB[(long long)jg * LSTM_H + h];
       float4 dv = dgp4[jg];
