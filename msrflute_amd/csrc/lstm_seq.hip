@@ -56,6 +56,13 @@ void k_lstm_seq_fwd(const float* __restrict__ xp,
   const float4* hp4 = reinterpret_cast<const float4*>(h_prev);
   if (tid < LSTM_H) h_prev[tid] = 0.f;
   float c = 0.f;                   // live in threads tid < H only
+  // first 16 float4 weight groups register-resident across all T steps
+  // (W_hh is time-invariant; same trick as the batched variant below —
+  // 16 is the 128-VGPR/16-wave occupancy limit, 24 spills)
+  float4 wregS[16];
+  #pragma unroll
+  for (int kk = 0; kk < 16; ++kk)
+    wregS[kk] = wp[(long long)kk * 4 * LSTM_H + tid];
   __syncthreads();
   for (int t = 0; t < T; ++t) {
     const float* xr = xp + ((long long)b * T + t) * 4 * LSTM_H;
@@ -63,8 +70,16 @@ void k_lstm_seq_fwd(const float* __restrict__ xp,
     // + 4 independent accumulator chains: the b32 form was bound by
     // exposed L2 latency x load count (256 loads/step/thread -> 64)
     float s0 = xr[tid], s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    #pragma unroll
+    for (int kk = 0; kk < 16; ++kk) {
+      float4 hv = hp4[kk];
+      s0 = fmaf(wregS[kk].x, hv.x, s0);
+      s1 = fmaf(wregS[kk].y, hv.y, s1);
+      s2 = fmaf(wregS[kk].z, hv.z, s2);
+      s3 = fmaf(wregS[kk].w, hv.w, s3);
+    }
     #pragma unroll 16
-    for (int kk = 0; kk < LSTM_H / 4; ++kk) {
+    for (int kk = 16; kk < LSTM_H / 4; ++kk) {
       float4 wv = wp[(long long)kk * 4 * LSTM_H + tid];
       float4 hv = hp4[kk];
       s0 = fmaf(wv.x, hv.x, s0);
@@ -111,6 +126,11 @@ void k_lstm_seq_bwd(const float* __restrict__ gates,
   const float4* dgp4 = reinterpret_cast<const float4*>(dg_l);
   float dc = 0.f;                            // live in threads tid < H
   if (tid < LSTM_H) dh_rec_l[tid] = 0.f;
+  // register-resident first 16 weight groups of this thread's quarter
+  float4 wregSB[16];
+  #pragma unroll
+  for (int jj = 0; jj < 16; ++jj)
+    wregSB[jj] = wpB[(long long)(q * (LSTM_H / 4) + jj) * LSTM_H + h];
   __syncthreads();
   for (int t = T - 1; t >= 0; --t) {
     long long base = ((long long)b * T + t) * 4 * LSTM_H;
@@ -143,8 +163,16 @@ void k_lstm_seq_bwd(const float* __restrict__ gates,
     // dh_rec = W_hh^T dg: thread (q, h) sums its 256-row quarter with
     // lane-coalesced w_hh[j*H + h] reads; 16 waves hide the L2 latency
     float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    #pragma unroll
+    for (int jj = 0; jj < 16; ++jj) {
+      float4 dv = dgp4[q * (LSTM_H / 4) + jj];
+      s0 = fmaf(wregSB[jj].x, dv.x, s0);
+      s1 = fmaf(wregSB[jj].y, dv.y, s1);
+      s2 = fmaf(wregSB[jj].z, dv.z, s2);
+      s3 = fmaf(wregSB[jj].w, dv.w, s3);
+    }
     #pragma unroll 16
-    for (int jj = 0; jj < LSTM_H / 4; ++jj) {
+    for (int jj = 16; jj < LSTM_H / 4; ++jj) {
       int jg = q * (LSTM_H / 4) + jj;    // float4 group of 4 j's
       float4 wv = wpB[(long long)jg * LSTM_H + h];
       float4 dv = dgp4[jg];
