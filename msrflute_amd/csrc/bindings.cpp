@@ -538,8 +538,9 @@ std::vector<torch::Tensor> gru_seq_fwd(torch::Tensor gi, torch::Tensor whh_t,
                                        torch::Tensor b_hh) {
   check_flat(gi, "gi"); check_flat(whh_t, "whh_t"); check_flat(b_hh, "b_hh");
   TORCH_CHECK(gi.dim() == 3 && gi.size(2) == 3 * 512 &&
-              whh_t.size(0) == 512 && whh_t.size(1) == 3 * 512,
-              "fused GRU supports hidden size 512; pass W_hh transposed");
+              whh_t.numel() == 512 * 3 * 512,
+              "fused GRU supports hidden size 512; pass W_hh "
+              "float4-packed [H/4, 3H, 4] (ops/lstm._GRUSeq)");
   long long B = gi.size(0), T = gi.size(1);
   auto h_seq = torch::empty({B, T, 512}, gi.options());
   auto gates = torch::empty({B, T, 3 * 512}, gi.options());
@@ -558,6 +559,8 @@ std::vector<torch::Tensor> gru_seq_bwd(torch::Tensor gates, torch::Tensor ghn,
   check_flat(gates, "gates"); check_flat(ghn, "ghn");
   check_flat(h_seq, "h_seq"); check_flat(w_hh, "w_hh");
   check_flat(dh_out, "dh_out");
+  TORCH_CHECK(w_hh.numel() == 3 * 512 * 512,
+              "gru_seq_bwd wants W_hh float4-packed [3H/4, H, 4]");
   long long B = gates.size(0), T = gates.size(1);
   auto dgi = torch::empty_like(gates);
   auto dgh = torch::empty_like(gates);

@@ -98,15 +98,20 @@ class _GRUSeq(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, gi, w_hh, b_hh):
+        # float4-packed layouts (see _pack_fwd/_pack_bwd; 3H gate blocks)
+        H3, H = w_hh.shape
+        w_p = w_hh.t().reshape(H // 4, 4, H3).permute(0, 2, 1).contiguous()
         h_seq, gates, ghn = _C.gru_seq_fwd(
-            gi.contiguous(), w_hh.t().contiguous(), b_hh.contiguous())
+            gi.contiguous(), w_p, b_hh.contiguous())
         ctx.save_for_backward(gates, ghn, h_seq, w_hh)
         return h_seq
 
     @staticmethod
     def backward(ctx, dh):
         gates, ghn, h_seq, w_hh = ctx.saved_tensors
-        dgi, dgh = _C.gru_seq_bwd(gates, ghn, h_seq, w_hh, dh.contiguous())
+        H3, H = w_hh.shape
+        w_pb = w_hh.reshape(H3 // 4, 4, H).permute(0, 2, 1).contiguous()
+        dgi, dgh = _C.gru_seq_bwd(gates, ghn, h_seq, w_pb, dh.contiguous())
         B, T, H = h_seq.shape
         h_prev = torch.cat([h_seq.new_zeros(B, 1, H), h_seq[:, :-1]], dim=1)
         dw_hh = dgh.reshape(-1, 3 * H).t().mm(h_prev.reshape(-1, H))
